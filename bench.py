@@ -1,0 +1,161 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: TIGER training step on Amazon-Beauty-shaped synthetic
+data (BASELINE.json: "Recall@10 + train samples/sec, TIGER Amazon-Beauty at
+1/2/4/8 MI355X").
+
+Measures whole-job train samples/sec for the reference's shipped TIGER
+config (config/tiger/amazon/tiger.gin: d_model 128 -> attn 384, 6 heads,
+4+4 layers, sem_id_dim 3, codebook 256, batch 256/GPU, 20-item histories =
+61 encoder tokens), bf16 autocast, full fwd+bwd+clip+AdamW step per
+iteration, DDP over RCCL/xGMI for N>1 (weak scaling: per-GPU batch fixed).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+For N>1 the driver launches via torch.distributed.run with one rank per GPU.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def build_batch(batch_size: int, n_items: int, sem_id_dim: int,
+                codebook_size: int, num_users: int, device, seed: int):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    L = n_items * sem_id_dim
+    item_ids = torch.randint(0, codebook_size, (batch_size, L), generator=g)
+    ttype = (torch.arange(L) % sem_id_dim).unsqueeze(0).expand(batch_size, -1)
+    tgt = torch.randint(0, codebook_size, (batch_size, sem_id_dim), generator=g)
+    tgt_type = torch.arange(sem_id_dim).unsqueeze(0).expand(batch_size, -1)
+    user = torch.randint(0, num_users, (batch_size, 1), generator=g)
+    # realistic ragged histories: ~30% of rows left-padded shorter
+    mask = torch.ones(batch_size, L, dtype=torch.long)
+    lens = torch.randint(5, n_items + 1, (batch_size,), generator=g)
+    for i in range(0, batch_size, 3):
+        mask[i, lens[i] * sem_id_dim:] = 0
+    return {
+        "user_input_ids": user.to(device),
+        "item_input_ids": item_ids.to(device),
+        "token_type_ids": ttype.contiguous().to(device),
+        "target_input_ids": tgt.to(device),
+        "target_token_type_ids": tgt_type.contiguous().to(device),
+        "seq_mask": mask.to(device),
+    }
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=256)
+    p.add_argument("--model", type=str, default="tiger")
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+
+    from genrec_amd.models.tiger import Tiger
+    from genrec_amd.parallel import GradReducer, init_distributed
+    from genrec_amd.parallel.ddp import broadcast_parameters
+
+    ctx = init_distributed()
+    device = ctx.device
+    use_gpu = device.type == "cuda"
+    torch.manual_seed(1234)
+
+    # reference TIGER Amazon-Beauty architecture (tiger.gin:16-25)
+    cfg = dict(embedding_dim=128, attn_dim=384, dropout=0.1, num_heads=6,
+               n_layers=8, num_item_embeddings=256, num_user_embeddings=10000,
+               sem_id_dim=3)
+    model = Tiger(**cfg).to(device)
+    broadcast_parameters(model)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-4, weight_decay=0.035)
+    reducer = GradReducer(model)
+
+    n_items_hist = 20
+    batches = [
+        build_batch(args.batch_size, n_items_hist, cfg["sem_id_dim"],
+                    cfg["num_item_embeddings"], cfg["num_user_embeddings"],
+                    device, seed=100 + rank * 1000 + i)
+        for i in range(8)
+    ]
+
+    amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16) \
+        if use_gpu else _null_ctx()
+
+    def step(i: int) -> None:
+        b = batches[i % len(batches)]
+        opt.zero_grad(set_to_none=False)
+        with amp:
+            out = model(**b)
+        out.loss.backward()
+        reducer.finalize()
+        torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+        opt.step()
+
+    model.train()
+    for i in range(args.warmup):
+        step(i)
+
+    ctx.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if use_gpu:
+        torch.cuda.synchronize()
+    ctx.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device
+                         if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    global_batch = args.batch_size * world
+    samples_per_s = global_batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train_samples_per_s",
+            "value": samples_per_s,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "tiger-amazon-beauty",
+                "global_batch": global_batch,
+                "seq_len": n_items_hist * cfg["sem_id_dim"] + 1,
+                "parallelism": f"dp{world}",
+            },
+        }))
+
+
+class _null_ctx:
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
+
+
+if __name__ == "__main__":
+    main()

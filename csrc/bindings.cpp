@@ -1,0 +1,52 @@
+// genrec_amd._C — single op library for the CDNA4 kernel layer.
+#include <torch/extension.h>
+
+namespace genrec {
+
+std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
+                                        double eps, bool t5_style);
+std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor w, torch::Tensor inv_rms,
+                                        bool t5_style);
+std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x, double eps);
+torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor x,
+                         torch::Tensor inv_norm, double eps);
+
+std::vector<torch::Tensor> attn_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
+    c10::optional<torch::Tensor> add_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, bool causal, int64_t act, double dropout_p, int64_t seed);
+std::vector<torch::Tensor> attn_bwd(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor p_saved, torch::Tensor drop_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, int64_t act, double dropout_p, int64_t seed,
+    bool bias_grad, int64_t bias_dim);
+
+std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
+                                          torch::Tensor targets,
+                                          int64_t ignore_index);
+torch::Tensor softmax_ce_bwd(torch::Tensor dloss, torch::Tensor logits,
+                             torch::Tensor targets, torch::Tensor lse,
+                             torch::Tensor n_valid, int64_t ignore_index);
+
+std::vector<torch::Tensor> sqdist_argmin(torch::Tensor x,
+                                         torch::Tensor codebook);
+torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
+
+}  // namespace genrec
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm_fwd", &genrec::rms_norm_fwd, "fused RMSNorm forward");
+  m.def("rms_norm_bwd", &genrec::rms_norm_bwd, "fused RMSNorm backward");
+  m.def("l2norm_fwd", &genrec::l2norm_fwd, "fused L2Norm forward");
+  m.def("l2norm_bwd", &genrec::l2norm_bwd, "fused L2Norm backward");
+  m.def("attn_fwd", &genrec::attn_fwd, "fused attention forward");
+  m.def("attn_bwd", &genrec::attn_bwd, "fused attention backward");
+  m.def("softmax_ce_fwd", &genrec::softmax_ce_fwd, "fused CE forward");
+  m.def("softmax_ce_bwd", &genrec::softmax_ce_bwd, "fused CE backward");
+  m.def("sqdist_argmin", &genrec::sqdist_argmin, "L2 dist + argmin");
+  m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
+}

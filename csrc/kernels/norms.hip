@@ -1,0 +1,306 @@
+// Fused norm kernels (K24 — SURVEY.md §2.4): RMSNorm (standard + T5 style)
+// and L2Norm, forward + backward, fp32 and bf16, gfx950.
+//
+// Design: one 64-lane wave per row (rows are small: D = 32..768 across the
+// zoo), 4 waves per 256-thread block, grid-stride over rows. bf16 rows are
+// loaded vectorized (short4 reinterpret = 8 B/lane) per Guideline 13; all
+// math accumulates in fp32, matching the reference's fp32-upcast semantics
+// (normalize.py:38-55, 73-95). Backward reduces dweight per-block in LDS
+// then atomically into a fp32 accumulator.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../core/common.h"
+
+namespace genrec {
+
+// ---------------------------------------------------------------- RMSNorm
+
+template <typename T, typename OutT, bool T5_STYLE>
+__global__ void rms_norm_fwd_kernel(const T* __restrict__ x,
+                                    const float* __restrict__ w_f32,
+                                    OutT* __restrict__ y,
+                                    float* __restrict__ inv_rms,
+                                    int64_t n_rows, int d, float eps,
+                                    bool w_is_half) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = gridDim.x * blockDim.x / WAVE;
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const T* xr = x + row * d;
+    float ss = 0.f;
+    for (int j = lane; j < d; j += WAVE) {
+      float v = to_f32(xr[j]);
+      ss += v * v;
+    }
+    ss = wave_sum(ss);
+    float r = rsqrtf(ss / d + eps);
+    if (lane == 0) inv_rms[row] = r;
+    OutT* yr = y + row * d;
+    for (int j = lane; j < d; j += WAVE) {
+      float xf = to_f32(xr[j]);
+      float wf = w_f32[j];
+      float out;
+      if (T5_STYLE) {
+        // t = x * r (promote), optional cast to weight dtype, then w * t
+        float t = to_f32(xr[j]) * r;
+        if (w_is_half) t = to_f32(from_f32<__hip_bfloat16>(t));
+        out = wf * t;
+      } else {
+        // y = cast_to_xdtype(xf * r) * w
+        float t = xf * r;
+        t = to_f32(from_f32<T>(t));
+        out = t * wf;
+      }
+      yr[j] = from_f32<OutT>(out);
+    }
+  }
+}
+
+// backward: dx_i = w_i*dy_i*r - r^3/d * x_i * sum_j(w_j*dy_j*x_j)
+// (intermediate rounding of the forward's casts is ignored in backward, as
+// eager autograd does for the same graph up to bf16 rounding)
+template <typename T, typename OutT>
+__global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
+                                    const T* __restrict__ x,
+                                    const float* __restrict__ w_f32,
+                                    const float* __restrict__ inv_rms,
+                                    T* __restrict__ dx,
+                                    float* __restrict__ dw,
+                                    int64_t n_rows, int d) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* dw_local = reinterpret_cast<float*>(smem_raw);  // [d]
+  for (int j = threadIdx.x; j < d; j += blockDim.x) dw_local[j] = 0.f;
+  __syncthreads();
+
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = (int64_t)blockIdx.x * waves_per_block + wave_in_block;
+  const int64_t n_waves = (int64_t)gridDim.x * waves_per_block;
+
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const T* xr = x + row * d;
+    const OutT* dyr = dy + row * d;
+    float r = inv_rms[row];
+    float dot = 0.f;
+    for (int j = lane; j < d; j += WAVE) {
+      dot += w_f32[j] * to_f32(dyr[j]) * to_f32(xr[j]);
+    }
+    dot = wave_sum(dot);
+    float c = r * r * r / d * dot;
+    T* dxr = dx + row * d;
+    for (int j = lane; j < d; j += WAVE) {
+      float xf = to_f32(xr[j]);
+      float dyf = to_f32(dyr[j]);
+      dxr[j] = from_f32<T>(w_f32[j] * dyf * r - c * xf);
+      atomicAdd(&dw_local[j], dyf * xf * r);
+    }
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < d; j += blockDim.x) {
+    if (dw_local[j] != 0.f) atomicAdd(&dw[j], dw_local[j]);
+  }
+}
+
+// ---------------------------------------------------------------- L2Norm
+
+template <typename T>
+__global__ void l2norm_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                  float* __restrict__ inv_norm, int64_t n_rows,
+                                  int d, float eps) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = gridDim.x * blockDim.x / WAVE;
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const T* xr = x + row * d;
+    float ss = 0.f;
+    for (int j = lane; j < d; j += WAVE) {
+      float v = to_f32(xr[j]);
+      ss += v * v;
+    }
+    ss = wave_sum(ss);
+    // F.normalize: x / max(||x||, eps)
+    float inv = 1.0f / fmaxf(sqrtf(ss), eps);
+    if (lane == 0) inv_norm[row] = inv;
+    T* yr = y + row * d;
+    for (int j = lane; j < d; j += WAVE) {
+      yr[j] = from_f32<T>(to_f32(xr[j]) * inv);
+    }
+  }
+}
+
+// dx = inv*(dy - y * dot(dy, y))  with y = x*inv  (when not eps-clamped;
+// clamped rows: dx = inv*dy since norm is constant w.r.t x below eps — the
+// clamp branch gradient through max() is zero. F.normalize autograd
+// behaves the same way for ||x|| < eps.)
+template <typename T>
+__global__ void l2norm_bwd_kernel(const T* __restrict__ dy,
+                                  const T* __restrict__ x,
+                                  const float* __restrict__ inv_norm,
+                                  T* __restrict__ dx, int64_t n_rows, int d,
+                                  float eps) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = gridDim.x * blockDim.x / WAVE;
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const T* xr = x + row * d;
+    const T* dyr = dy + row * d;
+    float inv = inv_norm[row];
+    float dot = 0.f, ss = 0.f;
+    for (int j = lane; j < d; j += WAVE) {
+      float xf = to_f32(xr[j]);
+      dot += to_f32(dyr[j]) * xf;
+      ss += xf * xf;
+    }
+    dot = wave_sum(dot);
+    ss = wave_sum(ss);
+    bool clamped = sqrtf(ss) < eps;
+    T* dxr = dx + row * d;
+    for (int j = lane; j < d; j += WAVE) {
+      float xf = to_f32(xr[j]);
+      float g = to_f32(dyr[j]) * inv;
+      if (!clamped) g -= xf * inv * inv * inv * dot;
+      dxr[j] = from_f32<T>(g);
+    }
+  }
+}
+
+// ---------------------------------------------------------------- host
+
+static int grid_for_rows(int64_t n_rows, int waves_per_block) {
+  int64_t blocks = (n_rows + waves_per_block - 1) / waves_per_block;
+  return (int)std::min<int64_t>(blocks, 8192);
+}
+
+std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
+                                        double eps, bool t5_style) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int d = x.size(-1);
+  const int64_t n_rows = x.numel() / d;
+  auto w_f32 = w.to(torch::kFloat32).contiguous();
+  bool w_half = w.scalar_type() == torch::kBFloat16 ||
+                w.scalar_type() == torch::kHalf;
+  // output dtype follows eager semantics: promote(x.dtype, w.dtype)
+  auto out_dtype = at::result_type(x, w);
+  auto y = torch::empty(x.sizes(), x.options().dtype(out_dtype));
+  auto inv_rms = torch::empty({n_rows}, x.options().dtype(torch::kFloat32));
+  dim3 block(256);
+  dim3 grid(grid_for_rows(n_rows, 4));
+  auto stream = at::cuda::getCurrentHIPStream();
+
+#define LAUNCH_RMS(T, OutT, T5)                                                \
+  hipLaunchKernelGGL((rms_norm_fwd_kernel<T, OutT, T5>), grid, block, 0,       \
+                     stream, reinterpret_cast<const T*>(x.data_ptr()),         \
+                     w_f32.data_ptr<float>(),                                  \
+                     reinterpret_cast<OutT*>(y.data_ptr()),                    \
+                     inv_rms.data_ptr<float>(), n_rows, d, (float)eps, w_half)
+
+  if (x.scalar_type() == torch::kFloat32) {
+    if (t5_style) LAUNCH_RMS(float, float, true);
+    else LAUNCH_RMS(float, float, false);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    if (out_dtype == torch::kFloat32) {
+      if (t5_style) LAUNCH_RMS(__hip_bfloat16, float, true);
+      else LAUNCH_RMS(__hip_bfloat16, float, false);
+    } else {
+      if (t5_style) LAUNCH_RMS(__hip_bfloat16, __hip_bfloat16, true);
+      else LAUNCH_RMS(__hip_bfloat16, __hip_bfloat16, false);
+    }
+  } else {
+    TORCH_CHECK(false, "rms_norm: unsupported dtype");
+  }
+#undef LAUNCH_RMS
+  return {y, inv_rms};
+}
+
+std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor w, torch::Tensor inv_rms,
+                                        bool t5_style) {
+  (void)t5_style;  // same analytic gradient for both styles
+  const int d = x.size(-1);
+  const int64_t n_rows = x.numel() / d;
+  auto w_f32 = w.to(torch::kFloat32).contiguous();
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({d}, x.options().dtype(torch::kFloat32));
+  dim3 block(256);
+  dim3 grid(grid_for_rows(n_rows, 4));
+  size_t smem = d * sizeof(float);
+  auto stream = at::cuda::getCurrentHIPStream();
+
+#define LAUNCH_RMSB(T, OutT)                                                   \
+  hipLaunchKernelGGL((rms_norm_bwd_kernel<T, OutT>), grid, block, smem,        \
+                     stream, reinterpret_cast<const OutT*>(dy.data_ptr()),     \
+                     reinterpret_cast<const T*>(x.data_ptr()),                 \
+                     w_f32.data_ptr<float>(), inv_rms.data_ptr<float>(),       \
+                     reinterpret_cast<T*>(dx.data_ptr()),                      \
+                     dw.data_ptr<float>(), n_rows, d)
+
+  if (x.scalar_type() == torch::kFloat32) {
+    LAUNCH_RMSB(float, float);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    if (dy.scalar_type() == torch::kFloat32) LAUNCH_RMSB(__hip_bfloat16, float);
+    else LAUNCH_RMSB(__hip_bfloat16, __hip_bfloat16);
+  } else {
+    TORCH_CHECK(false, "rms_norm_bwd: unsupported dtype");
+  }
+#undef LAUNCH_RMSB
+  auto dw_out = dw.to(w.scalar_type());
+  return {dx, dw_out};
+}
+
+std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int d = x.size(-1);
+  const int64_t n_rows = x.numel() / d;
+  auto y = torch::empty_like(x);
+  auto inv_norm = torch::empty({n_rows}, x.options().dtype(torch::kFloat32));
+  dim3 block(256);
+  dim3 grid(grid_for_rows(n_rows, 4));
+  auto stream = at::cuda::getCurrentHIPStream();
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((l2norm_fwd_kernel<float>), grid, block, 0, stream,
+                       x.data_ptr<float>(), y.data_ptr<float>(),
+                       inv_norm.data_ptr<float>(), n_rows, d, (float)eps);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((l2norm_fwd_kernel<__hip_bfloat16>), grid, block, 0,
+                       stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                       inv_norm.data_ptr<float>(), n_rows, d, (float)eps);
+  } else {
+    TORCH_CHECK(false, "l2norm: unsupported dtype");
+  }
+  return {y, inv_norm};
+}
+
+torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor x,
+                         torch::Tensor inv_norm, double eps_in) {
+  const int d = x.size(-1);
+  const int64_t n_rows = x.numel() / d;
+  auto dx = torch::empty_like(x);
+  dim3 block(256);
+  dim3 grid(grid_for_rows(n_rows, 4));
+  auto stream = at::cuda::getCurrentHIPStream();
+  float eps = (float)eps_in;
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((l2norm_bwd_kernel<float>), grid, block, 0, stream,
+                       dy.data_ptr<float>(), x.data_ptr<float>(),
+                       inv_norm.data_ptr<float>(), dx.data_ptr<float>(),
+                       n_rows, d, eps);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((l2norm_bwd_kernel<__hip_bfloat16>), grid, block, 0,
+                       stream,
+                       reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       inv_norm.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()),
+                       n_rows, d, eps);
+  } else {
+    TORCH_CHECK(false, "l2norm_bwd: unsupported dtype");
+  }
+  return dx;
+}
+
+}  // namespace genrec

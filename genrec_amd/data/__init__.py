@@ -1,0 +1,16 @@
+from genrec_amd.data.schemas import SeqData, SeqBatch, TokenizedSeqBatch
+from genrec_amd.data.synthetic import (
+    SyntheticSASRecDataset, SyntheticHSTUDataset, SyntheticItemDataset,
+    SyntheticSemIdSeqDataset,
+)
+from genrec_amd.data.collate import (
+    sasrec_collate_fn, sasrec_eval_collate_fn, hstu_collate_fn,
+    hstu_eval_collate_fn, tiger_pad_collate,
+)
+
+__all__ = [
+    "SeqData", "SeqBatch", "TokenizedSeqBatch",
+    "SyntheticSASRecDataset", "SyntheticHSTUDataset", "SyntheticItemDataset",
+    "SyntheticSemIdSeqDataset", "sasrec_collate_fn", "sasrec_eval_collate_fn",
+    "hstu_collate_fn", "hstu_eval_collate_fn", "tiger_pad_collate",
+]

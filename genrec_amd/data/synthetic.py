@@ -1,0 +1,192 @@
+"""Synthetic datasets shaped like Amazon-2014 5-core splits.
+
+The driver environment has no network, so benchmarks and tests run on
+synthetic user-item sequences with random-init weights (BASELINE.json). The
+generators reproduce the statistics that matter for kernel shapes and the
+training hot loop: Zipf-distributed item popularity, per-user sequence
+lengths like the 5-core data (mean ≈ 9, min 5), monotone timestamps, and —
+for TIGER — per-item semantic-ID tuples drawn from C codebooks of size V
+(collision rate matching a trained RQ-VAE's ≈ a few %).
+
+Amazon-Beauty reference stats: ~22.3k users, ~12.1k items, ~198k events.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+from torch.utils.data import Dataset
+
+from genrec_amd.config import ginlite
+from genrec_amd.data.schemas import SeqData
+
+BEAUTY_LIKE = dict(num_users=22363, num_items=12101, mean_len=8.9)
+
+
+def _zipf_sequences(num_users: int, num_items: int, mean_len: float,
+                    seed: int, min_len: int = 5, max_len: int = 50,
+                    alpha: float = 1.1) -> List[List[int]]:
+    """Per-user item sequences with Zipf item popularity. Item ids 1-based."""
+    rng = np.random.default_rng(seed)
+    ranks = np.arange(1, num_items + 1, dtype=np.float64)
+    probs = ranks ** (-alpha)
+    probs /= probs.sum()
+    perm = rng.permutation(num_items) + 1  # popularity -> shuffled item id
+    lens = np.clip(
+        rng.poisson(mean_len - min_len, size=num_users) + min_len,
+        min_len, max_len)
+    seqs = []
+    for L in lens:
+        picks = rng.choice(num_items, size=int(L), replace=True, p=probs)
+        seqs.append([int(perm[p]) for p in picks])
+    return seqs
+
+
+class SyntheticSeqDataset(Dataset):
+    """Raw item-ID sequences (SASRec/HSTU-style samples).
+
+    Train: sliding window over seq[:-2]; valid: leave-one-out on seq[:-1];
+    test: leave-one-out on the full sequence — matching the reference's
+    split logic (amazon_sasrec.py:80-112).
+    """
+
+    def __init__(self, num_users: int = 2000, num_items: int = 12101,
+                 mean_len: float = 8.9, max_seq_len: int = 50,
+                 split: str = "train", seed: int = 0,
+                 with_timestamps: bool = False) -> None:
+        self.num_items = num_items
+        self.max_seq_len = max_seq_len
+        self.with_timestamps = with_timestamps
+        seqs = _zipf_sequences(num_users, num_items, mean_len, seed)
+        base_ts = 1_400_000_000
+        self.samples: List[Dict] = []
+        for u, full in enumerate(seqs):
+            ts = [base_ts + u * 1000 + i * 86400 for i in range(len(full))]
+            if split == "train":
+                seq = full[:-2]
+                t = ts[:-2]
+                if len(seq) < 2:
+                    continue
+                for i in range(1, len(seq)):
+                    lo = max(0, i - max_seq_len)
+                    self.samples.append({
+                        "user": u,
+                        "history": seq[lo:i],
+                        "ts": t[lo:i],
+                        "target": seq[i],
+                    })
+            else:
+                seq = full[:-1] if split == "valid" else full
+                t = ts[:-1] if split == "valid" else ts
+                if len(seq) < 2:
+                    continue
+                lo = max(0, len(seq) - 1 - max_seq_len)
+                self.samples.append({
+                    "user": u,
+                    "history": seq[lo:-1],
+                    "ts": t[lo:-1],
+                    "target": seq[-1],
+                })
+
+    def __len__(self) -> int:
+        return len(self.samples)
+
+    def __getitem__(self, idx: int) -> Dict:
+        s = self.samples[idx]
+        out = {"history": s["history"], "target": s["target"]}
+        if self.with_timestamps:
+            out["timestamps"] = s["ts"]
+        return out
+
+
+@ginlite.configurable(name="SyntheticSASRecDataset")
+class SyntheticSASRecDataset(SyntheticSeqDataset):
+    pass
+
+
+@ginlite.configurable(name="SyntheticHSTUDataset")
+class SyntheticHSTUDataset(SyntheticSeqDataset):
+    def __init__(self, **kw):
+        kw.setdefault("with_timestamps", True)
+        super().__init__(**kw)
+
+
+@ginlite.configurable(name="SyntheticItemDataset")
+class SyntheticItemDataset(Dataset):
+    """Item feature vectors for RQ-VAE training (768-dim sentence-T5-like,
+    L2-normalized continuous part + {0,1} categorical tail)."""
+
+    def __init__(self, num_items: int = 12101, dim: int = 768,
+                 n_cat_features: int = 18, seed: int = 0,
+                 split: str = "train") -> None:
+        rng = np.random.default_rng(seed + (0 if split == "train" else 1))
+        n_cont = dim - n_cat_features
+        # cluster structure so kmeans/quantization have something to learn
+        n_clusters = 64
+        centers = rng.normal(size=(n_clusters, n_cont))
+        assign = rng.integers(0, n_clusters, size=num_items)
+        x = centers[assign] + 0.3 * rng.normal(size=(num_items, n_cont))
+        x /= np.linalg.norm(x, axis=1, keepdims=True) + 1e-12
+        feats = np.concatenate(
+            [x, (rng.random((num_items, n_cat_features)) < 0.2).astype(np.float32)],
+            axis=1) if n_cat_features > 0 else x
+        self.x = torch.tensor(feats, dtype=torch.float32)
+
+    def __len__(self) -> int:
+        return self.x.size(0)
+
+    def __getitem__(self, idx: int):
+        return self.x[idx]
+
+
+@ginlite.configurable(name="SyntheticSemIdSeqDataset")
+class SyntheticSemIdSeqDataset(Dataset):
+    """TIGER-style tokenized sequences: per-item sem-ID tuples (C codes of V)
+    flattened into the history, leave-one-out target.
+
+    Mirrors AmazonSeqDataset's output schema (amazon.py:392-444) on
+    synthetic data: each item maps to a fixed random (c_0..c_{C-1}) tuple.
+    """
+
+    def __init__(self, num_users: int = 2000, num_items: int = 12101,
+                 mean_len: float = 8.9, max_items_per_seq: int = 20,
+                 sem_id_dim: int = 3, codebook_size: int = 256,
+                 split: str = "train", seed: int = 0) -> None:
+        self.sem_id_dim = sem_id_dim
+        self.codebook_size = codebook_size
+        self.max_items = max_items_per_seq
+        rng = np.random.default_rng(seed + 17)
+        self.item_sem_ids = rng.integers(
+            0, codebook_size, size=(num_items + 1, sem_id_dim))
+        self.num_items = num_items
+        seqs = _zipf_sequences(num_users, num_items, mean_len, seed)
+        self.samples: List[SeqData] = []
+        for u, full in enumerate(seqs):
+            if split == "train":
+                seq = full[:-2]
+                if len(seq) < 2:
+                    continue
+                for i in range(1, len(seq)):
+                    lo = max(0, i - max_items_per_seq)
+                    self.samples.append(SeqData(u, seq[lo:i], [seq[i]]))
+            else:
+                seq = full[:-1] if split == "valid" else full
+                if len(seq) < 2:
+                    continue
+                lo = max(0, len(seq) - 1 - max_items_per_seq)
+                self.samples.append(SeqData(u, seq[lo:-1], [seq[-1]]))
+
+    def all_valid_sem_ids(self) -> torch.Tensor:
+        """[N_items, sem_id_dim] tuples for trie construction."""
+        return torch.tensor(self.item_sem_ids[1:], dtype=torch.long)
+
+    def __len__(self) -> int:
+        return len(self.samples)
+
+    def __getitem__(self, idx: int) -> SeqData:
+        s = self.samples[idx]
+        hist = [int(c) for it in s.item_ids for c in self.item_sem_ids[it]]
+        tgt = [int(c) for it in s.target_ids for c in self.item_sem_ids[it]]
+        return SeqData(user_id=s.user_id, item_ids=hist, target_ids=tgt)

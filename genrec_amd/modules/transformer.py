@@ -1,0 +1,276 @@
+"""T5-style encoder-decoder transformer (parity: reference transformer.py).
+
+Semantics preserved from the reference implementation:
+  * bidirectional log-bucket relative position bias, one per self-attention
+    layer, stored as a flat Embedding(n_heads*num_buckets, 1) gathered with
+    per-head offsets (transformer.py:13-41, 84-104)
+  * fused KV projection for self-attention (transformer.py:72)
+  * 1/sqrt(head_dim) score scale (transformer.py:63)
+  * key-padding mask at -1e9 then additive attention mask (transformer.py:144-151)
+  * pre-norm blocks with optional cross attention, ReLU FFN
+    (transformer.py:162-189, 256-324)
+
+MI355X redesign notes:
+  * the scores -> bias -> mask -> softmax -> PV chain dispatches to the fused
+    CDNA4 attention kernel via genrec_amd.ops.t5_attention
+  * bucket index tensors are input-independent per (Lq, Lk); they are cached
+    per module instead of recomputed every layer call (reference recomputes
+    per layer per forward, transformer.py:84-104)
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+from torch import Tensor, nn
+
+from genrec_amd import ops
+from genrec_amd.modules.norms import RMSNorm, T5RMSNorm
+
+
+def relative_position_bucket(relative_positions: Tensor, num_buckets: int = 32,
+                             max_distance: int = 128,
+                             bidirectional: bool = True) -> Tensor:
+    """Log-bucketing of relative positions (ref transformer.py:13-41)."""
+    ret = -relative_positions
+    if bidirectional:
+        num_buckets //= 2
+        sign = (ret < 0).long()
+        ret = ret.abs()
+    else:
+        ret = torch.clamp_min(ret, 0)
+    max_exact = num_buckets // 2
+    is_small = ret < max_exact
+    large = max_exact + (
+        (torch.log(ret.float() / max_exact + 1e-6)
+         / math.log(max_distance / max_exact))
+        * (num_buckets - max_exact)
+    ).long().clamp(max=num_buckets - max_exact - 1)
+    ret = torch.where(is_small, ret, large)
+    if bidirectional:
+        ret = ret + sign * num_buckets
+    return ret
+
+
+class T5Attention(nn.Module):
+    """Multi-head attention with optional per-layer relative bias."""
+
+    def __init__(self, d_model: int, n_heads: int, dropout: float = 0.0,
+                 is_cross_attention: bool = False, has_relative_bias: bool = True,
+                 num_relative_buckets: int = 32, max_distance: int = 128) -> None:
+        super().__init__()
+        assert d_model % n_heads == 0
+        self.d_model = d_model
+        self.n_heads = n_heads
+        self.head_dim = d_model // n_heads
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+        self.is_cross_attention = is_cross_attention
+
+        self.q = nn.Linear(d_model, d_model, bias=False)
+        if is_cross_attention:
+            self.k = nn.Linear(d_model, d_model, bias=False)
+            self.v = nn.Linear(d_model, d_model, bias=False)
+        else:
+            self.kv = nn.Linear(d_model, 2 * d_model, bias=False)
+        self.o = nn.Linear(d_model, d_model, bias=False)
+        self.dropout_p = dropout
+
+        if has_relative_bias and not is_cross_attention:
+            self.rel_bias = nn.Embedding(n_heads * num_relative_buckets, 1)
+            self.num_relative_buckets = num_relative_buckets
+            self.max_distance = max_distance
+        else:
+            self.rel_bias = None
+        self._bucket_cache: dict = {}
+
+    def _bias_indices(self, q_len: int, k_len: int, device) -> Tensor:
+        key = (q_len, k_len, str(device))
+        idx = self._bucket_cache.get(key)
+        if idx is None:
+            ctx = torch.arange(q_len, device=device)[:, None]
+            mem = torch.arange(k_len, device=device)[None, :]
+            buckets = relative_position_bucket(
+                mem - ctx, self.num_relative_buckets, self.max_distance,
+                bidirectional=True,
+            )
+            offs = (torch.arange(self.n_heads, device=device)
+                    * self.num_relative_buckets)[:, None, None]
+            idx = (buckets.unsqueeze(0) + offs).reshape(-1)
+            self._bucket_cache[key] = idx
+        return idx
+
+    def compute_bias(self, q_len: int, k_len: int, device) -> Tensor:
+        idx = self._bias_indices(q_len, k_len, device)
+        return self.rel_bias(idx).view(self.n_heads, q_len, k_len)
+
+    def _split(self, x: Tensor) -> Tensor:
+        b, l, _ = x.shape
+        return x.view(b, l, self.n_heads, self.head_dim).transpose(1, 2)
+
+    def forward(self, query: Tensor, key: Optional[Tensor] = None,
+                value: Optional[Tensor] = None,
+                attn_mask: Optional[Tensor] = None,
+                key_padding_mask: Optional[Tensor] = None,
+                kv_cache: Optional[dict] = None) -> Tensor:
+        if self.is_cross_attention:
+            if kv_cache is not None and "k" in kv_cache:
+                k, v = kv_cache["k"], kv_cache["v"]
+            else:
+                k, v = self._split(self.k(key)), self._split(self.v(value))
+                if kv_cache is not None:
+                    kv_cache["k"], kv_cache["v"] = k, v
+        else:
+            kv = self.kv(query)
+            k, v = map(self._split, kv.chunk(2, dim=-1))
+            if kv_cache is not None:
+                if "k" in kv_cache:
+                    k = torch.cat([kv_cache["k"], k], dim=2)
+                    v = torch.cat([kv_cache["v"], v], dim=2)
+                kv_cache["k"], kv_cache["v"] = k, v
+        q = self._split(self.q(query))
+
+        bias = None
+        if self.rel_bias is not None:
+            bias = self.compute_bias(q.size(2), k.size(2), q.device)
+
+        out = ops.t5_attention(
+            q, k, v, bias, key_padding_mask, attn_mask, self.scale,
+            self.dropout_p, self.training,
+        )
+        b = out.size(0)
+        out = out.transpose(1, 2).contiguous().view(b, -1, self.d_model)
+        return self.o(out)
+
+
+class FeedForward(nn.Module):
+    """T5-style FFN: wi -> relu -> dropout -> wo (ref transformer.py:162-189)."""
+
+    def __init__(self, dim: int, hidden_dim: int = 2048,
+                 dropout: float = 0.1) -> None:
+        super().__init__()
+        self.wi = nn.Linear(dim, hidden_dim, bias=False)
+        self.wo = nn.Linear(hidden_dim, dim, bias=False)
+        self.dropout = nn.Dropout(dropout)
+
+    def forward(self, x: Tensor) -> Tensor:
+        return self.wo(self.dropout(F.relu(self.wi(x))))
+
+
+class TransformerBlock(nn.Module):
+    """Pre-norm block: self-attn [+ cross-attn] + FFN (ref transformer.py:256-324)."""
+
+    def __init__(self, dim: int, num_heads: int, dropout: float = 0.1,
+                 norm_cls: type = RMSNorm, ff_hidden_dim: int = 2048,
+                 cross_attn: bool = False) -> None:
+        super().__init__()
+        self.cross_attn_enabled = cross_attn
+        self.self_attn = T5Attention(dim, num_heads, dropout)
+        self.norm1 = norm_cls(dim)
+        self.dropout1 = nn.Dropout(dropout)
+        if cross_attn:
+            self.cross_attn = T5Attention(dim, num_heads, dropout,
+                                          is_cross_attention=True,
+                                          has_relative_bias=False)
+            self.norm_cross = norm_cls(dim)
+            self.dropout_cross = nn.Dropout(dropout)
+        self.ff = FeedForward(dim, hidden_dim=ff_hidden_dim, dropout=dropout)
+        self.norm2 = norm_cls(dim)
+        self.dropout2 = nn.Dropout(dropout)
+
+    def forward(self, x: Tensor, *, context: Optional[Tensor] = None,
+                attn_mask: Optional[Tensor] = None,
+                key_padding_mask: Optional[Tensor] = None,
+                memory_key_padding_mask: Optional[Tensor] = None,
+                kv_cache: Optional[dict] = None) -> Tensor:
+        self_cache = cross_cache = None
+        if kv_cache is not None:
+            self_cache = kv_cache.setdefault("self", {})
+            cross_cache = kv_cache.setdefault("cross", {})
+        x = x + self.dropout1(self.self_attn(
+            self.norm1(x), attn_mask=attn_mask,
+            key_padding_mask=key_padding_mask, kv_cache=self_cache,
+        ))
+        if self.cross_attn_enabled and context is not None:
+            x = x + self.dropout_cross(self.cross_attn(
+                self.norm_cross(x), key=context, value=context,
+                key_padding_mask=memory_key_padding_mask, kv_cache=cross_cache,
+            ))
+        x = x + self.dropout2(self.ff(self.norm2(x)))
+        return x
+
+
+class TransformerEncoder(nn.Module):
+    def __init__(self, dim: int, depth: int, num_heads: int,
+                 dropout: float = 0.1, norm_cls: type = RMSNorm,
+                 ff_hidden_dim: int = 2048) -> None:
+        super().__init__()
+        self.layers = nn.ModuleList([
+            TransformerBlock(dim, num_heads, dropout, norm_cls=norm_cls,
+                             ff_hidden_dim=ff_hidden_dim, cross_attn=False)
+            for _ in range(depth)
+        ])
+
+    def forward(self, src: Tensor, *, attn_mask: Optional[Tensor] = None,
+                key_padding_mask: Optional[Tensor] = None) -> Tensor:
+        for layer in self.layers:
+            src = layer(src, attn_mask=attn_mask,
+                        key_padding_mask=key_padding_mask)
+        return src
+
+
+class TransformerDecoder(nn.Module):
+    def __init__(self, dim: int, depth: int, num_heads: int,
+                 dropout: float = 0.1, norm_cls: type = RMSNorm,
+                 ff_hidden_dim: int = 2048) -> None:
+        super().__init__()
+        self.layers = nn.ModuleList([
+            TransformerBlock(dim, num_heads, dropout, norm_cls=norm_cls,
+                             ff_hidden_dim=ff_hidden_dim, cross_attn=True)
+            for _ in range(depth)
+        ])
+
+    def forward(self, tgt: Tensor, *, memory: Tensor,
+                attn_mask: Optional[Tensor] = None,
+                key_padding_mask: Optional[Tensor] = None,
+                memory_key_padding_mask: Optional[Tensor] = None,
+                kv_caches: Optional[list] = None) -> Tensor:
+        for i, layer in enumerate(self.layers):
+            tgt = layer(tgt, context=memory, attn_mask=attn_mask,
+                        key_padding_mask=key_padding_mask,
+                        memory_key_padding_mask=memory_key_padding_mask,
+                        kv_cache=kv_caches[i] if kv_caches is not None else None)
+        return tgt
+
+
+class TransformerEncoderDecoder(nn.Module):
+    """Encoder-decoder wrapper (ref transformer.py:417-476)."""
+
+    def __init__(self, d_model: int, nhead: int, num_encoder_layers: int,
+                 num_decoder_layers: int, dim_feedforward: int = 2048,
+                 dropout: float = 0.1, norm_cls: type = RMSNorm) -> None:
+        super().__init__()
+        self.encoder = TransformerEncoder(
+            dim=d_model, depth=num_encoder_layers, num_heads=nhead,
+            dropout=dropout, norm_cls=norm_cls, ff_hidden_dim=dim_feedforward)
+        self.decoder = TransformerDecoder(
+            dim=d_model, depth=num_decoder_layers, num_heads=nhead,
+            dropout=dropout, norm_cls=norm_cls, ff_hidden_dim=dim_feedforward)
+
+    def forward(self, src: Tensor, tgt: Tensor, *,
+                src_key_padding_mask: Optional[Tensor] = None,
+                tgt_key_padding_mask: Optional[Tensor] = None,
+                memory_key_padding_mask: Optional[Tensor] = None,
+                src_mask: Optional[Tensor] = None,
+                tgt_mask: Optional[Tensor] = None) -> Tensor:
+        if tgt_mask is None:
+            t = tgt.size(1)
+            tgt_mask = torch.triu(
+                torch.full((t, t), float("-inf"), device=tgt.device), diagonal=1)
+        memory = self.encoder(src, attn_mask=src_mask,
+                              key_padding_mask=src_key_padding_mask)
+        return self.decoder(tgt, memory=memory, attn_mask=tgt_mask,
+                            key_padding_mask=tgt_key_padding_mask,
+                            memory_key_padding_mask=memory_key_padding_mask)

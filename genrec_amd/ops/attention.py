@@ -1,0 +1,135 @@
+"""Fused attention dispatch (K1 SASRec, K4 HSTU, K13 T5 — SURVEY.md §2.4).
+
+One generic fused-attention op covers the three attention families of the
+model zoo; the CDNA4 kernel (csrc/kernels/attention.hip) is templated on the
+score activation (softmax vs SiLU) and fuses the bias adds, the reference's
+exact masking semantics, and the PV matmul.
+
+Reference semantics fused here:
+  * SASRec (sasrec.py:201-245): scale -> key-mask(-1e9) -> causal(-1e9)
+    -> softmax -> POST-softmax query mask -> dropout -> @V
+  * T5 (transformer.py:106-159): scale -> +rel-bias -> key-pad-mask(-1e9)
+    -> +additive attn mask -> softmax -> dropout -> @V
+  * HSTU (hstu.py:232-276): QK^T (no scale) -> +pos bias -> +temporal bias
+    -> causal(-1e9) -> key-pad(-1e9) -> SiLU (no softmax!) -> @V
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import Tensor
+
+from genrec_amd.ops import eager
+
+_ACT_SOFTMAX = 0
+_ACT_SILU = 1
+
+
+def _kernel_available(name: str, *tensors: Tensor) -> bool:
+    from genrec_amd import ops
+
+    if not ops.use_hip(*tensors):
+        return False
+    return hasattr(ops.ext(), name)
+
+
+class _FusedAttnFn(torch.autograd.Function):
+    """Autograd wrapper over the HIP fused-attention fwd/bwd kernels."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, key_pad_mask, additive_mask, query_mask,
+                scale, causal, act, dropout_p, training):
+        from genrec_amd import ops
+
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) if (
+            dropout_p > 0 and training) else 0
+        out, probs, dmask = ops.ext().attn_fwd(
+            q, k, v, bias, key_pad_mask, additive_mask, query_mask,
+            scale, causal, act, dropout_p if training else 0.0, seed,
+        )
+        ctx.save_for_backward(q, k, v, probs, dmask,
+                              query_mask if query_mask is not None else torch.empty(0))
+        ctx.meta = (scale, causal, act, dropout_p if training else 0.0, seed,
+                    bias is not None and bias.requires_grad,
+                    bias.dim() if bias is not None else 0)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from genrec_amd import ops
+
+        q, k, v, probs, dmask, query_mask = ctx.saved_tensors
+        scale, causal, act, dropout_p, seed, bias_grad, bias_dim = ctx.meta
+        dq, dk, dv, dbias = ops.ext().attn_bwd(
+            dout.contiguous(), q, k, v, probs, dmask,
+            query_mask if query_mask.numel() else None,
+            scale, act, dropout_p, seed, bias_grad, bias_dim,
+        )
+        if bias_grad and bias_dim == 3:
+            dbias = dbias.to(q.dtype) if q.dtype != torch.float32 else dbias
+        return (dq, dk, dv, dbias if bias_grad else None,
+                None, None, None, None, None, None, None, None)
+
+
+def fused_attention(
+    q: Tensor,
+    k: Tensor,
+    v: Tensor,
+    *,
+    scale: float = 1.0,
+    bias: Optional[Tensor] = None,
+    key_pad_mask: Optional[Tensor] = None,
+    additive_mask: Optional[Tensor] = None,
+    causal: bool = False,
+    query_mask: Optional[Tensor] = None,
+    score_act: str = "softmax",
+    dropout_p: float = 0.0,
+    training: bool = False,
+) -> Tensor:
+    act = _ACT_SOFTMAX if score_act == "softmax" else _ACT_SILU
+    if _kernel_available("attn_fwd", q, k, v):
+        b = bias.contiguous() if bias is not None else None
+        kp = key_pad_mask.contiguous() if key_pad_mask is not None else None
+        am = additive_mask.contiguous() if additive_mask is not None else None
+        qm = query_mask.contiguous() if query_mask is not None else None
+        return _FusedAttnFn.apply(q, k, v, b, kp, am, qm,
+                                  scale, causal, act, dropout_p, training)
+    return eager.fused_attention(
+        q, k, v, scale=scale, bias=bias, key_pad_mask=key_pad_mask,
+        additive_mask=additive_mask, causal=causal, query_mask=query_mask,
+        score_act=score_act, dropout_p=dropout_p, training=training,
+    )
+
+
+def sasrec_attention(q, k, v, valid_mask, scale, dropout_p, training):
+    """K1: valid_mask [B, L] float (1=valid). Residual stays in the model."""
+    key_pad = valid_mask == 0
+    return fused_attention(
+        q, k, v, scale=scale, key_pad_mask=key_pad, causal=True,
+        query_mask=valid_mask, score_act="softmax",
+        dropout_p=dropout_p, training=training,
+    )
+
+
+def t5_attention(q, k, v, bias, key_pad_mask, additive_mask, scale,
+                 dropout_p, training):
+    """K13: bias [H,Lq,Lk] or None; additive_mask [Lq,Lk] float or None."""
+    return fused_attention(
+        q, k, v, scale=scale, bias=bias, key_pad_mask=key_pad_mask,
+        additive_mask=additive_mask, score_act="softmax",
+        dropout_p=dropout_p, training=training,
+    )
+
+
+def hstu_pointwise_attention(q, k, v, pos_bias, time_bias, key_pad_mask):
+    """K4 core: SiLU-score attention with two bias tensors, causal."""
+    bias = pos_bias.unsqueeze(0) if pos_bias.dim() == 3 else pos_bias
+    if time_bias is not None:
+        bias = bias + time_bias
+    return fused_attention(
+        q, k, v, scale=1.0, bias=bias, key_pad_mask=key_pad_mask,
+        causal=True, score_act="silu",
+    )

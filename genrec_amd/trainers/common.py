@@ -1,0 +1,144 @@
+"""Shared trainer runtime.
+
+The reference ships six copy-pasted ~200-490-line trainer scripts
+(SURVEY.md §1 notes the absence of a shared base). This module centralizes
+what they duplicate: seeding, logging, checkpoint I/O (dict format parity
+with the reference writers, e.g. tiger_trainer.py:258-269), distributed
+setup, AMP autocast and the DataLoader/sampler plumbing.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import random
+import time
+from typing import Any, Dict, Iterable, Optional
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader, Dataset, DistributedSampler
+
+from genrec_amd.parallel import DistributedContext, init_distributed
+
+logger = logging.getLogger("genrec_amd")
+
+
+def setup_logging(log_dir: Optional[str] = None, name: str = "train") -> None:
+    handlers: list = [logging.StreamHandler()]
+    if log_dir:
+        os.makedirs(log_dir, exist_ok=True)
+        stamp = time.strftime("%Y%m%d_%H%M%S")
+        handlers.append(logging.FileHandler(
+            os.path.join(log_dir, f"{name}_{stamp}.log")))
+    logging.basicConfig(
+        level=logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s: %(message)s",
+        handlers=handlers, force=True)
+
+
+def set_seed(seed: int, rank: int = 0) -> None:
+    random.seed(seed + rank)
+    np.random.seed(seed + rank)
+    torch.manual_seed(seed + rank)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(seed + rank)
+
+
+def autocast_ctx(device: torch.device, mixed_precision: Optional[str]):
+    if mixed_precision in ("bf16", "fp16") and device.type == "cuda":
+        dtype = torch.bfloat16 if mixed_precision == "bf16" else torch.float16
+        return torch.autocast(device_type="cuda", dtype=dtype)
+    import contextlib
+
+    return contextlib.nullcontext()
+
+
+def make_loader(dataset: Dataset, batch_size: int, ctx: DistributedContext,
+                shuffle: bool, collate_fn=None, num_workers: int = 4,
+                drop_last: bool = False, seed: int = 0) -> DataLoader:
+    sampler = None
+    if ctx.world_size > 1:
+        sampler = DistributedSampler(
+            dataset, num_replicas=ctx.world_size, rank=ctx.rank,
+            shuffle=shuffle, seed=seed, drop_last=drop_last)
+        shuffle = False
+    return DataLoader(
+        dataset, batch_size=batch_size, shuffle=shuffle, sampler=sampler,
+        collate_fn=collate_fn, num_workers=num_workers,
+        pin_memory=torch.cuda.is_available(), drop_last=drop_last,
+        persistent_workers=num_workers > 0)
+
+
+def save_checkpoint(path: str, model: torch.nn.Module, optimizer, scheduler,
+                    *, epoch: Optional[int] = None, step: Optional[int] = None,
+                    model_config: Optional[dict] = None,
+                    is_main: bool = True) -> None:
+    """Dict checkpoint, reference-compatible layout
+    ({epoch|iter, model, model_config, optimizer, scheduler})."""
+    if not is_main:
+        return
+    os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+    state: Dict[str, Any] = {"model": model.state_dict()}
+    if epoch is not None:
+        state["epoch"] = epoch
+    if step is not None:
+        state["iter"] = step
+    if model_config is not None:
+        state["model_config"] = model_config
+    if optimizer is not None:
+        state["optimizer"] = optimizer.state_dict()
+    if scheduler is not None:
+        state["scheduler"] = scheduler.state_dict()
+    tmp = path + ".tmp"
+    torch.save(state, tmp)
+    os.replace(tmp, path)
+    logger.info("saved checkpoint to %s", path)
+
+
+def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
+                    scheduler=None, map_location="cpu") -> Dict[str, Any]:
+    state = torch.load(path, map_location=map_location, weights_only=False)
+    model.load_state_dict(state["model"])
+    if optimizer is not None and "optimizer" in state:
+        optimizer.load_state_dict(state["optimizer"])
+    if scheduler is not None and "scheduler" in state:
+        scheduler.load_state_dict(state["scheduler"])
+    logger.info("resumed from %s (epoch=%s iter=%s)", path,
+                state.get("epoch"), state.get("iter"))
+    return state
+
+
+def to_device(batch, device: torch.device):
+    if isinstance(batch, dict):
+        return {k: to_device(v, device) for k, v in batch.items()}
+    if isinstance(batch, torch.Tensor):
+        return batch.to(device, non_blocking=True)
+    if isinstance(batch, (list, tuple)):
+        t = type(batch)
+        vals = [to_device(v, device) for v in batch]
+        return t(*vals) if hasattr(batch, "_fields") else t(vals)
+    return batch
+
+
+class WandbStub:
+    """No-op logger used when wandb is unavailable (offline env)."""
+
+    def log(self, *a, **k):
+        pass
+
+    def finish(self):
+        pass
+
+
+def init_wandb(project: str, config: dict, enabled: bool, is_main: bool):
+    if not (enabled and is_main):
+        return WandbStub()
+    try:
+        import wandb
+
+        wandb.init(project=project, config=config)
+        return wandb
+    except Exception:
+        logger.warning("wandb unavailable; logging disabled")
+        return WandbStub()

@@ -1,0 +1,177 @@
+"""HSTU trainer (parity: reference trainers/hstu_trainer.py, 214 LoC).
+
+SASRec trainer + timestamps + temporal-bias toggle (hstu_trainer.py:91,155).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+from torch.optim import Adam
+
+from genrec_amd.config import ginlite
+from genrec_amd.data.collate import hstu_collate_fn, hstu_eval_collate_fn
+from genrec_amd.data.synthetic import SyntheticHSTUDataset
+from genrec_amd.models.hstu import HSTU
+from genrec_amd.parallel import GradReducer, init_distributed, reduce_scalars
+from genrec_amd.parallel.ddp import broadcast_parameters
+from genrec_amd.trainers import common
+from genrec_amd.trainers.common import logger
+
+
+@torch.no_grad()
+def evaluate(model, loader, device, ks=(1, 5, 10)):
+    model.eval()
+    sums = {f"recall@{k}": 0.0 for k in ks}
+    sums.update({f"ndcg@{k}": 0.0 for k in ks})
+    n, max_k = 0, max(ks)
+    for batch in loader:
+        batch = common.to_device(batch, device)
+        logits, _ = model(batch["input_ids"], batch.get("timestamps"))
+        last = logits[:, -1, :].float()
+        last[:, 0] = float("-inf")
+        topk = torch.topk(last, max_k, dim=-1).indices
+        match = topk == batch["targets"].unsqueeze(1)
+        found = match.any(dim=1)
+        rank = torch.where(found, match.float().argmax(dim=1),
+                           torch.full_like(found.long(), max_k))
+        dcg = 1.0 / torch.log2(rank.float() + 2.0)
+        for k in ks:
+            hit = rank < k
+            sums[f"recall@{k}"] += hit.sum().item()
+            sums[f"ndcg@{k}"] += torch.where(
+                hit, dcg, torch.zeros_like(dcg)).sum().item()
+        n += batch["input_ids"].size(0)
+    sums["n"] = n
+    sums = reduce_scalars(sums, device)
+    n = max(sums.pop("n"), 1)
+    return {k: v / n for k, v in sums.items()}
+
+
+@ginlite.configurable(name="train")
+def train(
+    epochs: int = 10,
+    batch_size: int = 128,
+    learning_rate: float = 1e-3,
+    weight_decay: float = 0.0,
+    max_seq_len: int = 50,
+    embed_dim: int = 64,
+    num_heads: int = 2,
+    num_blocks: int = 2,
+    dropout: float = 0.2,
+    num_position_buckets: int = 32,
+    num_time_buckets: int = 64,
+    use_temporal_bias: bool = True,
+    dataset=None,
+    dataset_folder: str = "dataset/amazon",
+    split: str = "beauty",
+    do_eval: bool = True,
+    eval_every_epoch: int = 1,
+    eval_batch_size: int = 256,
+    save_dir_root: str = "out/hstu/amazon",
+    save_every_epoch: int = 50,
+    wandb_logging: bool = False,
+    wandb_project: str = "hstu",
+    wandb_log_interval: int = 100,
+    amp: bool = True,
+    mixed_precision_type: str = "bf16",
+    seed: int = 42,
+    max_steps: Optional[int] = None,
+    num_workers: int = 4,
+):
+    ctx = init_distributed()
+    common.setup_logging(save_dir_root if ctx.is_main else None, "hstu")
+    common.set_seed(seed, ctx.rank)
+    device = ctx.device
+
+    ds_cls = dataset or SyntheticHSTUDataset
+    import inspect
+
+    def mk(mode):
+        sig = inspect.signature(ds_cls.__init__)
+        kw = {"split": mode}
+        if "max_seq_len" in sig.parameters:
+            kw["max_seq_len"] = max_seq_len
+        if "root" in sig.parameters:
+            kw.update(root=dataset_folder, split=split, train_test_split=mode)
+        return ds_cls(**kw)
+
+    train_ds, valid_ds, test_ds = mk("train"), mk("valid"), mk("test")
+    model = HSTU(num_items=train_ds.num_items, max_seq_len=max_seq_len,
+                 embed_dim=embed_dim, num_heads=num_heads,
+                 num_blocks=num_blocks, dropout=dropout,
+                 num_position_buckets=num_position_buckets,
+                 num_time_buckets=num_time_buckets,
+                 use_temporal_bias=use_temporal_bias).to(device)
+    broadcast_parameters(model)
+    opt = Adam(model.parameters(), lr=learning_rate, betas=(0.9, 0.98),
+               weight_decay=weight_decay)
+    reducer = GradReducer(model)
+
+    coll = lambda b: hstu_collate_fn(b, max_seq_len)
+    ecoll = lambda b: hstu_eval_collate_fn(b, max_seq_len)
+    train_loader = common.make_loader(train_ds, batch_size, ctx, True, coll,
+                                      num_workers=num_workers, seed=seed)
+    valid_loader = common.make_loader(valid_ds, eval_batch_size, ctx, False,
+                                      ecoll, num_workers=num_workers)
+    test_loader = common.make_loader(test_ds, eval_batch_size, ctx, False,
+                                     ecoll, num_workers=num_workers)
+
+    wb = common.init_wandb(wandb_project, {"model": "hstu"},
+                           wandb_logging, ctx.is_main)
+    amp_ctx = common.autocast_ctx(device, mixed_precision_type if amp else None)
+    best_r10, step = -1.0, 0
+    best_path = os.path.join(save_dir_root, "best_model.pt")
+    for epoch in range(epochs):
+        model.train()
+        if hasattr(train_loader.sampler, "set_epoch"):
+            train_loader.sampler.set_epoch(epoch)
+        for batch in train_loader:
+            batch = common.to_device(batch, device)
+            opt.zero_grad(set_to_none=False)
+            with amp_ctx:
+                _, loss = model(batch["input_ids"], batch["timestamps"],
+                                batch["targets"])
+            loss.backward()
+            reducer.finalize()
+            opt.step()
+            step += 1
+            if ctx.is_main and step % wandb_log_interval == 0:
+                logger.info("epoch %d step %d loss %.4f", epoch, step,
+                            loss.item())
+                wb.log({"train/loss": loss.item()})
+            if max_steps is not None and step >= max_steps:
+                break
+        if do_eval and (epoch + 1) % eval_every_epoch == 0:
+            metrics = evaluate(model, valid_loader, device)
+            if ctx.is_main:
+                logger.info("epoch %d valid %s", epoch, metrics)
+                wb.log({f"eval/{k}": v for k, v in metrics.items()})
+            if metrics.get("recall@10", 0) > best_r10:
+                best_r10 = metrics["recall@10"]
+                if ctx.is_main:
+                    os.makedirs(save_dir_root, exist_ok=True)
+                    torch.save(model.state_dict(), best_path)
+        if ctx.is_main and (epoch + 1) % save_every_epoch == 0:
+            common.save_checkpoint(
+                os.path.join(save_dir_root, f"checkpoint_epoch_{epoch}.pt"),
+                model, opt, None, epoch=epoch, is_main=True)
+        if max_steps is not None and step >= max_steps:
+            break
+    if do_eval:
+        if os.path.exists(best_path):
+            model.load_state_dict(torch.load(best_path, map_location=device))
+        test_metrics = evaluate(model, test_loader, device)
+        if ctx.is_main:
+            logger.info("final test %s", test_metrics)
+        wb.finish()
+        return test_metrics
+    wb.finish()
+    return None
+
+
+if __name__ == "__main__":
+    ginlite.parse_config()
+    train()

@@ -1,0 +1,40 @@
+"""In-tree build of the genrec_amd HIP extension for gfx950 (MI355X).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built genrec_amd/_C*.so lives in the package so it travels with the
+repo snapshot to GPU boxes (no JIT cache dependency).
+"""
+
+import os
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+sources = [
+    "csrc/bindings.cpp",
+    "csrc/kernels/norms.hip",
+    "csrc/kernels/attention.hip",
+    "csrc/kernels/ce.hip",
+    "csrc/kernels/quantize.hip",
+    "csrc/kernels/metrics.hip",
+]
+
+setup(
+    name="genrec_amd_ext",
+    ext_modules=[
+        CUDAExtension(
+            name="genrec_amd._C",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -1,0 +1,24 @@
+import pytest
+import torch
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: needs an MI355X / ROCm GPU")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU in this environment")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+@pytest.fixture(autouse=True)
+def _clear_gin():
+    from genrec_amd.config import ginlite
+
+    ginlite.clear_config()
+    yield
+    ginlite.clear_config()

@@ -1,0 +1,136 @@
+"""ginlite config engine tests (UX parity with gin-config usage in the
+reference: utils.py:85-117, config/*.gin)."""
+
+import os
+import textwrap
+
+import pytest
+
+from genrec_amd.config import ginlite
+
+
+def _write(tmp_path, name, text):
+    p = tmp_path / name
+    p.write_text(textwrap.dedent(text))
+    return str(p)
+
+
+def test_macros_and_bindings(tmp_path):
+    cfg = _write(tmp_path, "a.gin", """
+        d_model = 128
+        train.lr = 1e-3
+        train.dims = [512, 256]
+        train.name = "hello"
+        train.d = %d_model
+        train.flag = True
+        train.none_val = None
+    """)
+    ginlite.parse_file(cfg)
+
+    @ginlite.configurable(name="train")
+    def train(lr=0.0, dims=None, name="", d=0, flag=False, none_val=1):
+        return lr, dims, name, d, flag, none_val
+
+    assert train() == (1e-3, [512, 256], "hello", 128, True, None)
+
+
+def test_explicit_args_beat_bindings(tmp_path):
+    cfg = _write(tmp_path, "b.gin", "f.x = 5\n")
+    ginlite.parse_file(cfg)
+
+    @ginlite.configurable(name="f")
+    def f(x=0, y=0):
+        return x, y
+
+    assert f() == (5, 0)
+    assert f(x=7) == (7, 0)
+
+
+def test_split_substitution(tmp_path):
+    cfg = _write(tmp_path, "c.gin", 'g.path = "out/{split}/ckpt"\n')
+    ginlite.parse_file(cfg, substitutions={"split": "beauty"})
+
+    @ginlite.configurable(name="g")
+    def g(path=""):
+        return path
+
+    assert g() == "out/beauty/ckpt"
+
+
+def test_class_reference(tmp_path):
+    cfg = _write(tmp_path, "d.gin", "h.dataset = @MyDs\n")
+
+    @ginlite.configurable(name="MyDs")
+    class MyDs:
+        def __init__(self, v=3):
+            self.v = v
+
+    ginlite.parse_file(cfg)
+
+    @ginlite.configurable(name="h")
+    def h(dataset=None):
+        return dataset
+
+    assert h() is MyDs
+
+
+def test_enum_constant(tmp_path):
+    cfg = _write(
+        tmp_path, "e.gin",
+        "q.mode = %genrec_amd.models.rqvae.QuantizeForwardMode.STE\n")
+    ginlite.parse_file(cfg)
+    from genrec_amd.models.rqvae import QuantizeForwardMode
+
+    @ginlite.configurable(name="q")
+    def q(mode=None):
+        return mode
+
+    assert q() is QuantizeForwardMode.STE
+
+
+def test_include(tmp_path):
+    _write(tmp_path, "base.gin", "base_val = 10\n")
+    cfg = _write(tmp_path, "main.gin", """
+        include "base.gin"
+        z.v = %base_val
+    """)
+    ginlite.parse_file(cfg)
+
+    @ginlite.configurable(name="z")
+    def z(v=0):
+        return v
+
+    assert z() == 10
+
+
+def test_cli_parse_config(tmp_path):
+    cfg = _write(tmp_path, "f.gin", 'w.a = 1\nw.p = "x/{split}"\n')
+    ginlite.parse_config([cfg, "--split", "toys", "--gin", "w.a=2"])
+
+    @ginlite.configurable(name="w")
+    def w(a=0, p=""):
+        return a, p
+
+    assert w() == (2, "x/toys")
+
+
+def test_class_binding_injection():
+    ginlite.bind("Klass.size", "9")
+
+    @ginlite.configurable(name="Klass")
+    class Klass:
+        def __init__(self, size=1):
+            self.size = size
+
+    assert Klass().size == 9
+    assert Klass(size=2).size == 2
+
+
+def test_shipped_configs_parse():
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for rel in ["config/sasrec/synthetic.gin", "config/hstu/synthetic.gin",
+                "config/tiger/synthetic/tiger.gin",
+                "config/tiger/synthetic/rqvae.gin"]:
+        ginlite.clear_config()
+        ginlite.parse_file(os.path.join(root, rel),
+                           substitutions={"split": "beauty"})

@@ -1,0 +1,141 @@
+"""Module-level numerics tests (semantics parity with the reference's
+modules; formulas cited per test)."""
+
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from genrec_amd.modules import (
+    L2Norm, RMSNorm, T5RMSNorm, SwishLayerNorm, ReconstructionLoss,
+    CategoricalReconstructionLoss, QuantizeLoss, TopKAccumulator,
+    SemIdEmbedding, UserIdEmbedding, Kmeans, kmeans_init_,
+    InverseSquareRootScheduler, relative_position_bucket,
+)
+
+torch.manual_seed(0)
+
+
+def test_rms_norm_matches_formula():
+    x = torch.randn(8, 32)
+    m = RMSNorm(32)
+    with torch.no_grad():
+        m.weight.copy_(torch.randn(32))
+    ref = (x.float() * torch.rsqrt(x.float().pow(2).mean(-1, keepdim=True)
+                                   + 1e-6)).to(x.dtype) * m.weight
+    assert torch.allclose(m(x), ref, atol=1e-6)
+
+
+def test_t5_rms_norm_matches_formula():
+    x = torch.randn(8, 16)
+    m = T5RMSNorm(16)
+    with torch.no_grad():
+        m.weight.copy_(torch.randn(16))
+    var = x.float().pow(2).mean(-1, keepdim=True)
+    ref = m.weight * (x * torch.rsqrt(var + 1e-6))
+    assert torch.allclose(m(x), ref, atol=1e-6)
+
+
+def test_l2norm_matches_f_normalize():
+    x = torch.randn(5, 7)
+    assert torch.allclose(L2Norm()(x), F.normalize(x, p=2, dim=-1),
+                          atol=1e-6)
+
+
+def test_swish_layer_norm():
+    x = torch.randn(4, 12)
+    m = SwishLayerNorm(12)
+    ref = F.silu(m.ln(x))
+    assert torch.allclose(m(x), ref, atol=1e-6)
+
+
+def test_losses():
+    x, y = torch.randn(6, 10), torch.randn(6, 10)
+    assert torch.allclose(ReconstructionLoss()(y, x), ((y - x) ** 2).sum(-1))
+    # categorical tail (loss.py:26-54)
+    xb = torch.cat([x, torch.randint(0, 2, (6, 3)).float()], dim=1)
+    yb = torch.cat([y, torch.randn(6, 3)], dim=1)
+    out = CategoricalReconstructionLoss(3)(yb, xb)
+    ref = ((y - x) ** 2).sum(-1) + F.binary_cross_entropy_with_logits(
+        yb[:, -3:], xb[:, -3:], reduction="none").sum(-1)
+    assert torch.allclose(out, ref, atol=1e-5)
+    # commitment loss (loss.py:57-77)
+    q, v = torch.randn(6, 4, requires_grad=True), torch.randn(6, 4,
+                                                              requires_grad=True)
+    ql = QuantizeLoss(0.25)(q, v)
+    ref = ((q.detach() - v) ** 2).sum(-1) + 0.25 * ((q - v.detach()) ** 2).sum(-1)
+    assert torch.allclose(ql, ref)
+
+
+def test_topk_accumulator_matches_reference_math():
+    # reproduces metrics.py:26-66 semantics
+    actual = torch.tensor([[1, 2, 3], [4, 5, 6], [7, 8, 9]])
+    top_k = torch.stack([
+        torch.tensor([[9, 9, 9], [1, 2, 3], [0, 0, 0]]),  # match at rank 1
+        torch.tensor([[4, 5, 6], [0, 0, 0], [1, 1, 1]]),  # match at rank 0
+        torch.tensor([[0, 0, 0], [1, 1, 1], [2, 2, 2]]),  # no match
+    ])
+    acc = TopKAccumulator(ks=[1, 2, 3])
+    acc.accumulate(actual, top_k)
+    m = acc.reduce()
+    assert m["Recall@1"] == pytest.approx(1 / 3)
+    assert m["Recall@2"] == pytest.approx(2 / 3)
+    assert m["NDCG@2"] == pytest.approx((1 / math.log2(3) + 1.0) / 3)
+
+
+def test_sem_id_embedding_flat_index():
+    emb = SemIdEmbedding(num_embeddings=4, sem_ids_dim=3, embeddings_dim=8)
+    ids = torch.tensor([[1, 2, 3]])
+    types = torch.tensor([[0, 1, 2]])
+    out = emb(ids, types)
+    expected = emb.emb(torch.tensor([[0 * 4 + 1, 1 * 4 + 2, 2 * 4 + 3]]))
+    assert torch.equal(out, expected)
+    assert emb.padding_idx == 12
+
+
+def test_user_id_embedding_hashing():
+    emb = UserIdEmbedding(num_embeddings=10, embeddings_dim=4)
+    out = emb(torch.tensor([[15]]))
+    assert torch.equal(out, emb.emb(torch.tensor([[5]])))
+
+
+def test_kmeans_converges_on_separated_clusters():
+    torch.manual_seed(1)
+    import numpy as np
+
+    np.random.seed(1)
+    a = torch.randn(50, 4) * 0.05 + torch.tensor([5.0, 0, 0, 0])
+    b = torch.randn(50, 4) * 0.05 + torch.tensor([-5.0, 0, 0, 0])
+    x = torch.cat([a, b])
+    out = Kmeans(k=2).run(x)
+    cents = out.centroids[:, 0].sort().values
+    assert abs(cents[0].item() + 5) < 0.5 and abs(cents[1].item() - 5) < 0.5
+    code = torch.empty(2, 4)
+    kmeans_init_(code, x)
+    assert code.shape == (2, 4)
+
+
+def test_inverse_sqrt_scheduler():
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=1.0)
+    sched = InverseSquareRootScheduler(opt, warmup_steps=10)
+    for _ in range(10):
+        opt.step()
+        sched.step()
+    assert opt.param_groups[0]["lr"] == pytest.approx(1.0)
+    for _ in range(30):
+        opt.step()
+        sched.step()
+    assert opt.param_groups[0]["lr"] == pytest.approx((10 / 40) ** 0.5)
+
+
+def test_relative_position_bucket_properties():
+    # bidirectional: bucket(d) in [0, 32); symmetric ranges split by sign
+    # (transformer.py:13-41)
+    rel = torch.arange(-200, 200).view(1, -1)
+    b = relative_position_bucket(rel, num_buckets=32, max_distance=128,
+                                 bidirectional=True)
+    assert b.min() >= 0 and b.max() < 32
+    assert b[0, 0] != b[0, -1]
+    # zero distance -> bucket 0
+    assert relative_position_bucket(torch.tensor([[0]])).item() == 0
