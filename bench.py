@@ -74,8 +74,6 @@ def main() -> None:
                sem_id_dim=3)
     model = Tiger(**cfg).to(device)
     broadcast_parameters(model)
-    opt = torch.optim.AdamW(model.parameters(), lr=1e-4, weight_decay=0.035)
-    reducer = GradReducer(model)
 
     n_items_hist = 20
     batches = [
@@ -85,18 +83,105 @@ def main() -> None:
         for i in range(8)
     ]
 
-    amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16) \
-        if use_gpu else _null_ctx()
+    # cache_enabled=False: required under hipGraph capture — autocast's
+    # weight-cast cache frees graph-pool blocks the graph still references.
+    amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                         cache_enabled=False) if use_gpu else _null_ctx()
 
-    def step(i: int) -> None:
-        b = batches[i % len(batches)]
-        opt.zero_grad(set_to_none=False)
+    # hipGraph capture of the whole train step (fwd+bwd+allreduce+clip+AdamW)
+    # removes per-launch host overhead — the step is ~1000 tiny dispatches.
+    use_graph = use_gpu and os.environ.get("GENREC_BENCH_GRAPH", "1") == "1"
+
+    if use_graph:
+        opt = torch.optim.AdamW(model.parameters(), lr=1e-4,
+                                weight_decay=0.035, capturable=True,
+                                foreach=True)
+        model.train()
+        params = [p for p in model.parameters() if p.requires_grad]
+
+        static = {k: v.clone() for k, v in batches[0].items()}
+
+        # grads live as views into ONE flat buffer: a single RCCL
+        # all-reduce moves the whole gradient (xGMI likes few large
+        # messages), zeroing is one fill, and the global-norm clip is a
+        # norm+scale on the flat buffer (same math as clip_grad_norm_).
         with amp:
-            out = model(**b)
-        out.loss.backward()
-        reducer.finalize()
-        torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
-        opt.step()
+            model(**static).loss.backward()
+        flat_grads = torch.zeros(sum(p.numel() for p in params),
+                                 device=device)
+        off = 0
+        for p in params:
+            p.grad = flat_grads[off:off + p.numel()].view_as(p)
+            off += p.numel()
+
+        def inner_step():
+            flat_grads.zero_()
+            with amp:
+                out = model(**static)
+            out.loss.backward()
+            if world > 1:
+                import torch.distributed as dist
+
+                dist.all_reduce(flat_grads)
+                flat_grads.mul_(1.0 / world)
+            norm = flat_grads.norm()
+            flat_grads.mul_(torch.clamp(1.0 / (norm + 1e-6), max=1.0))
+            opt.step()
+            return out.loss
+
+        debug = os.environ.get("GENREC_BENCH_DEBUG", "0") == "1"
+
+        def dbg(msg):
+            if debug:
+                torch.cuda.synchronize()
+                print(f"# bench-debug: {msg}", flush=True)
+
+        try:
+            # warmup on a side stream (required before capture)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    inner_step()
+            torch.cuda.current_stream().wait_stream(s)
+            dbg("warmup done")
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_loss = inner_step()
+            dbg("capture done")
+            graph.replay()
+            dbg("first replay done")
+
+            nocopy = os.environ.get("GENREC_BENCH_NOCOPY", "0") == "1"
+
+            def step(i: int) -> None:
+                if not nocopy:
+                    b = batches[i % len(batches)]
+                    for key in static:
+                        static[key].copy_(b[key], non_blocking=True)
+                graph.replay()
+                if debug:
+                    torch.cuda.synchronize()
+                    print(f"# bench-debug: replay {i} ok", flush=True)
+        except Exception as e:  # pragma: no cover - capture unsupported
+            print(f"# graph capture failed ({e}); falling back to eager",
+                  flush=True)
+            use_graph = False
+
+    if not use_graph:
+        opt = torch.optim.AdamW(model.parameters(), lr=1e-4,
+                                weight_decay=0.035)
+        reducer = GradReducer(model)
+
+        def step(i: int) -> None:
+            b = batches[i % len(batches)]
+            opt.zero_grad(set_to_none=False)
+            with amp:
+                out = model(**b)
+            out.loss.backward()
+            reducer.finalize()
+            torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+            opt.step()
 
     model.train()
     for i in range(args.warmup):
@@ -145,6 +230,7 @@ def main() -> None:
                 "global_batch": global_batch,
                 "seq_len": n_items_hist * cfg["sem_id_dim"] + 1,
                 "parallelism": f"dp{world}",
+                "hip_graph": bool(use_graph),
             },
         }))
 

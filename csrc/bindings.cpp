@@ -17,7 +17,8 @@ std::vector<torch::Tensor> attn_fwd(
     c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
     c10::optional<torch::Tensor> add_mask,
     c10::optional<torch::Tensor> query_mask,
-    double scale, bool causal, int64_t act, double dropout_p, int64_t seed);
+    double scale, bool causal, int64_t act, double dropout_p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev);
 std::vector<torch::Tensor> attn_bwd(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor p_saved, torch::Tensor drop_mask,
@@ -34,6 +35,9 @@ torch::Tensor softmax_ce_bwd(torch::Tensor dloss, torch::Tensor logits,
 
 std::vector<torch::Tensor> sqdist_argmin(torch::Tensor x,
                                          torch::Tensor codebook);
+torch::Tensor embedding_fwd(torch::Tensor weight, torch::Tensor indices);
+torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor indices,
+                            int64_t num_weights, int64_t padding_idx);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
 
 }  // namespace genrec
@@ -48,5 +52,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_fwd", &genrec::softmax_ce_fwd, "fused CE forward");
   m.def("softmax_ce_bwd", &genrec::softmax_ce_bwd, "fused CE backward");
   m.def("sqdist_argmin", &genrec::sqdist_argmin, "L2 dist + argmin");
+  m.def("embedding_fwd", &genrec::embedding_fwd, "embedding gather");
+  m.def("embedding_bwd", &genrec::embedding_bwd, "embedding scatter-add bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
 }

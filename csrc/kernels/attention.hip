@@ -47,9 +47,11 @@ __global__ void attn_fwd_kernel(
     T* __restrict__ out,            // [B,H,Lq,D]
     float* __restrict__ p_saved,    // [B,H,Lq,Lk] (softmax: P, silu: S)
     unsigned char* __restrict__ drop_mask,  // null | [B,H,Lq,Lk]
+    const unsigned int* __restrict__ seed_dev,  // null | device seed counter
     int B, int H, int Lq, int Lk, int D,
     float scale, int bias_dim, bool causal, int act,
     float dropout_p, unsigned int seed, int q_tile) {
+  if (seed_dev) seed += *seed_dev;  // hipGraph-replay-varying dropout
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh % H;
   const int q0 = blockIdx.y * q_tile;
@@ -147,8 +149,7 @@ __global__ void attn_bwd_kernel(
     const float* __restrict__ query_mask,
     const unsigned char* __restrict__ drop_mask,
     T* __restrict__ dq, T* __restrict__ dk, T* __restrict__ dv,
-    float* __restrict__ dbias3,     // null | [H,Lq,Lk] atomically summed
-    float* __restrict__ dbias4,     // null | [B,H,Lq,Lk]
+    float* __restrict__ ds_out,     // null | [B,H,Lq,Lk] (dS for bias grad)
     int B, int H, int Lq, int Lk, int D,
     float scale, int act, float dropout_p, unsigned int seed) {
   const int bh = blockIdx.x;
@@ -210,10 +211,7 @@ __global__ void attn_bwd_kernel(
       ds = col_ok ? dp * sg * (1.f + s * (1.f - sg)) : 0.f;
     }
     if (lane < MAXL) dss[i * (MAXL + 1) + lane] = col_ok ? ds : 0.f;
-    if (col_ok) {
-      if (dbias4) dbias4[IDX4(b, h, i, lane, H, Lq, Lk)] = ds;
-      else if (dbias3) atomicAdd(&dbias3[((int64_t)h * Lq + i) * Lk + lane], ds);
-    }
+    if (ds_out && col_ok) ds_out[IDX4(b, h, i, lane, H, Lq, Lk)] = ds;
   }
   __syncthreads();
 
@@ -260,7 +258,8 @@ std::vector<torch::Tensor> attn_fwd(
     c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
     c10::optional<torch::Tensor> add_mask,
     c10::optional<torch::Tensor> query_mask,
-    double scale, bool causal, int64_t act, double dropout_p, int64_t seed) {
+    double scale, bool causal, int64_t act, double dropout_p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 4);
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
@@ -311,6 +310,9 @@ std::vector<torch::Tensor> attn_fwd(
       query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,               \
       reinterpret_cast<T*>(out.data_ptr()), p_saved.data_ptr<float>(),         \
       dropout_p > 0 ? dmask.data_ptr<unsigned char>() : nullptr,               \
+      seed_dev.has_value()                                                     \
+          ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())        \
+          : nullptr,                                                           \
       B, H, Lq, Lk, D, (float)scale, bias_dim, causal, (int)act,               \
       (float)dropout_p, (unsigned int)seed, q_tile)
 
@@ -340,20 +342,15 @@ std::vector<torch::Tensor> attn_bwd(
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  torch::Tensor dbias;
-  float* db3 = nullptr;
-  float* db4 = nullptr;
+  // bias gradient: kernel writes per-(b,h) dS to scratch (no atomics —
+  // a [H,Lq,Lk] atomic target serializes all B blocks per location); the
+  // batch reduction for 3-dim biases is one ATen sum afterwards.
+  torch::Tensor ds_scratch;
+  float* ds_ptr = nullptr;
   if (bias_grad) {
-    if (bias_dim == 3) {
-      dbias = torch::zeros({H, Lq, Lk}, q.options().dtype(torch::kFloat32));
-      db3 = dbias.data_ptr<float>();
-    } else {
-      dbias = torch::empty({B, H, Lq, Lk},
-                           q.options().dtype(torch::kFloat32));
-      db4 = dbias.data_ptr<float>();
-    }
-  } else {
-    dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
+    ds_scratch = torch::empty({B, H, Lq, Lk},
+                              q.options().dtype(torch::kFloat32));
+    ds_ptr = ds_scratch.data_ptr<float>();
   }
   torch::Tensor qm_f;
   if (query_mask.has_value()) qm_f = query_mask->to(torch::kFloat32).contiguous();
@@ -375,7 +372,7 @@ std::vector<torch::Tensor> attn_bwd(
       dropout_p > 0 ? drop_mask.data_ptr<unsigned char>() : nullptr,           \
       reinterpret_cast<T*>(dq.data_ptr()),                                     \
       reinterpret_cast<T*>(dk.data_ptr()),                                     \
-      reinterpret_cast<T*>(dv.data_ptr()), db3, db4,                           \
+      reinterpret_cast<T*>(dv.data_ptr()), ds_ptr,                             \
       B, H, Lq, Lk, D, (float)scale, (int)act, (float)dropout_p,               \
       (unsigned int)seed)
 
@@ -384,6 +381,12 @@ std::vector<torch::Tensor> attn_bwd(
   else TORCH_CHECK(false, "attn_bwd: unsupported dtype");
 #undef LAUNCH_ATTN_BWD
 
+  torch::Tensor dbias;
+  if (bias_grad) {
+    dbias = (bias_dim == 3) ? ds_scratch.sum(0) : ds_scratch;
+  } else {
+    dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
+  }
   return {dq, dk, dv, dbias};
 }
 

@@ -61,6 +61,8 @@ __global__ void rms_norm_fwd_kernel(const T* __restrict__ x,
 // backward: dx_i = w_i*dy_i*r - r^3/d * x_i * sum_j(w_j*dy_j*x_j)
 // (intermediate rounding of the forward's casts is ignored in backward, as
 // eager autograd does for the same graph up to bf16 rounding)
+constexpr int RMS_MAX_COLS_PER_LANE = 16;  // supports d <= 1024
+
 template <typename T, typename OutT>
 __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
                                     const T* __restrict__ x,
@@ -69,16 +71,16 @@ __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
                                     T* __restrict__ dx,
                                     float* __restrict__ dw,
                                     int64_t n_rows, int d) {
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* dw_local = reinterpret_cast<float*>(smem_raw);  // [d]
-  for (int j = threadIdx.x; j < d; j += blockDim.x) dw_local[j] = 0.f;
-  __syncthreads();
-
   const int wave_in_block = threadIdx.x / WAVE;
   const int waves_per_block = blockDim.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = (int64_t)blockIdx.x * waves_per_block + wave_in_block;
   const int64_t n_waves = (int64_t)gridDim.x * waves_per_block;
+
+  // per-lane dw partials in registers (lane owns columns lane, lane+64, ...)
+  float dw_acc[RMS_MAX_COLS_PER_LANE];
+#pragma unroll
+  for (int c = 0; c < RMS_MAX_COLS_PER_LANE; ++c) dw_acc[c] = 0.f;
 
   for (int64_t row = wave_id; row < n_rows; row += n_waves) {
     const T* xr = x + row * d;
@@ -91,16 +93,17 @@ __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
     dot = wave_sum(dot);
     float c = r * r * r / d * dot;
     T* dxr = dx + row * d;
-    for (int j = lane; j < d; j += WAVE) {
+    int ci = 0;
+    for (int j = lane; j < d; j += WAVE, ++ci) {
       float xf = to_f32(xr[j]);
       float dyf = to_f32(dyr[j]);
       dxr[j] = from_f32<T>(w_f32[j] * dyf * r - c * xf);
-      atomicAdd(&dw_local[j], dyf * xf * r);
+      dw_acc[ci] += dyf * xf * r;
     }
   }
-  __syncthreads();
-  for (int j = threadIdx.x; j < d; j += blockDim.x) {
-    if (dw_local[j] != 0.f) atomicAdd(&dw[j], dw_local[j]);
+  int ci = 0;
+  for (int j = lane; j < d; j += WAVE, ++ci) {
+    if (dw_acc[ci] != 0.f) atomicAdd(&dw[j], dw_acc[ci]);
   }
 }
 
@@ -222,11 +225,12 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   const int d = x.size(-1);
   const int64_t n_rows = x.numel() / d;
   auto w_f32 = w.to(torch::kFloat32).contiguous();
+  TORCH_CHECK(d <= WAVE * 16, "rms_norm_bwd: d too large");
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros({d}, x.options().dtype(torch::kFloat32));
   dim3 block(256);
   dim3 grid(grid_for_rows(n_rows, 4));
-  size_t smem = d * sizeof(float);
+  size_t smem = 0;
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_RMSB(T, OutT)                                                   \

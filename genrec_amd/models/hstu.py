@@ -59,7 +59,7 @@ class RelativePositionBias(nn.Module):
             pos = torch.arange(seq_len, device=device)
             buckets = self._bucket(pos.unsqueeze(0) - pos.unsqueeze(1))
             self._bucket_cache[key] = buckets
-        bias = self.relative_attention_bias(buckets)  # [L, L, H]
+        bias = ops.embedding(self.relative_attention_bias.weight, buckets)
         return bias.permute(2, 0, 1)  # [H, L, L]
 
 
@@ -76,7 +76,7 @@ class TemporalBias(nn.Module):
         abs_diff = torch.clamp(diff.abs(), min=1).float()
         buckets = (torch.log(abs_diff) / 0.693).long().clamp(
             min=0, max=self.num_buckets - 1)
-        bias = self.temporal_attention_bias(buckets)  # [B, L, L, H]
+        bias = ops.embedding(self.temporal_attention_bias.weight, buckets)
         return bias.permute(0, 3, 1, 2)  # [B, H, L, L]
 
 
@@ -173,7 +173,8 @@ class HSTU(nn.Module):
                 targets: Optional[Tensor] = None
                 ) -> Tuple[Optional[Tensor], Optional[Tensor]]:
         padding_mask = input_ids == 0
-        x = self.emb_dropout(self.item_embedding(input_ids))
+        x = self.emb_dropout(ops.embedding(self.item_embedding.weight,
+                                           input_ids, padding_idx=0))
         for layer in self.layers:
             x = layer(x, padding_mask, timestamps)
         x = self.final_norm(x)

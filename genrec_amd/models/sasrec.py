@@ -125,9 +125,10 @@ class SASRec(nn.Module):
                 ) -> Tuple[Tensor, Optional[Tensor]]:
         b, l = input_ids.shape
         mask = (input_ids != 0).unsqueeze(-1).float()
-        x = self.item_embedding(input_ids) * (self.embed_dim ** 0.5)
+        x = ops.embedding(self.item_embedding.weight, input_ids,
+                          padding_idx=0) * (self.embed_dim ** 0.5)
         pos = torch.arange(l, device=input_ids.device).unsqueeze(0).expand(b, l)
-        x = x + self.position_embedding(pos)
+        x = x + ops.embedding(self.position_embedding.weight, pos)
         x = self.emb_dropout(x) * mask
         for block in self.blocks:
             x = block(x, mask) * mask

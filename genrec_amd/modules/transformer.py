@@ -104,7 +104,8 @@ class T5Attention(nn.Module):
 
     def compute_bias(self, q_len: int, k_len: int, device) -> Tensor:
         idx = self._bias_indices(q_len, k_len, device)
-        return self.rel_bias(idx).view(self.n_heads, q_len, k_len)
+        return ops.embedding(self.rel_bias.weight, idx).view(
+            self.n_heads, q_len, k_len)
 
     def _split(self, x: Tensor) -> Tensor:
         b, l, _ = x.shape

@@ -84,6 +84,7 @@ from genrec_amd.ops.attention import (  # noqa: E402
 from genrec_amd.ops.quantize import residual_quantize_step  # noqa: E402
 from genrec_amd.ops.losses import tied_softmax_ce, summed_ce  # noqa: E402
 from genrec_amd.ops.metrics import topk_hit_ranks  # noqa: E402
+from genrec_amd.ops.embedding import embedding  # noqa: E402
 
 __all__ = [
     "ext",
@@ -100,4 +101,5 @@ __all__ = [
     "tied_softmax_ce",
     "summed_ce",
     "topk_hit_ranks",
+    "embedding",
 ]

@@ -35,6 +35,19 @@ def _kernel_available(name: str, *tensors: Tensor) -> bool:
     return hasattr(ops.ext(), name)
 
 
+_seed_counters = {}
+
+
+def _seed_counter(device) -> torch.Tensor:
+    """Per-device dropout seed counter; incremented per call so dropout
+    masks vary across hipGraph replays (the increment is captured too)."""
+    t = _seed_counters.get(device)
+    if t is None:
+        t = torch.zeros(1, dtype=torch.int32, device=device)
+        _seed_counters[device] = t
+    return t
+
+
 class _FusedAttnFn(torch.autograd.Function):
     """Autograd wrapper over the HIP fused-attention fwd/bwd kernels."""
 
@@ -44,11 +57,17 @@ class _FusedAttnFn(torch.autograd.Function):
         from genrec_amd import ops
 
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) if (
-            dropout_p > 0 and training) else 0
+        seed_dev = None
+        seed = 0
+        if dropout_p > 0 and training:
+            seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) \
+                if not torch.cuda.is_current_stream_capturing() else 12345
+            seed_dev = _seed_counter(q.device)
+            seed_dev.add_(1)
         out, probs, dmask = ops.ext().attn_fwd(
             q, k, v, bias, key_pad_mask, additive_mask, query_mask,
             scale, causal, act, dropout_p if training else 0.0, seed,
+            seed_dev,
         )
         ctx.save_for_backward(q, k, v, probs, dmask,
                               query_mask if query_mask is not None else torch.empty(0))
@@ -89,8 +108,13 @@ def fused_attention(
     dropout_p: float = 0.0,
     training: bool = False,
 ) -> Tensor:
+    import os
+
     act = _ACT_SOFTMAX if score_act == "softmax" else _ACT_SILU
-    if _kernel_available("attn_fwd", q, k, v):
+    fits = (k.size(2) <= 64 and q.size(2) <= 64 and q.size(3) <= 64
+            and (additive_mask is None or additive_mask.dim() == 2)
+            and os.environ.get("GENREC_DISABLE_ATTN", "0") != "1")
+    if fits and _kernel_available("attn_fwd", q, k, v):
         b = bias.contiguous() if bias is not None else None
         kp = key_pad_mask.contiguous() if key_pad_mask is not None else None
         am = additive_mask.contiguous() if additive_mask is not None else None

@@ -45,9 +45,12 @@ class _SoftmaxCEFn(torch.autograd.Function):
 
 
 def softmax_ce(logits: Tensor, targets: Tensor, ignore_index: int = -100) -> Tensor:
+    import os
+
     from genrec_amd import ops
 
-    if ops.use_hip(logits) and hasattr(ops.ext(), "softmax_ce_fwd"):
+    if os.environ.get("GENREC_DISABLE_CE", "0") != "1" \
+            and ops.use_hip(logits) and hasattr(ops.ext(), "softmax_ce_fwd"):
         return _SoftmaxCEFn.apply(logits, targets, ignore_index)
     return torch.nn.functional.cross_entropy(
         logits, targets, ignore_index=ignore_index
@@ -65,8 +68,11 @@ def summed_ce(logits: Tensor, targets: Tensor) -> Tensor:
     (tiger.py:232-240)."""
     from genrec_amd import ops
 
+    import os
+
     B, T, V = logits.shape
-    if ops.use_hip(logits) and hasattr(ops.ext(), "softmax_ce_fwd"):
+    if os.environ.get("GENREC_DISABLE_CE", "0") != "1" \
+            and ops.use_hip(logits) and hasattr(ops.ext(), "softmax_ce_fwd"):
         # fused per-token CE (no ignore) * T gives sum-then-mean semantics:
         # mean over B*T tokens * T == sum over T, mean over B.
         flat_loss = _SoftmaxCEFn.apply(

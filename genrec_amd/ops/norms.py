@@ -13,8 +13,12 @@ from genrec_amd.ops import eager
 
 
 def _hip_ok(x: Tensor) -> bool:
+    import os
+
     from genrec_amd import ops
 
+    if os.environ.get("GENREC_DISABLE_NORM", "0") == "1":
+        return False
     return ops.use_hip(x)
 
 

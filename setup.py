@@ -22,6 +22,7 @@ sources = [
     "csrc/kernels/ce.hip",
     "csrc/kernels/quantize.hip",
     "csrc/kernels/metrics.hip",
+    "csrc/kernels/embedding.hip",
 ]
 
 setup(
