@@ -47,7 +47,20 @@ def build_batch(batch_size: int, n_items: int, sem_id_dim: int,
     }
 
 
+def _enable_tuned_gemms() -> None:
+    """Load the pre-tuned hipBLASLt solution table (benchmarks/tunableop0.csv,
+    produced once with PYTORCH_TUNABLEOP_TUNING=1 on an MI355X)."""
+    repo = os.path.dirname(os.path.abspath(__file__))
+    csv = os.path.join(repo, "benchmarks", "tunableop.csv")
+    if os.path.exists(os.path.join(repo, "benchmarks", "tunableop0.csv")) \
+            and os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") != "1":
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", csv)
+
+
 def main() -> None:
+    _enable_tuned_gemms()
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
@@ -93,40 +106,53 @@ def main() -> None:
     use_graph = use_gpu and os.environ.get("GENREC_BENCH_GRAPH", "1") == "1"
 
     if use_graph:
-        opt = torch.optim.AdamW(model.parameters(), lr=1e-4,
-                                weight_decay=0.035, capturable=True,
-                                foreach=True)
+        # Pure-bf16 compute with fp32 master weights (standard bf16 mixed
+        # precision a la Megatron): the model runs natively in bf16 — no
+        # autocast, so no per-layer weight-cast kernels (~2900/step) — the
+        # optimizer steps fp32 masters, and one fused foreach-copy refreshes
+        # the bf16 params after each step. All capturable.
+        model = model.to(torch.bfloat16)
         model.train()
         params = [p for p in model.parameters() if p.requires_grad]
+        masters = [p.detach().float().clone() for p in params]
+        opt = torch.optim.AdamW(masters, lr=1e-4, weight_decay=0.035,
+                                capturable=True, foreach=True)
 
         static = {k: v.clone() for k, v in batches[0].items()}
 
-        # grads live as views into ONE flat buffer: a single RCCL
+        # bf16 grads live as views into ONE flat buffer: a single RCCL
         # all-reduce moves the whole gradient (xGMI likes few large
-        # messages), zeroing is one fill, and the global-norm clip is a
-        # norm+scale on the flat buffer (same math as clip_grad_norm_).
-        with amp:
-            model(**static).loss.backward()
+        # messages; bf16 halves the bytes), zeroing is one fill, the
+        # global-norm clip is a norm+scale on the flat fp32 copy.
+        model(**static).loss.backward()
         flat_grads = torch.zeros(sum(p.numel() for p in params),
-                                 device=device)
+                                 device=device, dtype=torch.bfloat16)
         off = 0
         for p in params:
             p.grad = flat_grads[off:off + p.numel()].view_as(p)
             off += p.numel()
+        flat_master_grad = torch.zeros(flat_grads.numel(), device=device)
+        moff = 0
+        for m in masters:
+            m.grad = flat_master_grad[moff:moff + m.numel()].view_as(m)
+            moff += m.numel()
 
         def inner_step():
             flat_grads.zero_()
-            with amp:
-                out = model(**static)
+            out = model(**static)
             out.loss.backward()
             if world > 1:
                 import torch.distributed as dist
 
                 dist.all_reduce(flat_grads)
                 flat_grads.mul_(1.0 / world)
-            norm = flat_grads.norm()
-            flat_grads.mul_(torch.clamp(1.0 / (norm + 1e-6), max=1.0))
+            flat_master_grad.copy_(flat_grads)
+            norm = flat_master_grad.norm()
+            flat_master_grad.mul_(
+                torch.clamp(1.0 / (norm + 1e-6), max=1.0))
             opt.step()
+            with torch.no_grad():
+                torch._foreach_copy_(params, masters)
             return out.loss
 
         debug = os.environ.get("GENREC_BENCH_DEBUG", "0") == "1"

@@ -29,6 +29,7 @@
 namespace genrec {
 
 constexpr int MAXL = 64;   // max Lk per block (wave width)
+constexpr int NWAVES = 8;  // waves per workgroup (512 threads)
 constexpr int MAXD = 64;   // max head dim
 constexpr float NEG_BIG_F = -1e9f;
 
@@ -61,7 +62,7 @@ __global__ void attn_fwd_kernel(
   float* kt = reinterpret_cast<float*>(smem_raw);          // [D][MAXL+1]
   float* vv = kt + MAXD * (MAXL + 1);                      // [MAXL][MAXD]
   float* qq = vv + MAXL * MAXD;                            // [q_tile][MAXD]
-  float* pp = qq + q_tile * MAXD;                          // [4][MAXL+1] per-wave P rows
+  float* pp = qq + q_tile * MAXD;                          // [NWAVES][MAXL+1] per-wave P rows
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -82,7 +83,7 @@ __global__ void attn_fwd_kernel(
 
   const float inv_keep = dropout_p > 0.f ? 1.0f / (1.0f - dropout_p) : 1.0f;
 
-  for (int i = q0 + wid; i < q1; i += 4) {
+  for (int i = q0 + wid; i < q1; i += NWAVES) {
     // ----- scores: lane owns column `lane`
     float s = 0.f;
     const bool col_ok = lane < Lk;
@@ -162,6 +163,7 @@ __global__ void attn_bwd_kernel(
   float* dos = qs + MAXL * MAXD;                           // [MAXL][MAXD]
   float* ps = dos + MAXL * MAXD;                           // [MAXL][MAXL+1]
   float* dss = ps + MAXL * (MAXL + 1);                     // [MAXL][MAXL+1]
+  float* am = dss + MAXL * (MAXL + 1);                     // [MAXL][MAXL+1]
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -178,14 +180,26 @@ __global__ void attn_bwd_kernel(
     qs[i * MAXD + t] = to_f32(q[IDX4(b, h, i, t, H, Lq, D)]);
     dos[i * MAXD + t] = to_f32(dout[IDX4(b, h, i, t, H, Lq, D)]);
   }
+  // stage P and the combined post-softmax multiplier (query mask x
+  // dropout keep/scale) in one coalesced pass — the inner loops then only
+  // touch LDS (a per-element global drop_mask read inside phase 3 was the
+  // dominant cost of this kernel: ~500-cycle latency per (i,j) pair).
+  const bool has_am = (query_mask != nullptr) || (dropout_p > 0.f);
   for (int idx = tid; idx < Lq * Lk; idx += blockDim.x) {
     int i = idx / Lk, j = idx % Lk;
     ps[i * (MAXL + 1) + j] = p_saved[IDX4(b, h, i, j, H, Lq, Lk)];
+    if (has_am) {
+      float m = query_mask ? query_mask[(int64_t)b * Lq + i] : 1.f;
+      if (dropout_p > 0.f) {
+        m *= drop_mask[IDX4(b, h, i, j, H, Lq, Lk)] ? inv_keep : 0.f;
+      }
+      am[i * (MAXL + 1) + j] = m;
+    }
   }
   __syncthreads();
 
   // ---- phase 1: dS rows (wave per q-row, lane owns column)
-  for (int i = wid; i < Lq; i += 4) {
+  for (int i = wid; i < Lq; i += NWAVES) {
     const bool col_ok = lane < Lk;
     float dp = 0.f;
     if (col_ok) {
@@ -197,11 +211,7 @@ __global__ void attn_bwd_kernel(
     if (act == 0) {
       float pv = col_ok ? ps[i * (MAXL + 1) + lane] : 0.f;  // pre-mask P
       float da = dp;
-      if (dropout_p > 0.f && col_ok) {
-        unsigned long long gidx = IDX4(b, h, i, lane, H, Lq, Lk);
-        da = drop_mask[gidx] ? da * inv_keep : 0.f;
-      }
-      if (query_mask) da *= query_mask[(int64_t)b * Lq + i];
+      if (has_am && col_ok) da *= am[i * (MAXL + 1) + lane];
       float dot = wave_sum(da * pv);
       ds = pv * (da - dot);
     } else {
@@ -216,7 +226,7 @@ __global__ void attn_bwd_kernel(
   __syncthreads();
 
   // ---- phase 2: dQ rows (lane owns head-dim column t)
-  for (int i = wid; i < Lq; i += 4) {
+  for (int i = wid; i < Lq; i += NWAVES) {
     if (lane < D) {
       float acc = 0.f;
       const float* dsr = dss + i * (MAXL + 1);
@@ -227,7 +237,7 @@ __global__ void attn_bwd_kernel(
   }
 
   // ---- phase 3: dK, dV rows (wave per k-row j, lane owns column t)
-  for (int j = wid; j < Lk; j += 4) {
+  for (int j = wid; j < Lk; j += NWAVES) {
     if (lane < D) {
       float acc_k = 0.f, acc_v = 0.f;
       for (int i = 0; i < Lq; ++i) {
@@ -235,11 +245,7 @@ __global__ void attn_bwd_kernel(
         acc_k += ds * qs[i * MAXD + lane];
         float a = ps[i * (MAXL + 1) + j];
         if (act == 0) {
-          if (query_mask) a *= query_mask[(int64_t)b * Lq + i];
-          if (dropout_p > 0.f) {
-            unsigned long long gidx = IDX4(b, h, i, j, H, Lq, Lk);
-            a = drop_mask[gidx] ? a * inv_keep : 0.f;
-          }
+          if (has_am) a *= am[i * (MAXL + 1) + j];
         } else {
           a = a * sigmoidf_dev(a);  // A = silu(S)
         }
@@ -293,10 +299,10 @@ std::vector<torch::Tensor> attn_fwd(
   if (query_mask.has_value()) qm_f = query_mask->to(torch::kFloat32).contiguous();
 
   const int q_tile = std::min(Lq, MAXL);
-  dim3 block(256);
+  dim3 block(64 * NWAVES);
   dim3 grid(B * H, (Lq + q_tile - 1) / q_tile);
   size_t smem = (MAXD * (MAXL + 1) + MAXL * MAXD + q_tile * MAXD +
-                 4 * (MAXL + 1)) * sizeof(float);
+                 NWAVES * (MAXL + 1)) * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_ATTN_FWD(T)                                                     \
@@ -355,10 +361,10 @@ std::vector<torch::Tensor> attn_bwd(
   torch::Tensor qm_f;
   if (query_mask.has_value()) qm_f = query_mask->to(torch::kFloat32).contiguous();
 
-  dim3 block(256);
+  dim3 block(64 * NWAVES);
   dim3 grid(B * H);
   size_t smem = (MAXD * (MAXL + 1) + 3 * MAXL * MAXD +
-                 2 * MAXL * (MAXL + 1)) * sizeof(float);
+                 3 * MAXL * (MAXL + 1)) * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_ATTN_BWD(T)                                                     \

@@ -101,9 +101,19 @@ __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
       dw_acc[ci] += dyf * xf * r;
     }
   }
+  // block-level reduce of the 4 waves' partials in LDS, then ONE partial
+  // row per block into global scratch [n_blocks, d] (summed by ATen after
+  // the launch — no atomics, deterministic).
+  extern __shared__ __attribute__((aligned(16))) float dw_lds[];  // [4][d]
   int ci = 0;
   for (int j = lane; j < d; j += WAVE, ++ci) {
-    if (dw_acc[ci] != 0.f) atomicAdd(&dw[j], dw_acc[ci]);
+    dw_lds[wave_in_block * d + j] = dw_acc[ci];
+  }
+  __syncthreads();
+  for (int j = (int)threadIdx.x; j < d; j += (int)blockDim.x) {
+    float s = dw_lds[j] + dw_lds[d + j] + dw_lds[2 * d + j] +
+              dw_lds[3 * d + j];
+    dw[(int64_t)blockIdx.x * d + j] = s;
   }
 }
 
@@ -227,10 +237,12 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   auto w_f32 = w.to(torch::kFloat32).contiguous();
   TORCH_CHECK(d <= WAVE * 16, "rms_norm_bwd: d too large");
   auto dx = torch::empty_like(x);
-  auto dw = torch::zeros({d}, x.options().dtype(torch::kFloat32));
   dim3 block(256);
-  dim3 grid(grid_for_rows(n_rows, 4));
-  size_t smem = 0;
+  int n_blocks = std::min(grid_for_rows(n_rows, 4), 1024);
+  dim3 grid(n_blocks);
+  auto dw = torch::empty({n_blocks, (int64_t)d},
+                         x.options().dtype(torch::kFloat32));
+  size_t smem = 4 * (size_t)d * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_RMSB(T, OutT)                                                   \
@@ -250,7 +262,7 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
     TORCH_CHECK(false, "rms_norm_bwd: unsupported dtype");
   }
 #undef LAUNCH_RMSB
-  auto dw_out = dw.to(w.scalar_type());
+  auto dw_out = dw.sum(0).to(w.scalar_type());
   return {dx, dw_out};
 }
 
