@@ -1,0 +1,191 @@
+"""NoteLLM: query-to-embedding LLM (arXiv:2403.01744).
+
+Parity target: /root/reference/genrec/models/notellm.py (265 LoC):
+[EMB]-token pooling (notellm.py:113-129), pairwise contrastive loss with
+learnable temperature + hard-negative handling (notellm.py:170-189),
+optional category-generation auxiliary CE (notellm.py:196-202),
+device-cached shared-prefix KV cache (notellm.py:20-41), top-k retrieval
+accuracy metric (notellm.py:236-265). Model-only capability — the
+reference ships no trainer/config for it either.
+
+Offline-first like LCRec: with no pretrained directory the backbone is a
+random-init Qwen2 from config with a byte-level tokenizer.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional, Union
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from genrec_amd.config import ginlite
+
+
+class PrefixKVCache:
+    """Shared-prefix KV cache replicated per device on demand
+    (ref notellm.py:20-41)."""
+
+    def __init__(self, cache):
+        from transformers import cache_utils
+
+        self.fixed_len = cache.get_seq_length()
+        self._base = cache
+        self._per_device: Dict[torch.device, object] = {}
+
+    def to(self, device: torch.device):
+        from transformers import DynamicCache
+
+        obj = self._per_device.get(device)
+        if obj is None:
+            obj = DynamicCache()
+            legacy = self._base.to_legacy_cache()
+            obj = DynamicCache.from_legacy_cache(tuple(
+                (k.to(device), v.to(device)) for k, v in legacy))
+            self._per_device[device] = obj
+        obj.crop(self.fixed_len)
+        return obj
+
+
+@ginlite.configurable(name="Query2Embedding")
+class Query2Embedding(nn.Module):
+    def __init__(self, pretrained_path: Optional[str] = None,
+                 config=None, freeze_lm: bool = False,
+                 item_token: str = "[EMB]", pad_token: str = "<|pad|>",
+                 eos_token: str = "<|endoftext|>",
+                 gradient_checkpointing: bool = True, tau: float = 3.0,
+                 alpha: float = 0.01, hardneg_r: float = 0.1) -> None:
+        super().__init__()
+        self.tau = nn.Parameter(torch.tensor(tau))
+        self.alpha, self.hardneg_r = alpha, hardneg_r
+        self.item_token = item_token
+        if pretrained_path and os.path.isdir(pretrained_path):
+            from transformers import AutoModelForCausalLM, AutoTokenizer
+
+            self.tokenizer = AutoTokenizer.from_pretrained(
+                pretrained_path, pad_token=pad_token, eos_token=eos_token)
+            self.model = AutoModelForCausalLM.from_pretrained(pretrained_path)
+        else:
+            from transformers import AutoModelForCausalLM
+
+            from genrec_amd.models.lcrec import default_qwen_config
+            from genrec_amd.utils.tokenizer import build_offline_tokenizer
+
+            self.tokenizer = build_offline_tokenizer(eos_token=eos_token)
+            cfg = config or default_qwen_config(
+                vocab_size=max(len(self.tokenizer) + 8, 512))
+            self.model = AutoModelForCausalLM.from_config(cfg)
+        if freeze_lm:
+            for p in self.model.parameters():
+                p.requires_grad = False
+        self.tokenizer.add_tokens([item_token], special_tokens=True)
+        self.emb_id = self.tokenizer.convert_tokens_to_ids(item_token)
+        self.model.resize_token_embeddings(len(self.tokenizer))
+        self.model.config.vocab_size = len(self.tokenizer)
+        if gradient_checkpointing:
+            self.model.gradient_checkpointing_enable()
+
+    def tokenize(self, query: List[str], category: Optional[List[str]] = None,
+                 score: Optional[List[float]] = None) -> Dict[str, torch.Tensor]:
+        token = self.tokenizer(
+            query, category, padding=True,
+            return_token_type_ids=category is not None, return_tensors="pt")
+        token["emb_token_idx"] = (token.input_ids == self.emb_id).int() \
+            .argmax(1, keepdim=True)
+        if category:
+            token["labels"] = token.input_ids.where(
+                token.pop("token_type_ids").bool(), torch.tensor(-100))
+        if score:
+            token["hardneg"] = torch.tensor(score) < self.hardneg_r
+        return token
+
+    def get_embedding(self, input_ids, attention_mask, emb_token_idx,
+                      past_key_values=None):
+        outputs = self.model.model(
+            input_ids=input_ids, attention_mask=attention_mask,
+            past_key_values=past_key_values)
+        hidden = outputs.last_hidden_state
+        idx = emb_token_idx.repeat(1, self.model.config.hidden_size) \
+            .unsqueeze(1)
+        emb = hidden.gather(1, idx).squeeze(1)
+        return F.normalize(emb, p=2, dim=1), hidden
+
+    def forward(self, input_ids, attention_mask, emb_token_idx,
+                labels=None, hardneg=None, past_key_values=None,
+                return_loss: bool = True):
+        if past_key_values is not None:
+            if isinstance(past_key_values, PrefixKVCache):
+                past_key_values = past_key_values.to(input_ids.device)
+            attention_mask = torch.cat([
+                torch.ones(attention_mask.size(0),
+                           past_key_values.get_seq_length(),
+                           dtype=attention_mask.dtype,
+                           device=attention_mask.device),
+                attention_mask], dim=1)
+
+        outputs: Dict[str, torch.Tensor] = {}
+        outputs["sentence_embedding"], hidden = self.get_embedding(
+            input_ids, attention_mask, emb_token_idx, past_key_values)
+        if not return_loss:
+            return {"sentence_embedding": outputs["sentence_embedding"]}
+
+        # pairwise contrastive: rows alternate (anchor, positive)
+        emb = outputs["sentence_embedding"]
+        sim = torch.mm(F.normalize(emb[::2], p=2, dim=1),
+                       F.normalize(emb[1::2], p=2, dim=1).T)
+        log_softmax = -(sim * self.tau.exp()).softmax(dim=1).diag().log()
+        if hardneg is not None:
+            outputs["hardneg"] = hardneg
+            if hardneg.any():
+                keep = torch.ones_like(hardneg, dtype=torch.bool)
+                keep[hardneg] = False
+                log_softmax = torch.cat([
+                    log_softmax[keep],
+                    (sim[hardneg].mean(dim=1) + 1).log() * self.hardneg_r])
+        cl_loss = log_softmax.mean()
+
+        if labels is None or bool((labels < 0).all()):
+            outputs["loss"] = cl_loss
+            return outputs
+        logits = self.model.lm_head(hidden)
+        shift_logits = logits[:, :-1].contiguous()
+        shift_labels = labels[:, 1:].contiguous()
+        gen_loss = F.cross_entropy(
+            shift_logits.view(-1, shift_logits.size(-1)),
+            shift_labels.view(-1), ignore_index=-100)
+        outputs["loss"] = (cl_loss + gen_loss * self.alpha) / (1 + self.alpha)
+        return outputs
+
+    def save_pretrained(self, save_dir: str, **kwargs):
+        self.model.save_pretrained(save_dir, **kwargs)
+        self.tokenizer.save_pretrained(save_dir)
+
+    @torch.no_grad()
+    def generate(self, inputs: Union[torch.Tensor, Dict], **kwargs):
+        if isinstance(inputs, torch.Tensor):
+            inputs = {"input_ids": inputs}
+        elif not isinstance(inputs, dict):
+            raise ValueError("inputs must be Tensor or Dict[str, Tensor]")
+        return self.model.generate(**inputs, **kwargs)
+
+    @staticmethod
+    def topk_retrieval_accuracy(pred: torch.Tensor, topk: int = 5,
+                                batch_size: int = 64,
+                                hardneg: Optional[torch.Tensor] = None) -> float:
+        """Top-k accuracy of anchor->positive retrieval in eval batches
+        (ref notellm.py:236-265)."""
+        if hardneg is not None:
+            pred1, pred2 = pred[::2][~hardneg], pred[1::2][~hardneg]
+        else:
+            pred1, pred2 = pred[::2], pred[1::2]
+        correct = 0
+        n = pred1.size(0) // batch_size * batch_size
+        for i in range(0, n, batch_size):
+            sim = torch.mm(F.normalize(pred1[i:i + batch_size], p=2, dim=1),
+                           F.normalize(pred2[i:i + batch_size], p=2, dim=1).T)
+            topk_idx = sim.topk(topk, dim=0).indices
+            true_idx = torch.arange(sim.size(0), device=topk_idx.device)
+            correct += (topk_idx == true_idx).sum().item()
+        return correct / max(pred1.size(0), 1)

@@ -142,3 +142,16 @@ def init_wandb(project: str, config: dict, enabled: bool, is_main: bool):
     except Exception:
         logger.warning("wandb unavailable; logging disabled")
         return WandbStub()
+
+
+def dataset_kwargs(ds_cls, wanted: Dict[str, Any]) -> Dict[str, Any]:
+    """Filter dataset-constructor kwargs by signature; classes accepting
+    **kwargs receive everything (subclasses wrapping a base via **kw)."""
+    import inspect
+
+    sig = inspect.signature(ds_cls.__init__)
+    has_var_kw = any(p.kind == inspect.Parameter.VAR_KEYWORD
+                     for p in sig.parameters.values())
+    if has_var_kw:
+        return dict(wanted)
+    return {k: v for k, v in wanted.items() if k in sig.parameters}

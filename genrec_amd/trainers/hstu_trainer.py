@@ -91,12 +91,11 @@ def train(
 
     def mk(mode):
         sig = inspect.signature(ds_cls.__init__)
-        kw = {"split": mode}
-        if "max_seq_len" in sig.parameters:
-            kw["max_seq_len"] = max_seq_len
-        if "root" in sig.parameters:
-            kw.update(root=dataset_folder, split=split, train_test_split=mode)
-        return ds_cls(**kw)
+        if "root" in sig.parameters:  # real-data pipelines
+            return ds_cls(root=dataset_folder, split=split,
+                          train_test_split=mode, max_seq_len=max_seq_len)
+        return ds_cls(**common.dataset_kwargs(
+            ds_cls, {"split": mode, "max_seq_len": max_seq_len}))
 
     train_ds, valid_ds, test_ds = mk("train"), mk("valid"), mk("test")
     model = HSTU(num_items=train_ds.num_items, max_seq_len=max_seq_len,

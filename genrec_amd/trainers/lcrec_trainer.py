@@ -276,18 +276,15 @@ def _make_dataset(ds_cls, mode, folder, split, n_codebooks, codebook_size,
     import inspect
 
     sig = inspect.signature(ds_cls.__init__)
-    kw = {"split": mode}
-    if "sem_id_dim" in sig.parameters:
-        kw["sem_id_dim"] = n_codebooks
-    if "codebook_size" in sig.parameters:
-        kw["codebook_size"] = codebook_size
-    if "max_samples" in sig.parameters:
-        kw["max_samples"] = max_samples
-    if "root" in sig.parameters:
-        kw.update(root=folder, split=split, train_test_split=mode)
-    if "pretrained_rqvae_path" in sig.parameters and rqvae_path:
-        kw["pretrained_rqvae_path"] = rqvae_path
-    return ds_cls(**kw)
+    if "root" in sig.parameters:  # real-data pipelines
+        kw = dict(root=folder, split=split, train_test_split=mode,
+                  max_samples=max_samples)
+        if rqvae_path:
+            kw["pretrained_rqvae_path"] = rqvae_path
+        return ds_cls(**kw)
+    return ds_cls(**common.dataset_kwargs(ds_cls, {
+        "split": mode, "sem_id_dim": n_codebooks,
+        "codebook_size": codebook_size, "max_samples": max_samples}))
 
 
 if __name__ == "__main__":

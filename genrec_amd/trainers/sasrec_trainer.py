@@ -176,15 +176,12 @@ def train(
 def _make(ds_cls, split: str, mode: str, max_seq_len: int, folder: str):
     import inspect
 
-    kwargs = dict(split=mode)
     sig = inspect.signature(ds_cls.__init__)
-    if "max_seq_len" in sig.parameters:
-        kwargs["max_seq_len"] = max_seq_len
-    if "root" in sig.parameters:
-        kwargs["root"] = folder
-        kwargs["split"] = split
-        kwargs["train_test_split"] = mode
-    return ds_cls(**kwargs)
+    if "root" in sig.parameters:  # real-data pipelines
+        return ds_cls(root=folder, split=split, train_test_split=mode,
+                      max_seq_len=max_seq_len)
+    return ds_cls(**common.dataset_kwargs(
+        ds_cls, {"split": mode, "max_seq_len": max_seq_len}))
 
 
 if __name__ == "__main__":

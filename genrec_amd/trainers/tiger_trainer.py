@@ -192,22 +192,16 @@ def _make_dataset(ds_cls, split, mode, folder, max_items, sem_id_dim,
     import inspect
 
     sig = inspect.signature(ds_cls.__init__)
-    kwargs = {}
-    if "split" in sig.parameters:
-        kwargs["split"] = mode
-    if "max_items_per_seq" in sig.parameters:
-        kwargs["max_items_per_seq"] = max_items
-    if "sem_id_dim" in sig.parameters:
-        kwargs["sem_id_dim"] = sem_id_dim
-    if "codebook_size" in sig.parameters:
-        kwargs["codebook_size"] = codebook_size
-    if "root" in sig.parameters:
-        kwargs["root"] = folder
-        kwargs["split"] = split
-        kwargs["train_test_split"] = mode
-    if "rqvae_path" in sig.parameters and rqvae_path:
-        kwargs["rqvae_path"] = rqvae_path
-    return ds_cls(**kwargs)
+    if "root" in sig.parameters:  # real-data pipelines
+        kw = dict(root=folder, split=split, train_test_split=mode,
+                  max_items_per_seq=max_items)
+        if rqvae_path:
+            kw["pretrained_rqvae_path"] = rqvae_path
+        return ds_cls(**kw)
+    kw = common.dataset_kwargs(ds_cls, {
+        "split": mode, "max_items_per_seq": max_items,
+        "sem_id_dim": sem_id_dim, "codebook_size": codebook_size})
+    return ds_cls(**kw)
 
 
 if __name__ == "__main__":

@@ -1,0 +1,134 @@
+"""COBRA + NoteLLM CPU tests."""
+
+import pytest
+import torch
+
+torch.manual_seed(0)
+
+TINY_COBRA = dict(encoder_n_layers=1, encoder_hidden_dim=32,
+                  encoder_num_heads=4, encoder_vocab_size=100,
+                  id_vocab_size=16, n_codebooks=3, d_model=32,
+                  decoder_n_layers=2, decoder_num_heads=4,
+                  decoder_dropout=0.0)
+
+
+@pytest.fixture(scope="module")
+def cobra():
+    from genrec_amd.models.cobra import Cobra
+
+    torch.manual_seed(0)
+    return Cobra(**TINY_COBRA)
+
+
+def _batch(model, B=3, T=4, L=6):
+    C = model.C
+    ids = torch.randint(0, 16, (B, T * C))
+    enc = torch.randint(1, 100, (B, T, L))
+    return ids, enc
+
+
+def test_cobra_interleave_mask(cobra):
+    m = torch.tensor([[1, 1, 1, 1, 1, 1], [1, 1, 1, 0, 0, 0]]).bool()
+    out = cobra.interleave_seq_mask(m, 3)
+    # [s s s d s s s d]
+    assert out.shape == (2, 8)
+    assert out[0].tolist() == [True] * 8
+    assert out[1].tolist() == [True, True, True, True, False, False, False,
+                               False]
+
+
+def test_cobra_forward_losses(cobra):
+    ids, enc = _batch(cobra)
+    out = cobra(ids, enc)
+    assert torch.isfinite(out.loss)
+    assert out.acc_total > 0 and out.recall_total > 0
+    assert out.codebook_entropy > 0
+    out.loss.backward()
+
+
+def test_cobra_padded_rows(cobra):
+    ids, enc = _batch(cobra, B=2, T=4)
+    ids[1, 6:] = cobra.pad_id  # second row has only 2 complete items
+    out = cobra(ids, enc)
+    assert torch.isfinite(out.loss)
+
+
+def test_cobra_generate_and_fusion(cobra):
+    cobra.eval()
+    ids, enc = _batch(cobra, B=2, T=3)
+    gen = cobra.generate(ids, enc, n_candidates=4)
+    assert gen.sem_ids.shape == (2, 4, 3)
+    assert gen.dense_vecs.shape == (2, 4, 32)
+    assert torch.allclose(gen.dense_vecs.norm(dim=-1),
+                          torch.ones(2, 4), atol=1e-4)
+    # scores sorted desc
+    assert (gen.scores[:, :-1] >= gen.scores[:, 1:]).all()
+
+    n_items = 20
+    item_vecs = torch.nn.functional.normalize(torch.randn(n_items, 32), dim=-1)
+    item_sem = torch.randint(0, 16, (n_items, 3))
+    fus = cobra.beam_fusion(ids, enc, item_vecs, item_sem,
+                            n_candidates=5, n_beam=8)
+    assert fus.item_ids.shape == (2, 5)
+    assert fus.sem_ids.shape == (2, 5, 3)
+    assert (fus.item_ids < n_items).all()
+
+
+def test_cobra_trainer_smoke(tmp_path):
+    from genrec_amd.data.cobra_synthetic import SyntheticCobraDataset
+    from genrec_amd.trainers import cobra_trainer
+
+    class Tiny(SyntheticCobraDataset):
+        def __init__(self, **kw):
+            kw.update(num_users=40, num_items=50, text_vocab_size=100,
+                      max_text_len=6)
+            super().__init__(**kw)
+
+    cobra_trainer.train(
+        dataset=Tiny, epochs=1, max_steps=2, batch_size=8,
+        n_codebooks=3, id_vocab_size=16, d_model=32, decoder_n_layers=2,
+        decoder_num_heads=4, encoder_n_layers=1, amp=False,
+        save_dir_root=str(tmp_path), eval_every_epoch=1, eval_n_beam=6,
+        eval_max_batches=1, num_warmup_steps=1)
+    import os
+
+    assert os.path.exists(os.path.join(str(tmp_path), "checkpoint_final.pt"))
+
+
+def test_notellm_contrastive():
+    from genrec_amd.models.lcrec import default_qwen_config
+    from genrec_amd.models.notellm import Query2Embedding
+
+    torch.manual_seed(0)
+    m = Query2Embedding(config=default_qwen_config(
+        vocab_size=512, hidden_size=32, num_layers=2, num_heads=4,
+        num_kv_heads=2, intermediate_size=64), gradient_checkpointing=False)
+    queries = []
+    for i in range(4):
+        queries.append(f"note {i} text [EMB]")
+        queries.append(f"related note {i} [EMB]")
+    tok = m.tokenize(queries)
+    out = m(tok["input_ids"], tok["attention_mask"], tok["emb_token_idx"])
+    assert torch.isfinite(out["loss"])
+    out["loss"].backward()
+    assert m.tau.grad is not None
+    emb = out["sentence_embedding"].detach()
+    assert torch.allclose(emb.norm(dim=1), torch.ones(8), atol=1e-4)
+    acc = Query2Embedding.topk_retrieval_accuracy(emb, topk=2, batch_size=4)
+    assert 0.0 <= acc <= 1.0
+
+
+def test_notellm_hardneg_and_labels():
+    from genrec_amd.models.lcrec import default_qwen_config
+    from genrec_amd.models.notellm import Query2Embedding
+
+    torch.manual_seed(1)
+    m = Query2Embedding(config=default_qwen_config(
+        vocab_size=512, hidden_size=32, num_layers=2, num_heads=4,
+        num_kv_heads=2, intermediate_size=64), gradient_checkpointing=False)
+    queries = [f"q{i} [EMB]" for i in range(6)]
+    tok = m.tokenize(queries, score=[0.5, 0.05, 0.9])
+    assert tok["hardneg"].tolist() == [False, True, False]
+    out = m(tok["input_ids"], tok["attention_mask"], tok["emb_token_idx"],
+            hardneg=tok["hardneg"])
+    assert torch.isfinite(out["loss"])
