@@ -26,6 +26,20 @@ std::vector<torch::Tensor> attn_bwd(
     double scale, int64_t act, double dropout_p, int64_t seed,
     bool bias_grad, int64_t bias_dim);
 
+std::vector<torch::Tensor> attn_fwd_mfma(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
+    c10::optional<torch::Tensor> add_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, bool causal, int64_t act, double dropout_p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev);
+std::vector<torch::Tensor> attn_bwd_mfma(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor p_saved, torch::Tensor drop_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, int64_t act, double dropout_p, int64_t seed,
+    bool bias_grad, int64_t bias_dim);
+
 std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
                                           torch::Tensor targets,
                                           int64_t ignore_index);
@@ -49,6 +63,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_bwd", &genrec::l2norm_bwd, "fused L2Norm backward");
   m.def("attn_fwd", &genrec::attn_fwd, "fused attention forward");
   m.def("attn_bwd", &genrec::attn_bwd, "fused attention backward");
+  m.def("attn_fwd_mfma", &genrec::attn_fwd_mfma, "MFMA attention forward");
+  m.def("attn_bwd_mfma", &genrec::attn_bwd_mfma, "MFMA attention backward");
   m.def("softmax_ce_fwd", &genrec::softmax_ce_fwd, "fused CE forward");
   m.def("softmax_ce_bwd", &genrec::softmax_ce_bwd, "fused CE backward");
   m.def("sqdist_argmin", &genrec::sqdist_argmin, "L2 dist + argmin");

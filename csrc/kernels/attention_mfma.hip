@@ -1,0 +1,493 @@
+// MFMA fused attention (bf16) — gfx950 matrix-core path for K1/K4/K13.
+//
+// Forward: one 256-thread workgroup (4 waves) per (batch, head). Q, K and
+// V^T are staged in LDS as bf16 with the T2 XOR swizzle (byte ^= (row&7)<<4)
+// so ds_read_b128 fragment loads are conflict-free. Each wave owns a
+// 16-row q-strip: S = Q K^T runs as 4 column-fragments x (D/32) K-steps of
+// v_mfma_f32_16x16x32_bf16; scale/bias/masks/softmax(or SiLU) are applied
+// on the accumulator fragments in-register (the C/D layout places row
+// (lane>>4)*4+reg, col lane&15 — a row reduction is an fmax/fadd over the
+// 4 column fragments followed by shuffle-xor over the 16-lane group);
+// P is written to LDS (bf16, swizzled) and consumed directly as the
+// A-operand of the P V MFMA. Scores never touch HBM; P is saved fp32 for
+// backward.
+//
+// Backward: attn_bwd_ds_kernel computes dP = dO V^T with the same MFMA
+// geometry, applies the dropout/query-mask multiplier and the softmax
+// Jacobian (or SiLU') in-register, and writes dS (bf16) and the masked
+// probabilities A_d (bf16) to global. dQ/dK/dV are then three plain
+// strided-batched GEMMs done by hipBLASLt from the host wrapper — exactly
+// the "library GEMMs stay in the library" split.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../core/common.h"
+
+namespace genrec {
+
+constexpr int TILE = 64;               // max Lq/Lk per block
+constexpr float NEG_BIG_MF = -1e9f;
+
+typedef __attribute__((ext_vector_type(8))) short short8v;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+
+#define IDX4M(b, h, i, j, H, I, J) \
+  ((((int64_t)(b) * (H) + (h)) * (I) + (i)) * (J) + (j))
+
+// LDS tile: [64][64] bf16, row stride 128 B, XOR-swizzled byte offset.
+__device__ __forceinline__ int swz(int row, int byte_in_row) {
+  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
+}
+
+// load one 16x32 A/B fragment (8 bf16 = 16 B per lane) from a swizzled tile
+__device__ __forceinline__ short8v frag_load(const char* base, int row0,
+                                             int k0, int lane) {
+  int row = row0 + (lane & 15);
+  int byte = (k0 + ((lane >> 4) << 3)) * 2;
+  return *reinterpret_cast<const short8v*>(base + swz(row, byte));
+}
+
+template <bool IS_SILU>
+__global__ void __launch_bounds__(256)
+attn_fwd_mfma_kernel(
+    const __hip_bfloat16* __restrict__ q,   // [B,H,Lq,D]
+    const __hip_bfloat16* __restrict__ k,   // [B,H,Lk,D]
+    const __hip_bfloat16* __restrict__ v,   // [B,H,Lk,D]
+    const float* __restrict__ bias,         // null | [H,Lq,Lk] | [B,H,Lq,Lk]
+    const bool* __restrict__ key_pad,       // null | [B,Lk]
+    const float* __restrict__ add_mask,     // null | [Lq,Lk]
+    const float* __restrict__ query_mask,   // null | [B,Lq]
+    __hip_bfloat16* __restrict__ out,       // [B,H,Lq,D]
+    float* __restrict__ p_saved,            // [B,H,Lq,Lk]
+    unsigned char* __restrict__ drop_mask,
+    const unsigned int* __restrict__ seed_dev,
+    int B, int H, int Lq, int Lk, int D,
+    float scale, int bias_dim, bool causal,
+    float dropout_p, unsigned int seed, int q_tile) {
+  if (seed_dev) seed += *seed_dev;
+  const int bh = blockIdx.x;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.y * q_tile;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* qs = smem;                 // [64][128B] swizzled bf16
+  char* ks = qs + TILE * 128;      // [64][128B]
+  char* vt = ks + TILE * 128;      // [64(d)][128B(j)] transposed V
+  char* ps = vt + TILE * 128;      // [64][128B] P (bf16)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  // ---- stage: coalesced global rows -> swizzled LDS rows (zero-padded)
+  // each thread copies 8 bf16 (16 B); row-major walk
+  for (int idx = tid; idx < TILE * (TILE / 8); idx += blockDim.x) {
+    int row = idx / (TILE / 8);
+    int c8 = idx % (TILE / 8);        // 8-elem chunk
+    int d0 = c8 * 8;
+    short8v z = {};
+    // Q tile
+    short8v val = z;
+    int qi = q0 + row;
+    if (qi < Lq && d0 < D) {
+      val = *reinterpret_cast<const short8v*>(
+          &q[IDX4M(b, h, qi, d0, H, Lq, D)]);
+    }
+    *reinterpret_cast<short8v*>(qs + swz(row, d0 * 2)) = val;
+    // K tile
+    val = z;
+    if (row < Lk && d0 < D) {
+      val = *reinterpret_cast<const short8v*>(
+          &k[IDX4M(b, h, row, d0, H, Lk, D)]);
+    }
+    *reinterpret_cast<short8v*>(ks + swz(row, d0 * 2)) = val;
+    // V^T tile: vt[d][j] = V[j][d]; this thread owns (d=row, j=d0..d0+7)
+    __hip_bfloat16 tmp[8];
+    for (int jj = 0; jj < 8; ++jj) {
+      int j = d0 + jj;
+      tmp[jj] = (j < Lk && row < D)
+          ? v[IDX4M(b, h, j, row, H, Lk, D)] : __hip_bfloat16(0.f);
+    }
+    *reinterpret_cast<short8v*>(vt + swz(row, d0 * 2)) =
+        *reinterpret_cast<short8v*>(tmp);
+  }
+  __syncthreads();
+
+  const int strip = wid * 16;            // this wave's q rows [strip, strip+16)
+  const float inv_keep = dropout_p > 0.f ? 1.0f / (1.0f - dropout_p) : 1.0f;
+
+  // ---- S = Q K^T : 4 column fragments
+  float4v acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    for (int kk = 0; kk < D; kk += 32) {
+      short8v a = frag_load(qs, strip, kk, lane);
+      short8v bfr = frag_load(ks, f * 16, kk, lane);
+      acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc[f], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue on fragments: scale + bias + masks
+  const int col_base = lane & 15;
+  const int row_grp = (lane >> 4) << 2;  // rows row_grp..row_grp+3 (in strip)
+  float s_val[4][4];                     // [fragment][reg]
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = q0 + strip + row_grp + r;   // global q row
+      int j = f * 16 + col_base;          // key col
+      float s = acc[f][r];
+      bool ok = (i < Lq) && (j < Lk);
+      if (ok) {
+        s *= scale;
+        if (bias_dim == 3) s += bias[((int64_t)h * Lq + i) * Lk + j];
+        else if (bias_dim == 4) s += bias[IDX4M(b, h, i, j, H, Lq, Lk)];
+        if (causal && j > i) s = NEG_BIG_MF;
+        if (key_pad && key_pad[(int64_t)b * Lk + j]) s = NEG_BIG_MF;
+        if (add_mask) s += add_mask[(int64_t)i * Lk + j];
+      } else {
+        s = -INFINITY;
+      }
+      s_val[f][r] = s;
+    }
+  }
+
+  float p_val[4][4];
+  if (IS_SILU) {
+#pragma unroll
+    for (int f = 0; f < 4; ++f)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = s_val[f][r];
+        p_val[f][r] = (s == -INFINITY) ? 0.f : s * sigmoidf_dev(s);
+      }
+  } else {
+    // row softmax: combine 4 col fragments per reg, reduce over 16-lane grp
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float m = fmaxf(fmaxf(s_val[0][r], s_val[1][r]),
+                      fmaxf(s_val[2][r], s_val[3][r]));
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        m = fmaxf(m, __shfl_xor(m, off, 64));
+      float sum = 0.f;
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        float e = (s_val[f][r] == -INFINITY) ? 0.f
+                                             : __expf(s_val[f][r] - m);
+        p_val[f][r] = e;
+        sum += e;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        sum += __shfl_xor(sum, off, 64);
+      float inv = (sum > 0.f) ? 1.0f / sum : 0.f;
+#pragma unroll
+      for (int f = 0; f < 4; ++f) p_val[f][r] *= inv;
+    }
+  }
+
+  // ---- save P/S, apply post-softmax query mask + dropout, stash P in LDS
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = q0 + strip + row_grp + r;
+      int j = f * 16 + col_base;
+      float p = p_val[f][r];
+      if (i < Lq && j < Lk) {
+        p_saved[IDX4M(b, h, i, j, H, Lq, Lk)] = IS_SILU ? s_val[f][r] : p;
+        if (query_mask) p *= query_mask[(int64_t)b * Lq + i];
+        if (dropout_p > 0.f) {
+          unsigned long long gidx = IDX4M(b, h, i, j, H, Lq, Lk);
+          bool keep = (hash_rng(seed, gidx) & 0xFFFFFF) >=
+                      (unsigned int)(dropout_p * 16777216.0f);
+          drop_mask[gidx] = keep;
+          p = keep ? p * inv_keep : 0.f;
+        }
+      } else {
+        p = 0.f;
+      }
+      // bf16 scatter into swizzled P tile (2 B per element)
+      int row = strip + row_grp + r;
+      *reinterpret_cast<__hip_bfloat16*>(ps + swz(row, j * 2)) =
+          __float2bfloat16(p);
+    }
+  }
+  // wave-local: this wave only reads its own strip rows back
+  __builtin_amdgcn_wave_barrier();
+
+  // ---- O = P V : A = P strip rows (k=j), B = vt rows (n=d)
+  float4v acc2[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                     {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  const int nfrag_d = (D + 15) / 16;
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    if (f >= nfrag_d) break;
+    for (int kk = 0; kk < TILE; kk += 32) {
+      short8v a = frag_load(ps, strip, kk, lane);
+      short8v bfr = frag_load(vt, f * 16, kk, lane);
+      acc2[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc2[f],
+                                                        0, 0, 0);
+    }
+  }
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    if (f >= nfrag_d) break;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = q0 + strip + row_grp + r;
+      int d = f * 16 + col_base;
+      if (i < Lq && d < D) {
+        out[IDX4M(b, h, i, d, H, Lq, D)] = __float2bfloat16(acc2[f][r]);
+      }
+    }
+  }
+}
+
+// ---- backward dS kernel: dP = dO V^T (MFMA), then Jacobian in-register.
+// Writes dS (bf16, unscaled) and A_d (bf16, the post-mask probabilities)
+// to global for the host-side batched GEMMs.
+template <bool IS_SILU>
+__global__ void __launch_bounds__(256)
+attn_bwd_ds_kernel(
+    const __hip_bfloat16* __restrict__ dout,  // [B,H,Lq,D]
+    const __hip_bfloat16* __restrict__ v,     // [B,H,Lk,D]
+    const float* __restrict__ p_saved,        // [B,H,Lq,Lk] (P or S)
+    const float* __restrict__ query_mask,     // null | [B,Lq]
+    const unsigned char* __restrict__ drop_mask,
+    __hip_bfloat16* __restrict__ ds_out,      // [B,H,Lq,Lk]
+    __hip_bfloat16* __restrict__ ad_out,      // [B,H,Lq,Lk]
+    int B, int H, int Lq, int Lk, int D,
+    int act_silu_unused, float dropout_p) {
+  const int bh = blockIdx.x;
+  const int b = bh / H, h = bh % H;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* dos = smem;               // [64][128B] dO swizzled
+  char* vs = dos + TILE * 128;    // [64][128B] V rows
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  for (int idx = tid; idx < TILE * (TILE / 8); idx += blockDim.x) {
+    int row = idx / (TILE / 8);
+    int d0 = (idx % (TILE / 8)) * 8;
+    short8v val = {};
+    if (row < Lq && d0 < D) {
+      val = *reinterpret_cast<const short8v*>(
+          &dout[IDX4M(b, h, row, d0, H, Lq, D)]);
+    }
+    *reinterpret_cast<short8v*>(dos + swz(row, d0 * 2)) = val;
+    short8v val2 = {};
+    if (row < Lk && d0 < D) {
+      val2 = *reinterpret_cast<const short8v*>(
+          &v[IDX4M(b, h, row, d0, H, Lk, D)]);
+    }
+    *reinterpret_cast<short8v*>(vs + swz(row, d0 * 2)) = val2;
+  }
+  __syncthreads();
+
+  const int strip = wid * 16;
+  const float inv_keep = dropout_p > 0.f ? 1.0f / (1.0f - dropout_p) : 1.0f;
+
+  // dP = dO V^T : A = dO strip rows (k=d), B = V rows (n=j, k=d)
+  float4v acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    for (int kk = 0; kk < D; kk += 32) {
+      short8v a = frag_load(dos, strip, kk, lane);
+      short8v bfr = frag_load(vs, f * 16, kk, lane);
+      acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc[f], 0, 0, 0);
+    }
+  }
+
+  const int col_base = lane & 15;
+  const int row_grp = (lane >> 4) << 2;
+  float pv[4][4], da[4][4], ad[4][4];
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = strip + row_grp + r;
+      int j = f * 16 + col_base;
+      bool ok = (i < Lq) && (j < Lk);
+      float p = ok ? p_saved[IDX4M(b, h, i, j, H, Lq, Lk)] : 0.f;
+      float m = 1.f;
+      if (!IS_SILU && ok) {
+        if (query_mask) m *= query_mask[(int64_t)b * Lq + i];
+        if (dropout_p > 0.f) {
+          m *= drop_mask[IDX4M(b, h, i, j, H, Lq, Lk)] ? inv_keep : 0.f;
+        }
+      }
+      pv[f][r] = p;          // P (softmax) or S (silu)
+      da[f][r] = ok ? acc[f][r] * (IS_SILU ? 1.f : m) : 0.f;
+      ad[f][r] = IS_SILU ? (ok ? p * sigmoidf_dev(p) : 0.f) : p * m;
+    }
+  }
+
+  float ds[4][4];
+  if (IS_SILU) {
+#pragma unroll
+    for (int f = 0; f < 4; ++f)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = pv[f][r];
+        float sg = sigmoidf_dev(s);
+        ds[f][r] = da[f][r] * sg * (1.f + s * (1.f - sg));
+      }
+  } else {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float dot = da[0][r] * pv[0][r] + da[1][r] * pv[1][r] +
+                  da[2][r] * pv[2][r] + da[3][r] * pv[3][r];
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        dot += __shfl_xor(dot, off, 64);
+#pragma unroll
+      for (int f = 0; f < 4; ++f)
+        ds[f][r] = pv[f][r] * (da[f][r] - dot);
+    }
+  }
+
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = strip + row_grp + r;
+      int j = f * 16 + col_base;
+      if (i < Lq && j < Lk) {
+        int64_t gi = IDX4M(b, h, i, j, H, Lq, Lk);
+        ds_out[gi] = __float2bfloat16(ds[f][r]);
+        ad_out[gi] = __float2bfloat16(ad[f][r]);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------ hosts
+
+std::vector<torch::Tensor> attn_fwd_mfma(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
+    c10::optional<torch::Tensor> add_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, bool causal, int64_t act, double dropout_p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev) {
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
+  const int Lk = k.size(2);
+  TORCH_CHECK(Lk <= TILE && D <= TILE && D % 32 == 0);
+  auto out = torch::empty_like(q);
+  auto p_saved = torch::empty({B, H, Lq, Lk},
+                              q.options().dtype(torch::kFloat32));
+  torch::Tensor dmask;
+  if (dropout_p > 0) {
+    dmask = torch::empty({B, H, Lq, Lk}, q.options().dtype(torch::kUInt8));
+  } else {
+    dmask = torch::empty({0}, q.options().dtype(torch::kUInt8));
+  }
+  torch::Tensor bias_f;
+  int bias_dim = 0;
+  if (bias.has_value()) {
+    bias_f = bias->to(torch::kFloat32).contiguous();
+    bias_dim = bias_f.dim();
+  }
+  torch::Tensor am_f, qm_f;
+  if (add_mask.has_value()) am_f = add_mask->to(torch::kFloat32).contiguous();
+  if (query_mask.has_value())
+    qm_f = query_mask->to(torch::kFloat32).contiguous();
+
+  const int q_tile = TILE;
+  dim3 block(256);
+  dim3 grid(B * H, (Lq + q_tile - 1) / q_tile);
+  size_t smem = 4 * TILE * 128;
+  auto stream = at::cuda::getCurrentHIPStream();
+
+#define LAUNCH_FWD_MFMA(SILU)                                                  \
+  hipLaunchKernelGGL((attn_fwd_mfma_kernel<SILU>), grid, block, smem, stream,  \
+      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),                   \
+      reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),                   \
+      reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),                   \
+      bias_dim ? bias_f.data_ptr<float>() : nullptr,                           \
+      key_pad.has_value() ? key_pad->data_ptr<bool>() : nullptr,               \
+      add_mask.has_value() ? am_f.data_ptr<float>() : nullptr,                 \
+      query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,               \
+      reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),                       \
+      p_saved.data_ptr<float>(),                                               \
+      dropout_p > 0 ? dmask.data_ptr<unsigned char>() : nullptr,               \
+      seed_dev.has_value()                                                     \
+          ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())        \
+          : nullptr,                                                           \
+      B, H, Lq, Lk, D, (float)scale, bias_dim, causal, (float)dropout_p,       \
+      (unsigned int)seed, q_tile)
+
+  if (act == 0) LAUNCH_FWD_MFMA(false);
+  else LAUNCH_FWD_MFMA(true);
+#undef LAUNCH_FWD_MFMA
+  return {out, p_saved, dmask};
+}
+
+std::vector<torch::Tensor> attn_bwd_mfma(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor p_saved, torch::Tensor drop_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, int64_t act, double dropout_p, int64_t seed,
+    bool bias_grad, int64_t bias_dim) {
+  (void)seed;
+  const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
+  const int Lk = k.size(2);
+  TORCH_CHECK(Lq <= TILE && Lk <= TILE && D % 32 == 0);
+  auto opts_bf = q.options();
+  auto ds = torch::empty({B, H, Lq, Lk}, opts_bf);
+  auto ad = torch::empty({B, H, Lq, Lk}, opts_bf);
+  torch::Tensor qm_f;
+  if (query_mask.has_value())
+    qm_f = query_mask->to(torch::kFloat32).contiguous();
+  dim3 block(256);
+  dim3 grid(B * H);
+  size_t smem = 2 * TILE * 128;
+  auto stream = at::cuda::getCurrentHIPStream();
+
+#define LAUNCH_BWD_DS(SILU)                                                    \
+  hipLaunchKernelGGL((attn_bwd_ds_kernel<SILU>), grid, block, smem, stream,    \
+      reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),                 \
+      reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),                   \
+      p_saved.data_ptr<float>(),                                               \
+      query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,               \
+      dropout_p > 0 ? drop_mask.data_ptr<unsigned char>() : nullptr,           \
+      reinterpret_cast<__hip_bfloat16*>(ds.data_ptr()),                        \
+      reinterpret_cast<__hip_bfloat16*>(ad.data_ptr()),                        \
+      B, H, Lq, Lk, D, 0, (float)dropout_p)
+
+  if (act == 0) LAUNCH_BWD_DS(false);
+  else LAUNCH_BWD_DS(true);
+#undef LAUNCH_BWD_DS
+
+  // dQ = scale * dS @ K ; dK = scale * dS^T @ Q ; dV = A_d^T @ dO
+  auto ds3 = ds.reshape({B * H, Lq, Lk});
+  auto ad3 = ad.reshape({B * H, Lq, Lk});
+  auto k3 = k.reshape({B * H, Lk, D});
+  auto q3 = q.reshape({B * H, Lq, D});
+  auto do3 = dout.reshape({B * H, Lq, D});
+  auto dss = (scale == 1.0) ? ds3 : ds3 * scale;
+  auto dq = torch::bmm(dss, k3).reshape_as(q);
+  auto dk = torch::bmm(dss.transpose(1, 2), q3).reshape_as(k);
+  auto dv = torch::bmm(ad3.transpose(1, 2), do3).reshape_as(v);
+
+  torch::Tensor dbias;
+  if (bias_grad) {
+    dbias = (bias_dim == 3) ? ds.to(torch::kFloat32).sum(0)
+                            : ds.to(torch::kFloat32);
+  } else {
+    dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
+  }
+  return {dq, dk, dv, dbias};
+}
+
+}  // namespace genrec

@@ -16,6 +16,7 @@ Reference semantics fused here:
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -64,11 +65,15 @@ class _FusedAttnFn(torch.autograd.Function):
                 if not torch.cuda.is_current_stream_capturing() else 12345
             seed_dev = _seed_counter(q.device)
             seed_dev.add_(1)
-        out, probs, dmask = ops.ext().attn_fwd(
+        use_mfma = (q.dtype == torch.bfloat16 and q.size(3) % 32 == 0
+                    and os.environ.get("GENREC_DISABLE_MFMA", "0") != "1")
+        fwd = ops.ext().attn_fwd_mfma if use_mfma else ops.ext().attn_fwd
+        out, probs, dmask = fwd(
             q, k, v, bias, key_pad_mask, additive_mask, query_mask,
             scale, causal, act, dropout_p if training else 0.0, seed,
             seed_dev,
         )
+        ctx.use_mfma = use_mfma
         ctx.save_for_backward(q, k, v, probs, dmask,
                               query_mask if query_mask is not None else torch.empty(0))
         ctx.meta = (scale, causal, act, dropout_p if training else 0.0, seed,
@@ -82,7 +87,8 @@ class _FusedAttnFn(torch.autograd.Function):
 
         q, k, v, probs, dmask, query_mask = ctx.saved_tensors
         scale, causal, act, dropout_p, seed, bias_grad, bias_dim = ctx.meta
-        dq, dk, dv, dbias = ops.ext().attn_bwd(
+        bwd = ops.ext().attn_bwd_mfma if ctx.use_mfma else ops.ext().attn_bwd
+        dq, dk, dv, dbias = bwd(
             dout.contiguous(), q, k, v, probs, dmask,
             query_mask if query_mask.numel() else None,
             scale, act, dropout_p, seed, bias_grad, bias_dim,
@@ -108,8 +114,6 @@ def fused_attention(
     dropout_p: float = 0.0,
     training: bool = False,
 ) -> Tensor:
-    import os
-
     act = _ACT_SOFTMAX if score_act == "softmax" else _ACT_SILU
     fits = (k.size(2) <= 64 and q.size(2) <= 64 and q.size(3) <= 64
             and (additive_mask is None or additive_mask.dim() == 2)

@@ -19,6 +19,7 @@ sources = [
     "csrc/bindings.cpp",
     "csrc/kernels/norms.hip",
     "csrc/kernels/attention.hip",
+    "csrc/kernels/attention_mfma.hip",
     "csrc/kernels/ce.hip",
     "csrc/kernels/quantize.hip",
     "csrc/kernels/metrics.hip",
