@@ -255,20 +255,30 @@ template <bool IS_SILU>
 __global__ void __launch_bounds__(256)
 attn_bwd_ds_kernel(
     const __hip_bfloat16* __restrict__ dout,  // [B,H,Lq,D]
+    const __hip_bfloat16* __restrict__ q,     // [B,H,Lq,D]
+    const __hip_bfloat16* __restrict__ k,     // [B,H,Lk,D]
     const __hip_bfloat16* __restrict__ v,     // [B,H,Lk,D]
     const float* __restrict__ p_saved,        // [B,H,Lq,Lk] (P or S)
     const float* __restrict__ query_mask,     // null | [B,Lq]
     const unsigned char* __restrict__ drop_mask,
-    __hip_bfloat16* __restrict__ ds_out,      // [B,H,Lq,Lk]
-    __hip_bfloat16* __restrict__ ad_out,      // [B,H,Lq,Lk]
+    __hip_bfloat16* __restrict__ dq_out,      // [B,H,Lq,D]
+    __hip_bfloat16* __restrict__ dk_out,      // [B,H,Lk,D]
+    __hip_bfloat16* __restrict__ dv_out,      // [B,H,Lk,D]
+    float* __restrict__ ds_saved,             // null | [B,H,Lq,Lk] bias grad
     int B, int H, int Lq, int Lk, int D,
-    int act_silu_unused, float dropout_p) {
+    float scale, float dropout_p) {
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh % H;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* dos = smem;               // [64][128B] dO swizzled
-  char* vs = dos + TILE * 128;    // [64][128B] V rows
+  char* dos = smem;                 // [64][128B] dO rows (A for dP)
+  char* vs = dos + TILE * 128;      // [64][128B] V rows (B for dP)
+  char* kt = vs + TILE * 128;       // [64(d)][128B(j)] K^T (B for dQ)
+  char* qt = kt + TILE * 128;       // [64(d)][128B(i)] Q^T (B for dK)
+  char* dot = qt + TILE * 128;      // [64(d)][128B(i)] dO^T (B for dV)
+  char* dsn = dot + TILE * 128;     // [64(i)][128B(j)] dS (A for dQ)
+  char* dst = dsn + TILE * 128;     // [64(j)][128B(i)] dS^T (A for dK)
+  char* adt = dst + TILE * 128;     // [64(j)][128B(i)] A_d^T (A for dV)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -289,6 +299,23 @@ attn_bwd_ds_kernel(
           &v[IDX4M(b, h, row, d0, H, Lk, D)]);
     }
     *reinterpret_cast<short8v*>(vs + swz(row, d0 * 2)) = val2;
+    // transposed images: row = d, cols = sequence positions
+    __hip_bfloat16 tk[8], tq[8], td[8];
+    for (int jj = 0; jj < 8; ++jj) {
+      int p = d0 + jj;
+      tk[jj] = (p < Lk && row < D)
+          ? k[IDX4M(b, h, p, row, H, Lk, D)] : __hip_bfloat16(0.f);
+      tq[jj] = (p < Lq && row < D)
+          ? q[IDX4M(b, h, p, row, H, Lq, D)] : __hip_bfloat16(0.f);
+      td[jj] = (p < Lq && row < D)
+          ? dout[IDX4M(b, h, p, row, H, Lq, D)] : __hip_bfloat16(0.f);
+    }
+    *reinterpret_cast<short8v*>(kt + swz(row, d0 * 2)) =
+        *reinterpret_cast<short8v*>(tk);
+    *reinterpret_cast<short8v*>(qt + swz(row, d0 * 2)) =
+        *reinterpret_cast<short8v*>(tq);
+    *reinterpret_cast<short8v*>(dot + swz(row, d0 * 2)) =
+        *reinterpret_cast<short8v*>(td);
   }
   __syncthreads();
 
@@ -355,16 +382,88 @@ attn_bwd_ds_kernel(
     }
   }
 
+  // stash dS (natural + transposed) and A_d^T in LDS; optional global dS
 #pragma unroll
   for (int f = 0; f < 4; ++f) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int i = strip + row_grp + r;
       int j = f * 16 + col_base;
-      if (i < Lq && j < Lk) {
-        int64_t gi = IDX4M(b, h, i, j, H, Lq, Lk);
-        ds_out[gi] = __float2bfloat16(ds[f][r]);
-        ad_out[gi] = __float2bfloat16(ad[f][r]);
+      float dval = (i < Lq && j < Lk) ? ds[f][r] : 0.f;
+      float aval = (i < Lq && j < Lk) ? ad[f][r] : 0.f;
+      *reinterpret_cast<__hip_bfloat16*>(dsn + swz(i, j * 2)) =
+          __float2bfloat16(dval * scale);
+      *reinterpret_cast<__hip_bfloat16*>(dst + swz(j, i * 2)) =
+          __float2bfloat16(dval * scale);
+      *reinterpret_cast<__hip_bfloat16*>(adt + swz(j, i * 2)) =
+          __float2bfloat16(aval);
+      if (ds_saved && i < Lq && j < Lk) {
+        ds_saved[IDX4M(b, h, i, j, H, Lq, Lk)] = dval;
+      }
+    }
+  }
+  // dQ uses only this wave's dS rows — no barrier needed yet
+  __builtin_amdgcn_wave_barrier();
+
+  const int nfrag_d = (D + 15) / 16;
+  {  // dQ[strip rows] = (scale*dS) @ K  : A = dsn strip, B = kt d-rows
+    float4v accq[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      if (f >= nfrag_d) break;
+      for (int kk = 0; kk < TILE; kk += 32) {
+        short8v a = frag_load(dsn, strip, kk, lane);
+        short8v bfr = frag_load(kt, f * 16, kk, lane);
+        accq[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, accq[f],
+                                                          0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      if (f >= nfrag_d) break;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int i = strip + row_grp + r;
+        int d = f * 16 + col_base;
+        if (i < Lq && d < D) {
+          dq_out[IDX4M(b, h, i, d, H, Lq, D)] = __float2bfloat16(accq[f][r]);
+        }
+      }
+    }
+  }
+  __syncthreads();  // dK/dV read other waves' dS^T / A_d^T columns
+
+  {  // dK[j strip] = (scale*dS)^T @ Q ; dV[j strip] = A_d^T @ dO
+    float4v acck[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    float4v accv[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      if (f >= nfrag_d) break;
+      for (int kk = 0; kk < TILE; kk += 32) {
+        short8v a1 = frag_load(dst, strip, kk, lane);
+        short8v b1 = frag_load(qt, f * 16, kk, lane);
+        acck[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acck[f],
+                                                          0, 0, 0);
+        short8v a2 = frag_load(adt, strip, kk, lane);
+        short8v b2 = frag_load(dot, f * 16, kk, lane);
+        accv[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, accv[f],
+                                                          0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      if (f >= nfrag_d) break;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int j = strip + row_grp + r;
+        int d = f * 16 + col_base;
+        if (j < Lk && d < D) {
+          dk_out[IDX4M(b, h, j, d, H, Lk, D)] = __float2bfloat16(acck[f][r]);
+          dv_out[IDX4M(b, h, j, d, H, Lk, D)] = __float2bfloat16(accv[f][r]);
+        }
       }
     }
   }
@@ -443,47 +542,45 @@ std::vector<torch::Tensor> attn_bwd_mfma(
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
   TORCH_CHECK(Lq <= TILE && Lk <= TILE && D % 32 == 0);
-  auto opts_bf = q.options();
-  auto ds = torch::empty({B, H, Lq, Lk}, opts_bf);
-  auto ad = torch::empty({B, H, Lq, Lk}, opts_bf);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  torch::Tensor ds_saved;
+  float* ds_ptr = nullptr;
+  if (bias_grad) {
+    ds_saved = torch::empty({B, H, Lq, Lk},
+                            q.options().dtype(torch::kFloat32));
+    ds_ptr = ds_saved.data_ptr<float>();
+  }
   torch::Tensor qm_f;
   if (query_mask.has_value())
     qm_f = query_mask->to(torch::kFloat32).contiguous();
   dim3 block(256);
   dim3 grid(B * H);
-  size_t smem = 2 * TILE * 128;
+  size_t smem = 8 * TILE * 128;
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_BWD_DS(SILU)                                                    \
   hipLaunchKernelGGL((attn_bwd_ds_kernel<SILU>), grid, block, smem, stream,    \
-      reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),                 \
+      reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),                \
+      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),                   \
+      reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),                   \
       reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),                   \
       p_saved.data_ptr<float>(),                                               \
       query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,               \
       dropout_p > 0 ? drop_mask.data_ptr<unsigned char>() : nullptr,           \
-      reinterpret_cast<__hip_bfloat16*>(ds.data_ptr()),                        \
-      reinterpret_cast<__hip_bfloat16*>(ad.data_ptr()),                        \
-      B, H, Lq, Lk, D, 0, (float)dropout_p)
+      reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),                        \
+      reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),                        \
+      reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()), ds_ptr,                \
+      B, H, Lq, Lk, D, (float)scale, (float)dropout_p)
 
   if (act == 0) LAUNCH_BWD_DS(false);
   else LAUNCH_BWD_DS(true);
 #undef LAUNCH_BWD_DS
 
-  // dQ = scale * dS @ K ; dK = scale * dS^T @ Q ; dV = A_d^T @ dO
-  auto ds3 = ds.reshape({B * H, Lq, Lk});
-  auto ad3 = ad.reshape({B * H, Lq, Lk});
-  auto k3 = k.reshape({B * H, Lk, D});
-  auto q3 = q.reshape({B * H, Lq, D});
-  auto do3 = dout.reshape({B * H, Lq, D});
-  auto dss = (scale == 1.0) ? ds3 : ds3 * scale;
-  auto dq = torch::bmm(dss, k3).reshape_as(q);
-  auto dk = torch::bmm(dss.transpose(1, 2), q3).reshape_as(k);
-  auto dv = torch::bmm(ad3.transpose(1, 2), do3).reshape_as(v);
-
   torch::Tensor dbias;
   if (bias_grad) {
-    dbias = (bias_dim == 3) ? ds.to(torch::kFloat32).sum(0)
-                            : ds.to(torch::kFloat32);
+    dbias = (bias_dim == 3) ? ds_saved.sum(0) : ds_saved;
   } else {
     dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
   }
