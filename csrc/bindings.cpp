@@ -52,6 +52,14 @@ std::vector<torch::Tensor> sqdist_argmin(torch::Tensor x,
 torch::Tensor embedding_fwd(torch::Tensor weight, torch::Tensor indices);
 torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor indices,
                             int64_t num_weights, int64_t padding_idx);
+std::vector<torch::Tensor> dropout_add_fwd(
+    torch::Tensor x, torch::Tensor residual, double p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev);
+std::vector<torch::Tensor> relu_dropout_fwd(
+    torch::Tensor x, double p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev);
+torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
+                               double p, bool relu);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
 
 }  // namespace genrec
@@ -70,5 +78,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sqdist_argmin", &genrec::sqdist_argmin, "L2 dist + argmin");
   m.def("embedding_fwd", &genrec::embedding_fwd, "embedding gather");
   m.def("embedding_bwd", &genrec::embedding_bwd, "embedding scatter-add bwd");
+  m.def("dropout_add_fwd", &genrec::dropout_add_fwd, "residual+dropout fwd");
+  m.def("relu_dropout_fwd", &genrec::relu_dropout_fwd, "dropout(relu) fwd");
+  m.def("dropout_fuse_bwd", &genrec::dropout_fuse_bwd, "fused dropout bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
 }

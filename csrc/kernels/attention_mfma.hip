@@ -54,7 +54,7 @@ attn_fwd_mfma_kernel(
     const __hip_bfloat16* __restrict__ q,   // [B,H,Lq,D]
     const __hip_bfloat16* __restrict__ k,   // [B,H,Lk,D]
     const __hip_bfloat16* __restrict__ v,   // [B,H,Lk,D]
-    const float* __restrict__ bias,         // null | [H,Lq,Lk] | [B,H,Lq,Lk]
+    const void* __restrict__ bias,          // null | [H,Lq,Lk] | [B,H,Lq,Lk]
     const bool* __restrict__ key_pad,       // null | [B,Lk]
     const float* __restrict__ add_mask,     // null | [Lq,Lk]
     const float* __restrict__ query_mask,   // null | [B,Lq]
@@ -63,9 +63,11 @@ attn_fwd_mfma_kernel(
     unsigned char* __restrict__ drop_mask,
     const unsigned int* __restrict__ seed_dev,
     int B, int H, int Lq, int Lk, int D,
-    float scale, int bias_dim, bool causal,
+    float scale, int bias_dim, bool bias_bf16, bool causal,
     float dropout_p, unsigned int seed, int q_tile) {
   if (seed_dev) seed += *seed_dev;
+  const float* bias_f = reinterpret_cast<const float*>(bias);
+  const __hip_bfloat16* bias_b = reinterpret_cast<const __hip_bfloat16*>(bias);
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh % H;
   const int q0 = blockIdx.y * q_tile;
@@ -143,8 +145,11 @@ attn_fwd_mfma_kernel(
       bool ok = (i < Lq) && (j < Lk);
       if (ok) {
         s *= scale;
-        if (bias_dim == 3) s += bias[((int64_t)h * Lq + i) * Lk + j];
-        else if (bias_dim == 4) s += bias[IDX4M(b, h, i, j, H, Lq, Lk)];
+        if (bias_dim) {
+          int64_t bi = (bias_dim == 3) ? ((int64_t)h * Lq + i) * Lk + j
+                                       : IDX4M(b, h, i, j, H, Lq, Lk);
+          s += bias_bf16 ? to_f32(bias_b[bi]) : bias_f[bi];
+        }
         if (causal && j > i) s = NEG_BIG_MF;
         if (key_pad && key_pad[(int64_t)b * Lk + j]) s = NEG_BIG_MF;
         if (add_mask) s += add_mask[(int64_t)i * Lk + j];
@@ -491,11 +496,15 @@ std::vector<torch::Tensor> attn_fwd_mfma(
   } else {
     dmask = torch::empty({0}, q.options().dtype(torch::kUInt8));
   }
-  torch::Tensor bias_f;
+  torch::Tensor bias_c;
   int bias_dim = 0;
+  bool bias_bf16 = false;
   if (bias.has_value()) {
-    bias_f = bias->to(torch::kFloat32).contiguous();
-    bias_dim = bias_f.dim();
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32 ||
+                bias->scalar_type() == torch::kBFloat16);
+    bias_c = bias->contiguous();
+    bias_dim = bias_c.dim();
+    bias_bf16 = bias_c.scalar_type() == torch::kBFloat16;
   }
   torch::Tensor am_f, qm_f;
   if (add_mask.has_value()) am_f = add_mask->to(torch::kFloat32).contiguous();
@@ -513,7 +522,7 @@ std::vector<torch::Tensor> attn_fwd_mfma(
       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),                   \
       reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),                   \
       reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),                   \
-      bias_dim ? bias_f.data_ptr<float>() : nullptr,                           \
+      bias_dim ? bias_c.data_ptr() : nullptr,                                  \
       key_pad.has_value() ? key_pad->data_ptr<bool>() : nullptr,               \
       add_mask.has_value() ? am_f.data_ptr<float>() : nullptr,                 \
       query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,               \
@@ -523,8 +532,8 @@ std::vector<torch::Tensor> attn_fwd_mfma(
       seed_dev.has_value()                                                     \
           ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())        \
           : nullptr,                                                           \
-      B, H, Lq, Lk, D, (float)scale, bias_dim, causal, (float)dropout_p,       \
-      (unsigned int)seed, q_tile)
+      B, H, Lq, Lk, D, (float)scale, bias_dim, bias_bf16, causal,              \
+      (float)dropout_p, (unsigned int)seed, q_tile)
 
   if (act == 0) LAUNCH_FWD_MFMA(false);
   else LAUNCH_FWD_MFMA(true);

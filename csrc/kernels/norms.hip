@@ -17,9 +17,9 @@ namespace genrec {
 
 // ---------------------------------------------------------------- RMSNorm
 
-template <typename T, typename OutT, bool T5_STYLE>
+template <typename T, typename WT, typename OutT, bool T5_STYLE>
 __global__ void rms_norm_fwd_kernel(const T* __restrict__ x,
-                                    const float* __restrict__ w_f32,
+                                    const WT* __restrict__ w,
                                     OutT* __restrict__ y,
                                     float* __restrict__ inv_rms,
                                     int64_t n_rows, int d, float eps,
@@ -40,7 +40,7 @@ __global__ void rms_norm_fwd_kernel(const T* __restrict__ x,
     OutT* yr = y + row * d;
     for (int j = lane; j < d; j += WAVE) {
       float xf = to_f32(xr[j]);
-      float wf = w_f32[j];
+      float wf = to_f32(w[j]);
       float out;
       if (T5_STYLE) {
         // t = x * r (promote), optional cast to weight dtype, then w * t
@@ -63,10 +63,10 @@ __global__ void rms_norm_fwd_kernel(const T* __restrict__ x,
 // eager autograd does for the same graph up to bf16 rounding)
 constexpr int RMS_MAX_COLS_PER_LANE = 16;  // supports d <= 1024
 
-template <typename T, typename OutT>
+template <typename T, typename WT, typename OutT>
 __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
                                     const T* __restrict__ x,
-                                    const float* __restrict__ w_f32,
+                                    const WT* __restrict__ w,
                                     const float* __restrict__ inv_rms,
                                     T* __restrict__ dx,
                                     float* __restrict__ dw,
@@ -88,7 +88,7 @@ __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
     float r = inv_rms[row];
     float dot = 0.f;
     for (int j = lane; j < d; j += WAVE) {
-      dot += w_f32[j] * to_f32(dyr[j]) * to_f32(xr[j]);
+      dot += to_f32(w[j]) * to_f32(dyr[j]) * to_f32(xr[j]);
     }
     dot = wave_sum(dot);
     float c = r * r * r / d * dot;
@@ -97,7 +97,7 @@ __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
     for (int j = lane; j < d; j += WAVE, ++ci) {
       float xf = to_f32(xr[j]);
       float dyf = to_f32(dyr[j]);
-      dxr[j] = from_f32<T>(w_f32[j] * dyf * r - c * xf);
+      dxr[j] = from_f32<T>(to_f32(w[j]) * dyf * r - c * xf);
       dw_acc[ci] += dyf * xf * r;
     }
   }
@@ -192,9 +192,10 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
   const int d = x.size(-1);
   const int64_t n_rows = x.numel() / d;
-  auto w_f32 = w.to(torch::kFloat32).contiguous();
-  bool w_half = w.scalar_type() == torch::kBFloat16 ||
-                w.scalar_type() == torch::kHalf;
+  auto wc = w.contiguous();
+  bool w_bf16 = w.scalar_type() == torch::kBFloat16;
+  bool w_half = w_bf16 || w.scalar_type() == torch::kHalf;
+  TORCH_CHECK(w_bf16 || w.scalar_type() == torch::kFloat32);
   // output dtype follows eager semantics: promote(x.dtype, w.dtype)
   auto out_dtype = at::result_type(x, w);
   auto y = torch::empty(x.sizes(), x.options().dtype(out_dtype));
@@ -203,27 +204,34 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
   dim3 grid(grid_for_rows(n_rows, 4));
   auto stream = at::cuda::getCurrentHIPStream();
 
-#define LAUNCH_RMS(T, OutT, T5)                                                \
-  hipLaunchKernelGGL((rms_norm_fwd_kernel<T, OutT, T5>), grid, block, 0,       \
+#define LAUNCH_RMS(T, WT, OutT, T5)                                            \
+  hipLaunchKernelGGL((rms_norm_fwd_kernel<T, WT, OutT, T5>), grid, block, 0,   \
                      stream, reinterpret_cast<const T*>(x.data_ptr()),         \
-                     w_f32.data_ptr<float>(),                                  \
+                     reinterpret_cast<const WT*>(wc.data_ptr()),               \
                      reinterpret_cast<OutT*>(y.data_ptr()),                    \
                      inv_rms.data_ptr<float>(), n_rows, d, (float)eps, w_half)
 
+#define LAUNCH_RMS_W(T, OutT, T5)                                              \
+  do {                                                                         \
+    if (w_bf16) LAUNCH_RMS(T, __hip_bfloat16, OutT, T5);                       \
+    else LAUNCH_RMS(T, float, OutT, T5);                                       \
+  } while (0)
+
   if (x.scalar_type() == torch::kFloat32) {
-    if (t5_style) LAUNCH_RMS(float, float, true);
-    else LAUNCH_RMS(float, float, false);
+    if (t5_style) LAUNCH_RMS_W(float, float, true);
+    else LAUNCH_RMS_W(float, float, false);
   } else if (x.scalar_type() == torch::kBFloat16) {
     if (out_dtype == torch::kFloat32) {
-      if (t5_style) LAUNCH_RMS(__hip_bfloat16, float, true);
-      else LAUNCH_RMS(__hip_bfloat16, float, false);
+      if (t5_style) LAUNCH_RMS_W(__hip_bfloat16, float, true);
+      else LAUNCH_RMS_W(__hip_bfloat16, float, false);
     } else {
-      if (t5_style) LAUNCH_RMS(__hip_bfloat16, __hip_bfloat16, true);
-      else LAUNCH_RMS(__hip_bfloat16, __hip_bfloat16, false);
+      if (t5_style) LAUNCH_RMS_W(__hip_bfloat16, __hip_bfloat16, true);
+      else LAUNCH_RMS_W(__hip_bfloat16, __hip_bfloat16, false);
     }
   } else {
     TORCH_CHECK(false, "rms_norm: unsupported dtype");
   }
+#undef LAUNCH_RMS_W
 #undef LAUNCH_RMS
   return {y, inv_rms};
 }
@@ -234,8 +242,9 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   (void)t5_style;  // same analytic gradient for both styles
   const int d = x.size(-1);
   const int64_t n_rows = x.numel() / d;
-  auto w_f32 = w.to(torch::kFloat32).contiguous();
   TORCH_CHECK(d <= WAVE * 16, "rms_norm_bwd: d too large");
+  auto wc = w.contiguous();
+  bool w_bf16 = w.scalar_type() == torch::kBFloat16;
   auto dx = torch::empty_like(x);
   dim3 block(256);
   int n_blocks = std::min(grid_for_rows(n_rows, 4), 1024);
@@ -245,22 +254,31 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   size_t smem = 4 * (size_t)d * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
 
-#define LAUNCH_RMSB(T, OutT)                                                   \
-  hipLaunchKernelGGL((rms_norm_bwd_kernel<T, OutT>), grid, block, smem,        \
+#define LAUNCH_RMSB(T, WT, OutT)                                               \
+  hipLaunchKernelGGL((rms_norm_bwd_kernel<T, WT, OutT>), grid, block, smem,    \
                      stream, reinterpret_cast<const OutT*>(dy.data_ptr()),     \
                      reinterpret_cast<const T*>(x.data_ptr()),                 \
-                     w_f32.data_ptr<float>(), inv_rms.data_ptr<float>(),       \
+                     reinterpret_cast<const WT*>(wc.data_ptr()),               \
+                     inv_rms.data_ptr<float>(),                                \
                      reinterpret_cast<T*>(dx.data_ptr()),                      \
                      dw.data_ptr<float>(), n_rows, d)
 
+#define LAUNCH_RMSB_W(T, OutT)                                                 \
+  do {                                                                         \
+    if (w_bf16) LAUNCH_RMSB(T, __hip_bfloat16, OutT);                          \
+    else LAUNCH_RMSB(T, float, OutT);                                          \
+  } while (0)
+
   if (x.scalar_type() == torch::kFloat32) {
-    LAUNCH_RMSB(float, float);
+    LAUNCH_RMSB_W(float, float);
   } else if (x.scalar_type() == torch::kBFloat16) {
-    if (dy.scalar_type() == torch::kFloat32) LAUNCH_RMSB(__hip_bfloat16, float);
-    else LAUNCH_RMSB(__hip_bfloat16, __hip_bfloat16);
+    if (dy.scalar_type() == torch::kFloat32)
+      LAUNCH_RMSB_W(__hip_bfloat16, float);
+    else LAUNCH_RMSB_W(__hip_bfloat16, __hip_bfloat16);
   } else {
     TORCH_CHECK(false, "rms_norm_bwd: unsupported dtype");
   }
+#undef LAUNCH_RMSB_W
 #undef LAUNCH_RMSB
   auto dw_out = dw.sum(0).to(w.scalar_type());
   return {dx, dw_out};

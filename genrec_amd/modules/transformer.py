@@ -157,7 +157,8 @@ class FeedForward(nn.Module):
         self.dropout = nn.Dropout(dropout)
 
     def forward(self, x: Tensor) -> Tensor:
-        return self.wo(self.dropout(F.relu(self.wi(x))))
+        return self.wo(ops.relu_dropout(self.wi(x), self.dropout.p,
+                                        self.training))
 
 
 class TransformerBlock(nn.Module):
@@ -190,16 +191,17 @@ class TransformerBlock(nn.Module):
         if kv_cache is not None:
             self_cache = kv_cache.setdefault("self", {})
             cross_cache = kv_cache.setdefault("cross", {})
-        x = x + self.dropout1(self.self_attn(
+        x = ops.dropout_add(self.self_attn(
             self.norm1(x), attn_mask=attn_mask,
             key_padding_mask=key_padding_mask, kv_cache=self_cache,
-        ))
+        ), x, self.dropout1.p, self.training)
         if self.cross_attn_enabled and context is not None:
-            x = x + self.dropout_cross(self.cross_attn(
+            x = ops.dropout_add(self.cross_attn(
                 self.norm_cross(x), key=context, value=context,
                 key_padding_mask=memory_key_padding_mask, kv_cache=cross_cache,
-            ))
-        x = x + self.dropout2(self.ff(self.norm2(x)))
+            ), x, self.dropout_cross.p, self.training)
+        x = ops.dropout_add(self.ff(self.norm2(x)), x, self.dropout2.p,
+                            self.training)
         return x
 
 

@@ -85,6 +85,7 @@ from genrec_amd.ops.quantize import residual_quantize_step  # noqa: E402
 from genrec_amd.ops.losses import tied_softmax_ce, summed_ce  # noqa: E402
 from genrec_amd.ops.metrics import topk_hit_ranks  # noqa: E402
 from genrec_amd.ops.embedding import embedding  # noqa: E402
+from genrec_amd.ops.fused import dropout_add, relu_dropout  # noqa: E402
 
 __all__ = [
     "ext",
@@ -102,4 +103,6 @@ __all__ = [
     "summed_ce",
     "topk_hit_ranks",
     "embedding",
+    "dropout_add",
+    "relu_dropout",
 ]

@@ -78,7 +78,8 @@ class _FusedAttnFn(torch.autograd.Function):
                               query_mask if query_mask is not None else torch.empty(0))
         ctx.meta = (scale, causal, act, dropout_p if training else 0.0, seed,
                     bias is not None and bias.requires_grad,
-                    bias.dim() if bias is not None else 0)
+                    bias.dim() if bias is not None else 0,
+                    bias.dtype if bias is not None else None)
         return out
 
     @staticmethod
@@ -86,15 +87,16 @@ class _FusedAttnFn(torch.autograd.Function):
         from genrec_amd import ops
 
         q, k, v, probs, dmask, query_mask = ctx.saved_tensors
-        scale, causal, act, dropout_p, seed, bias_grad, bias_dim = ctx.meta
+        (scale, causal, act, dropout_p, seed, bias_grad, bias_dim,
+         bias_dtype) = ctx.meta
         bwd = ops.ext().attn_bwd_mfma if ctx.use_mfma else ops.ext().attn_bwd
         dq, dk, dv, dbias = bwd(
             dout.contiguous(), q, k, v, probs, dmask,
             query_mask if query_mask.numel() else None,
             scale, act, dropout_p, seed, bias_grad, bias_dim,
         )
-        if bias_grad and bias_dim == 3:
-            dbias = dbias.to(q.dtype) if q.dtype != torch.float32 else dbias
+        if bias_grad and dbias.dtype != bias_dtype:
+            dbias = dbias.to(bias_dtype)
         return (dq, dk, dv, dbias if bias_grad else None,
                 None, None, None, None, None, None, None, None)
 

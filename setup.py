@@ -24,6 +24,7 @@ sources = [
     "csrc/kernels/quantize.hip",
     "csrc/kernels/metrics.hip",
     "csrc/kernels/embedding.hip",
+    "csrc/kernels/fused_elementwise.hip",
 ]
 
 setup(

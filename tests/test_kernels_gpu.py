@@ -416,3 +416,32 @@ def test_mfma_attention_small_shapes():
         assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2), \
             (lq, lk, d, (out.float() - ref).abs().max())
         assert torch.isfinite(q.grad).all()
+
+
+def test_fused_dropout_add_and_relu_dropout():
+    from genrec_amd.ops.fused import dropout_add, relu_dropout
+
+    torch.manual_seed(11)
+    x = torch.randn(4096, 384, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    res = torch.randn_like(x, requires_grad=True)
+    out = dropout_add(x, res, 0.5, training=True)
+    # kept elements = res + 2x; dropped = res
+    diff = (out - res).float()
+    frac_zero = (diff == 0).float().mean().item()
+    assert 0.4 < frac_zero < 0.6
+    kept = diff != 0
+    assert torch.allclose(diff[kept], 2.0 * x.float()[kept], atol=2e-2,
+                          rtol=2e-2)
+    out.sum().backward()
+    assert torch.allclose(res.grad.float(), torch.ones_like(res).float())
+    # dx = 2*mask -> mean ~1
+    assert abs(x.grad.float().mean().item() - 1.0) < 0.05
+
+    y = torch.randn(4096, 1024, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    o2 = relu_dropout(y, 0.3, training=True)
+    assert (o2.float() >= 0).all()
+    o2.sum().backward()
+    g = y.grad.float()
+    assert ((g == 0) | (g > 1.0)).all()  # 0 or 1/(1-p)
