@@ -124,7 +124,8 @@ class SASRec(nn.Module):
                 targets: Optional[Tensor] = None
                 ) -> Tuple[Tensor, Optional[Tensor]]:
         b, l = input_ids.shape
-        mask = (input_ids != 0).unsqueeze(-1).float()
+        mask = (input_ids != 0).unsqueeze(-1) \
+            .to(self.item_embedding.weight.dtype)
         x = ops.embedding(self.item_embedding.weight, input_ids,
                           padding_idx=0) * (self.embed_dim ** 0.5)
         pos = torch.arange(l, device=input_ids.device).unsqueeze(0).expand(b, l)
