@@ -40,6 +40,18 @@ std::vector<torch::Tensor> attn_bwd_mfma(
     double scale, int64_t act, double dropout_p, int64_t seed,
     bool bias_grad, int64_t bias_dim);
 
+std::vector<torch::Tensor> hstu_attn_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor pos_bucket, torch::Tensor pos_table,
+    c10::optional<torch::Tensor> time_table,
+    c10::optional<torch::Tensor> timestamps,
+    c10::optional<torch::Tensor> key_pad);
+std::vector<torch::Tensor> hstu_attn_bwd(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor s_saved, torch::Tensor pos_bucket,
+    c10::optional<torch::Tensor> timestamps,
+    int64_t n_pos, int64_t n_time);
+
 std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
                                           torch::Tensor targets,
                                           int64_t ignore_index);
@@ -73,6 +85,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &genrec::attn_bwd, "fused attention backward");
   m.def("attn_fwd_mfma", &genrec::attn_fwd_mfma, "MFMA attention forward");
   m.def("attn_bwd_mfma", &genrec::attn_bwd_mfma, "MFMA attention backward");
+  m.def("hstu_attn_fwd", &genrec::hstu_attn_fwd, "HSTU fused attention fwd");
+  m.def("hstu_attn_bwd", &genrec::hstu_attn_bwd, "HSTU fused attention bwd");
   m.def("softmax_ce_fwd", &genrec::softmax_ce_fwd, "fused CE forward");
   m.def("softmax_ce_bwd", &genrec::softmax_ce_bwd, "fused CE backward");
   m.def("sqdist_argmin", &genrec::sqdist_argmin, "L2 dist + argmin");

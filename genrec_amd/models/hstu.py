@@ -52,13 +52,17 @@ class RelativePositionBias(nn.Module):
         large = torch.clamp(large, max=self.num_buckets - 1)
         return torch.where(is_small, rel, large)
 
-    def forward(self, seq_len: int, device) -> Tensor:
+    def bucket_table(self, seq_len: int, device) -> Tensor:
         key = (seq_len, str(device))
         buckets = self._bucket_cache.get(key)
         if buckets is None:
             pos = torch.arange(seq_len, device=device)
             buckets = self._bucket(pos.unsqueeze(0) - pos.unsqueeze(1))
             self._bucket_cache[key] = buckets
+        return buckets
+
+    def forward(self, seq_len: int, device) -> Tensor:
+        buckets = self.bucket_table(seq_len, device)
         bias = ops.embedding(self.relative_attention_bias.weight, buckets)
         return bias.permute(2, 0, 1)  # [H, L, L]
 
@@ -115,12 +119,20 @@ class HSTULayer(nn.Module):
             return t.view(b, l, self.num_heads, self.head_dim).transpose(1, 2)
 
         q, k, v = split(q), split(k), split(v)
-        pos_bias = self.position_bias(l, x.device)  # [H, L, L]
-        time_bias = None
-        if self.use_temporal_bias and timestamps is not None:
-            time_bias = self.temporal_bias(timestamps)  # [B, H, L, L]
-        attn = ops.hstu_pointwise_attention(q, k, v, pos_bias, time_bias,
-                                            padding_mask)
+        use_time = self.use_temporal_bias and timestamps is not None
+        # fully-fused path: bias bucketing + gathers + SiLU attention in
+        # one MFMA kernel (falls through to the bias-tensor composition)
+        attn = ops.hstu_fused_attention(
+            q, k, v, self.position_bias.bucket_table(l, x.device),
+            self.position_bias.relative_attention_bias.weight,
+            self.temporal_bias.temporal_attention_bias.weight
+            if use_time else None,
+            timestamps if use_time else None, padding_mask)
+        if attn is None:
+            pos_bias = self.position_bias(l, x.device)  # [H, L, L]
+            time_bias = self.temporal_bias(timestamps) if use_time else None
+            attn = ops.hstu_pointwise_attention(q, k, v, pos_bias, time_bias,
+                                                padding_mask)
         attn = attn.transpose(1, 2).reshape(b, l, d)
         attn = self.attn_norm(attn) * u
         x = residual + self.dropout(attn)

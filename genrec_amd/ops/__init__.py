@@ -80,6 +80,7 @@ from genrec_amd.ops.attention import (  # noqa: E402
     sasrec_attention,
     t5_attention,
     hstu_pointwise_attention,
+    hstu_fused_attention,
 )
 from genrec_amd.ops.quantize import residual_quantize_step  # noqa: E402
 from genrec_amd.ops.losses import tied_softmax_ce, summed_ce  # noqa: E402
@@ -98,6 +99,7 @@ __all__ = [
     "sasrec_attention",
     "t5_attention",
     "hstu_pointwise_attention",
+    "hstu_fused_attention",
     "residual_quantize_step",
     "tied_softmax_ce",
     "summed_ce",

@@ -163,3 +163,59 @@ def hstu_pointwise_attention(q, k, v, pos_bias, time_bias, key_pad_mask):
         q, k, v, scale=1.0, bias=bias, key_pad_mask=key_pad_mask,
         causal=True, score_act="silu",
     )
+
+
+class _HstuAttnFn(torch.autograd.Function):
+    """Fully-fused HSTU attention: bias bucketing/gather + SiLU scores +
+    table gradients all in-kernel (K4+K5+K6)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, pos_weight, time_weight, pos_bucket,
+                timestamps, key_pad):
+        from genrec_amd import ops
+
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        out, s_saved = ops.ext().hstu_attn_fwd(
+            q, k, v, pos_bucket,
+            pos_weight.contiguous(),
+            time_weight.contiguous() if time_weight is not None else None,
+            timestamps, key_pad)
+        ctx.save_for_backward(q, k, v, s_saved, pos_bucket,
+                              timestamps if timestamps is not None
+                              else torch.empty(0))
+        ctx.n_pos = pos_weight.size(0)
+        ctx.n_time = time_weight.size(0) if time_weight is not None else 0
+        ctx.dtypes = (pos_weight.dtype,
+                      time_weight.dtype if time_weight is not None else None)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from genrec_amd import ops
+
+        q, k, v, s_saved, pos_bucket, ts = ctx.saved_tensors
+        dq, dk, dv, dpos, dtime = ops.ext().hstu_attn_bwd(
+            dout.contiguous(), q, k, v, s_saved, pos_bucket,
+            ts if ts.numel() else None, ctx.n_pos, ctx.n_time)
+        dpos = dpos.to(ctx.dtypes[0])
+        dtime_out = dtime.to(ctx.dtypes[1]) if ctx.n_time else None
+        return dq, dk, dv, dpos, dtime_out, None, None, None
+
+
+def hstu_fused_attention(q, k, v, pos_bucket, pos_weight, time_weight,
+                         timestamps, key_pad_mask):
+    """Dispatch for the fully-fused HSTU kernel; falls back to the bias-
+    tensor composition (hstu_pointwise_attention) off-GPU / odd shapes."""
+    from genrec_amd import ops
+
+    fits = (q.dtype == torch.bfloat16 and q.size(2) <= 64
+            and q.size(3) % 32 == 0 and q.size(3) <= 64
+            and pos_weight.dtype == torch.bfloat16
+            and (time_weight is None or time_weight.dtype == torch.bfloat16)
+            and os.environ.get("GENREC_DISABLE_MFMA", "0") != "1")
+    if fits and _kernel_available("hstu_attn_fwd", q, k, v):
+        ts = timestamps.contiguous() if timestamps is not None else None
+        kp = key_pad_mask.contiguous() if key_pad_mask is not None else None
+        return _HstuAttnFn.apply(q, k, v, pos_weight, time_weight,
+                                 pos_bucket.contiguous(), ts, kp)
+    return None  # caller composes the bias-tensor path

@@ -20,6 +20,7 @@ sources = [
     "csrc/kernels/norms.hip",
     "csrc/kernels/attention.hip",
     "csrc/kernels/attention_mfma.hip",
+    "csrc/kernels/hstu_attn.hip",
     "csrc/kernels/ce.hip",
     "csrc/kernels/quantize.hip",
     "csrc/kernels/metrics.hip",

@@ -445,3 +445,84 @@ def test_fused_dropout_add_and_relu_dropout():
     o2.sum().backward()
     g = y.grad.float()
     assert ((g == 0) | (g > 1.0)).all()  # 0 or 1/(1-p)
+
+
+def test_hstu_fused_model_vs_fp32():
+    """Full HSTU model on the fused bias+SiLU MFMA kernel vs fp32 eager."""
+    from genrec_amd.models.hstu import HSTU
+
+    torch.manual_seed(21)
+    m = HSTU(num_items=500, max_seq_len=50, embed_dim=64, num_heads=2,
+             num_blocks=2, dropout=0.0, use_temporal_bias=True)
+    m.eval()
+    ids = torch.randint(1, 501, (4, 50))
+    ids[0, :10] = 0
+    ts = (torch.arange(50) * 86400 + 10 ** 9).unsqueeze(0).expand(4, -1) \
+        .contiguous()
+    with torch.no_grad():
+        logits_cpu, _ = m(ids, ts, ids)
+    mg = m.to(DEV).to(torch.bfloat16)
+    mg.train()
+    idg, tsg = ids.to(DEV), ts.to(DEV)
+    logits_gpu, loss = mg(idg, tsg, idg)
+    # training path returns loss only; eval for logits
+    mg.eval()
+    with torch.no_grad():
+        logits_gpu, _ = mg(idg, tsg, idg)
+    assert torch.allclose(logits_gpu.float().cpu(), logits_cpu, atol=0.5,
+                          rtol=0.1), (logits_gpu.float().cpu()
+                                      - logits_cpu).abs().max()
+    # gradient flow incl. both bias tables
+    mg.train()
+    _, loss = mg(idg, tsg, idg)
+    loss.backward()
+    for layer in mg.layers:
+        g1 = layer.position_bias.relative_attention_bias.weight.grad
+        g2 = layer.temporal_bias.temporal_attention_bias.weight.grad
+        assert g1 is not None and torch.isfinite(g1).all() and g1.abs().sum() > 0
+        assert g2 is not None and torch.isfinite(g2).all() and g2.abs().sum() > 0
+
+
+def test_hstu_fused_grads_vs_composed():
+    """Fused kernel grads vs the bias-tensor composition (same bf16 path)."""
+    import os
+
+    from genrec_amd.models.hstu import HSTULayer
+
+    torch.manual_seed(22)
+    layer = HSTULayer(embed_dim=64, num_heads=2, dropout=0.0,
+                      num_position_buckets=32, num_time_buckets=64,
+                      max_position_distance=128, use_temporal_bias=True) \
+        .to(DEV).to(torch.bfloat16)
+    x = torch.randn(4, 50, 64, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    ts = (torch.arange(50, device=DEV) * 3600 + 10 ** 9).unsqueeze(0) \
+        .expand(4, -1).contiguous()
+    pad = torch.zeros(4, 50, dtype=torch.bool, device=DEV)
+    pad[1, 40:] = True
+    layer.train()
+
+    out1 = layer(x, pad, ts)
+    out1.float().sum().backward()
+    g_fused = {n: p.grad.clone() for n, p in layer.named_parameters()
+               if p.grad is not None}
+    gx_fused = x.grad.clone()
+    for p in layer.parameters():
+        p.grad = None
+    x.grad = None
+
+    os.environ["GENREC_DISABLE_MFMA"] = "1"
+    try:
+        out2 = layer(x, pad, ts)
+        out2.float().sum().backward()
+    finally:
+        os.environ.pop("GENREC_DISABLE_MFMA")
+    assert torch.allclose(out1.float(), out2.float(), atol=0.5, rtol=0.1)
+    assert torch.allclose(gx_fused.float(), x.grad.float(), atol=0.5,
+                          rtol=0.1)
+    for n, p in layer.named_parameters():
+        if p.grad is None:
+            continue
+        assert torch.allclose(g_fused[n].float(), p.grad.float(), atol=0.5,
+                              rtol=0.1), (n, (g_fused[n].float()
+                                              - p.grad.float()).abs().max())
