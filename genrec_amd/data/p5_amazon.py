@@ -1,0 +1,225 @@
+"""P5-preprocessed Amazon pipeline (alternative to data/amazon.py).
+
+Parity target: /root/reference/genrec/data/p5_amazon.py (504 LoC). The
+reference builds on torch-geometric's InMemoryDataset + polars rolling
+windows over the P5 preprocessing zip; neither torch-geometric nor polars
+exists in this image, so this is a dependency-free re-implementation of
+the same behavior:
+
+  * inputs: the P5 preprocessing outputs placed under
+    `<root>/<split>/` — `sequential_data.txt` (lines: "user item1 item2
+    ...", ids already remapped dense), `datamaps.json` (id2item/item2id),
+    and `meta.json.gz` (python-literal metadata lines)
+  * item text = "title brand categories price" strings embedded with a
+    local sentence-T5 checkpoint when available, else the deterministic
+    hashed-char-3-gram features (same offline fallback as data/amazon.py)
+  * rolling-window user histories (window_size, stride) with an ordered
+    train/test split on the window end position (p5_amazon.py:83-123)
+  * P5AmazonReviewsItemDataset: item embedding rows for RQ-VAE
+    (p5_amazon.py:371-406)
+  * P5AmazonReviewsSeqDataset: RQ-VAE-tokenized windows with random-crop
+    subsampling for training (p5_amazon.py:410-500)
+"""
+
+from __future__ import annotations
+
+import gzip
+import json
+import os
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+from torch.utils.data import Dataset
+
+from genrec_amd.config import ginlite
+from genrec_amd.data.schemas import SeqData
+
+
+def _read_sequential(path: str) -> List[List[int]]:
+    seqs = []
+    with open(path) as f:
+        for line in f:
+            parts = line.split()
+            if len(parts) >= 3:
+                seqs.append([int(x) for x in parts[1:]])
+    return seqs
+
+
+def _read_meta(path: str) -> Dict[str, dict]:
+    import ast
+
+    out = {}
+    opener = gzip.open if path.endswith(".gz") else open
+    with opener(path, "rt", encoding="utf-8") as f:
+        for line in f:
+            line = line.strip()
+            if not line:
+                continue
+            try:
+                m = json.loads(line)
+            except json.JSONDecodeError:
+                try:
+                    m = ast.literal_eval(line)
+                except (ValueError, SyntaxError):
+                    continue
+            if m.get("asin"):
+                out[m["asin"]] = m
+    return out
+
+
+class AmazonReviews:
+    """Loader for a P5-preprocessed split (ref AmazonReviews,
+    p5_amazon.py:233-368), without torch-geometric."""
+
+    def __init__(self, root: str, split: str = "beauty",
+                 encoder_model_name: str = "light",
+                 embed_dim: int = 768) -> None:
+        self.root, self.split = root, split
+        base = os.path.join(root, split)
+        seq_path = os.path.join(base, "sequential_data.txt")
+        maps_path = os.path.join(base, "datamaps.json")
+        if not os.path.exists(seq_path):
+            raise FileNotFoundError(
+                f"P5 preprocessing outputs missing under {base} "
+                f"(sequential_data.txt / datamaps.json / meta.json.gz); "
+                f"no network in this environment — place them manually.")
+        self.sequences = _read_sequential(seq_path)
+        with open(maps_path) as f:
+            maps = json.load(f)
+        self.id2item: Dict[str, str] = maps.get("id2item", {})
+        n_items = max(int(i) for i in self.id2item) if self.id2item else \
+            max(max(s) for s in self.sequences)
+        self.num_items = n_items
+
+        meta_path = None
+        for cand in ("meta.json.gz", "meta.json"):
+            p = os.path.join(base, cand)
+            if os.path.exists(p):
+                meta_path = p
+                break
+        meta = _read_meta(meta_path) if meta_path else {}
+
+        texts = []
+        for i in range(1, n_items + 1):
+            asin = self.id2item.get(str(i), "")
+            m = meta.get(asin, {})
+            cats = m.get("categories") or m.get("category") or ""
+            texts.append(f"{m.get('title') or ''} {m.get('brand') or ''} "
+                         f"{cats} {m.get('price') or ''}".strip()
+                         or f"item {i}")
+        if encoder_model_name and encoder_model_name != "light" \
+                and os.path.exists(encoder_model_name):
+            from sentence_transformers import SentenceTransformer
+
+            model = SentenceTransformer(encoder_model_name)
+            self.item_embeddings = np.asarray(
+                model.encode(texts, batch_size=64), dtype=np.float32)
+        else:
+            from genrec_amd.data.amazon import _hashed_text_embeddings
+
+            self.item_embeddings = _hashed_text_embeddings(texts, embed_dim)
+
+    def rolling_windows(self, window_size: int = 20, stride: int = 1,
+                        train_split: float = 0.8):
+        """[(items_window, is_train)] with ordered split on window end
+        position within each user's sequence (ref p5_amazon.py:113-123)."""
+        out = []
+        for seq in self.sequences:
+            n = len(seq)
+            if n < 2:
+                continue
+            ends = list(range(2, n + 1, stride))
+            thresh = ends[int(np.ceil(train_split * len(ends))) - 1] \
+                if ends else n
+            for e in ends:
+                lo = max(0, e - window_size - 1)
+                out.append((seq[lo:e], e <= thresh))
+        return out
+
+
+@ginlite.configurable(name="P5AmazonReviewsItemDataset")
+class P5AmazonReviewsItemDataset(Dataset):
+    """Item embedding rows for RQ-VAE (ref p5_amazon.py:371-406)."""
+
+    def __init__(self, root: str = "dataset/p5_amazon",
+                 split: str = "beauty", train_test_split: str = "all",
+                 encoder_model_name: str = "light",
+                 embed_dim: int = 768) -> None:
+        data = AmazonReviews(root, split, encoder_model_name, embed_dim)
+        embs = data.item_embeddings
+        if train_test_split != "all":
+            gen = torch.Generator().manual_seed(42)
+            is_train = (torch.rand(len(embs), generator=gen) > 0.05).numpy()
+            embs = embs[is_train] if train_test_split == "train" \
+                else embs[~is_train]
+        self.embeddings = embs
+
+    def __len__(self) -> int:
+        return len(self.embeddings)
+
+    def __getitem__(self, idx: int):
+        return torch.tensor(self.embeddings[idx], dtype=torch.float32)
+
+
+@ginlite.configurable(name="P5AmazonReviewsSeqDataset")
+class P5AmazonReviewsSeqDataset(Dataset):
+    """RQ-VAE-tokenized rolling windows with random-crop subsampling
+    (ref p5_amazon.py:410-500)."""
+
+    def __init__(self, root: str = "dataset/p5_amazon",
+                 split: str = "beauty", train_test_split: str = "train",
+                 max_items_per_seq: int = 20, subsample: bool = True,
+                 pretrained_rqvae_path: str = "./out/rqvae/{split}/checkpoint_final.pt",
+                 encoder_model_name: str = "light",
+                 rqvae_input_dim: int = 768, rqvae_embed_dim: int = 32,
+                 rqvae_hidden_dims: List[int] = [512, 256, 128, 64],
+                 rqvae_codebook_size: int = 256, rqvae_n_layers: int = 3,
+                 seed: int = 0) -> None:
+        from genrec_amd.data.amazon import tokenize_items_with_rqvae
+
+        data = AmazonReviews(root, split, encoder_model_name,
+                             rqvae_input_dim)
+        embs = torch.tensor(data.item_embeddings, dtype=torch.float32)
+        # item ids in sequential_data are 1-based
+        self.sem_ids_list = tokenize_items_with_rqvae(
+            embs, pretrained_rqvae_path.format(split=split),
+            rqvae_input_dim=rqvae_input_dim, rqvae_embed_dim=rqvae_embed_dim,
+            rqvae_hidden_dims=rqvae_hidden_dims,
+            rqvae_codebook_size=rqvae_codebook_size,
+            rqvae_n_layers=rqvae_n_layers)
+        self.sem_id_dim = len(self.sem_ids_list[0])
+        self.subsample = subsample and train_test_split == "train"
+        self.max_items = max_items_per_seq
+        self._rng = np.random.default_rng(seed)
+
+        windows = data.rolling_windows(window_size=max_items_per_seq)
+        want_train = train_test_split == "train"
+        self.samples = [w for w, is_train in windows
+                        if is_train == want_train and len(w) >= 2]
+        if train_test_split == "test":
+            # last window per user only (leave-one-out on the full seq)
+            self.samples = []
+            for seq in data.sequences:
+                if len(seq) >= 2:
+                    lo = max(0, len(seq) - max_items_per_seq - 1)
+                    self.samples.append(seq[lo:])
+
+    def all_valid_sem_ids(self) -> torch.Tensor:
+        return torch.tensor(self.sem_ids_list, dtype=torch.long)
+
+    def __len__(self) -> int:
+        return len(self.samples)
+
+    def __getitem__(self, idx: int) -> SeqData:
+        w = self.samples[idx]
+        hist, target = w[:-1], w[-1]
+        if self.subsample and len(hist) > 2:
+            # random crop of the history (ref p5_amazon.py:469-500)
+            n = int(self._rng.integers(2, len(hist) + 1))
+            start = int(self._rng.integers(0, len(hist) - n + 1))
+            hist = hist[start:start + n]
+        hist = hist[-self.max_items:]
+        flat = [c for it in hist for c in self.sem_ids_list[it - 1]]
+        tgt = list(self.sem_ids_list[target - 1])
+        return SeqData(user_id=idx % 10000, item_ids=flat, target_ids=tgt)
