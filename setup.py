@@ -28,16 +28,21 @@ sources = [
     "csrc/kernels/fused_elementwise.hip",
 ]
 
+cxx_flags = ["-O3", "-std=c++17"]
+hip_flags = ["-O3", "-std=c++17"]
+if os.environ.get("GENREC_DEBUG_BUILD", "0") == "1":
+    # debug / sanitizer-friendly build: symbols, no
+    # aggressive opt, frame pointers for rocgdb
+    cxx_flags = ["-O1", "-g", "-std=c++17", "-fno-omit-frame-pointer"]
+    hip_flags = ["-O1", "-g", "-std=c++17", "-fno-omit-frame-pointer"]
+
 setup(
     name="genrec_amd_ext",
     ext_modules=[
         CUDAExtension(
             name="genrec_amd._C",
             sources=sources,
-            extra_compile_args={
-                "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17"],
-            },
+            extra_compile_args={"cxx": cxx_flags, "nvcc": hip_flags},
         )
     ],
     cmdclass={"build_ext": BuildExtension},
