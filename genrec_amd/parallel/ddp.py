@@ -145,8 +145,8 @@ class GradReducer:
         if self._pending[bi] == 0:
             self._launch(bi)
 
-    def _launch(self, bi: int) -> None:
-        ps = self._buckets[bi]
+    def _launch(self, bi: int, params=None) -> None:
+        ps = params if params is not None else self._buckets[bi]
         grads = [p.grad for p in ps]
         flat = torch._utils._flatten_dense_tensors(grads)
         if self._comm_stream is not None:
@@ -158,28 +158,32 @@ class GradReducer:
         else:
             work = dist.all_reduce(flat, group=self.group, async_op=True)
         self._flat[bi] = flat
-        self._works.append((bi, work))
+        self._works.append((bi, ps, work))
 
     def finalize(self) -> None:
-        """Wait for all in-flight reductions; write averaged grads back."""
+        """Wait for all in-flight reductions; write averaged grads back.
+
+        Buckets that never completed through the hooks (parameters the
+        forward did not touch, e.g. TIGER's unused positional embeddings)
+        are reduced here over the SUBSET of parameters that do have
+        grads — otherwise a mixed used/unused bucket would silently stay
+        unsynchronized.
+        """
         if self.world <= 1 or self.skip_sync:
             return
-        # launch any buckets whose grads never materialized as a group
         for bi in range(len(self._buckets)):
-            if self._pending[bi] > 0 and any(
-                    p.grad is not None for p in self._buckets[bi]):
-                grads_ready = all(
-                    p.grad is not None for p in self._buckets[bi])
-                if grads_ready:
-                    self._launch(bi)
-                    self._pending[bi] = 0
-        for bi, work in self._works:
+            if self._pending[bi] > 0:
+                with_grads = [p for p in self._buckets[bi]
+                              if p.grad is not None]
+                if with_grads:
+                    self._launch(bi, with_grads)
+                self._pending[bi] = 0
+        for bi, ps, work in self._works:
             work.wait()
         if self._comm_stream is not None:
             torch.cuda.current_stream().wait_stream(self._comm_stream)
         inv = 1.0 / self.world
-        for bi, _ in self._works:
-            ps = self._buckets[bi]
+        for bi, ps, _ in self._works:
             flat = self._flat[bi]
             flat.mul_(inv)
             for p, g in zip(ps, torch._utils._unflatten_dense_tensors(
