@@ -97,18 +97,71 @@ def extract_sem_ids(text: str, n_codebooks: int) -> Optional[List[int]]:
 
 
 @torch.no_grad()
+def evaluate_aux_tasks(model: LCRec, metas, device,
+                       helper: ConstrainedDecodingHelper, n_codebooks: int,
+                       max_samples: int = 64) -> Dict[str, float]:
+    """item2index greedy accuracy + index2item substring match
+    (ref lcrec_trainer.py:190-222)."""
+    from genrec_amd.data.lcrec_sft import TEMPLATES, sem_ids_to_tokens
+
+    allowed = helper.allowed_token_ids()
+    metas = metas[:max_samples]
+    out = {"item2index_correct": 0.0, "index2item_correct": 0.0,
+           "aux_total": float(len(metas))}
+    if not metas:
+        return out
+    # --- item2index: greedy constrained decode from the title prompt
+    prompts = [TEMPLATES["item2index"][0].format(
+        title=m.get("target_title", ""), description="") for m in metas]
+    batch = sft_collate([{"prompt": p, "response": ""} for p in prompts],
+                        model, for_generation=True)
+    ids = batch["input_ids"].to(device)
+    res = model.generate_topk(ids,
+                              attention_mask=batch["attention_mask"]
+                              .to(device),
+                              max_new_tokens=n_codebooks, beam_width=1,
+                              allowed_token_ids=allowed)
+    L = ids.size(1)
+    for m, row in zip(metas, res):
+        text = model.decode(row[0][0][L:], skip_special_tokens=False)
+        if extract_sem_ids(text, n_codebooks) == m["target_sem_ids"]:
+            out["item2index_correct"] += 1.0
+    # --- index2item: free generation, substring match of the title
+    prompts = [TEMPLATES["index2item"][0].format(
+        index=sem_ids_to_tokens(m["target_sem_ids"])) for m in metas]
+    batch = sft_collate([{"prompt": p, "response": ""} for p in prompts],
+                        model, for_generation=True)
+    ids = batch["input_ids"].to(device)
+    res = model.generate_topk(ids,
+                              attention_mask=batch["attention_mask"]
+                              .to(device),
+                              max_new_tokens=24, beam_width=1)
+    L = ids.size(1)
+    for m, row in zip(metas, res):
+        text = model.decode(row[0][0][L:], skip_special_tokens=True)
+        title = str(m.get("target_title", "")).strip().lower()
+        if title and (title in text.lower() or text.strip().lower() in title):
+            out["index2item_correct"] += 1.0
+    return out
+
+
+@torch.no_grad()
 def evaluate(model: LCRec, loader, device, helper: ConstrainedDecodingHelper,
              n_codebooks: int, ks=(1, 5, 10), beam_width: int = 10,
-             max_batches: Optional[int] = None) -> Dict[str, float]:
+             max_batches: Optional[int] = None,
+             aux_tasks: bool = True) -> Dict[str, float]:
     model.eval()
     acc = TopKAccumulator(ks=list(ks))
     allowed = helper.allowed_token_ids()
+    aux_metas = []
     for bi, batch in enumerate(loader):
         if max_batches is not None and bi >= max_batches:
             break
         ids = batch["input_ids"].to(device)
         attn = batch["attention_mask"].to(device)
         metas = batch["meta"]
+        if aux_tasks and len(aux_metas) < 64:
+            aux_metas.extend(metas)
         res = model.generate_topk(
             ids, attention_mask=attn, max_new_tokens=n_codebooks,
             beam_width=beam_width, allowed_token_ids=allowed)
@@ -126,7 +179,15 @@ def evaluate(model: LCRec, loader, device, helper: ConstrainedDecodingHelper,
             preds.append(beams)
         acc.accumulate(torch.tensor(tgts, device=device),
                        torch.tensor(preds, device=device))
-    return acc.reduce(all_reduce=True)
+    metrics = acc.reduce(all_reduce=True)
+    if aux_tasks:
+        aux = evaluate_aux_tasks(model, aux_metas, device, helper,
+                                 n_codebooks)
+        aux = reduce_scalars(aux, device)
+        t = max(aux.pop("aux_total"), 1.0)
+        metrics["item2index_acc"] = aux["item2index_correct"] / t
+        metrics["index2item_acc"] = aux["index2item_correct"] / t
+    return metrics
 
 
 @ginlite.configurable(name="train")
