@@ -28,7 +28,12 @@ class KmeansOutput(NamedTuple):
 
 
 class Kmeans:
-    def __init__(self, k: int, max_iters: Optional[int] = None,
+    """max_iters defaults to 200: the reference iterates unboundedly to a
+    1e-10 movement threshold (kmeans.py:84-97), which fp32 Lloyd's on GPU
+    can oscillate around forever — a hard hang observed on MI355X. 200
+    iterations is far past practical convergence for codebook init."""
+
+    def __init__(self, k: int, max_iters: Optional[int] = 200,
                  stop_threshold: float = 1e-10) -> None:
         self.k = k
         self.iters = max_iters
