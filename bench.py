@@ -103,9 +103,12 @@ def main() -> None:
 
     # hipGraph capture of the whole train step (fwd+bwd+allreduce+clip+AdamW)
     # removes per-launch host overhead — the step is ~1000 tiny dispatches.
+    # On capture failure (e.g. an RCCL build without graph support) the SAME
+    # bf16/master-weight step runs eagerly, so N=1 and N>1 numbers stay
+    # mode-comparable.
     use_graph = use_gpu and os.environ.get("GENREC_BENCH_GRAPH", "1") == "1"
 
-    if use_graph:
+    if use_gpu:
         # Pure-bf16 compute with fp32 master weights (standard bf16 mixed
         # precision a la Megatron): the model runs natively in bf16 — no
         # autocast, so no per-layer weight-cast kernels (~2900/step) — the
@@ -162,39 +165,48 @@ def main() -> None:
                 torch.cuda.synchronize()
                 print(f"# bench-debug: {msg}", flush=True)
 
-        try:
-            # warmup on a side stream (required before capture)
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
-                for _ in range(3):
-                    inner_step()
-            torch.cuda.current_stream().wait_stream(s)
-            dbg("warmup done")
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                static_loss = inner_step()
-            dbg("capture done")
-            graph.replay()
-            dbg("first replay done")
+        def eager_step(i: int) -> None:
+            b = batches[i % len(batches)]
+            for key in static:
+                static[key].copy_(b[key], non_blocking=True)
+            inner_step()
 
-            nocopy = os.environ.get("GENREC_BENCH_NOCOPY", "0") == "1"
-
-            def step(i: int) -> None:
-                if not nocopy:
-                    b = batches[i % len(batches)]
-                    for key in static:
-                        static[key].copy_(b[key], non_blocking=True)
+        step = eager_step
+        if use_graph:
+            try:
+                # warmup on a side stream (required before capture)
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    for _ in range(3):
+                        inner_step()
+                torch.cuda.current_stream().wait_stream(s)
+                dbg("warmup done")
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    static_loss = inner_step()
+                dbg("capture done")
                 graph.replay()
-                if debug:
-                    torch.cuda.synchronize()
-                    print(f"# bench-debug: replay {i} ok", flush=True)
-        except Exception as e:  # pragma: no cover - capture unsupported
-            print(f"# graph capture failed ({e}); falling back to eager",
-                  flush=True)
-            use_graph = False
+                dbg("first replay done")
 
-    if not use_graph:
+                nocopy = os.environ.get("GENREC_BENCH_NOCOPY", "0") == "1"
+
+                def graph_step(i: int) -> None:
+                    if not nocopy:
+                        b = batches[i % len(batches)]
+                        for key in static:
+                            static[key].copy_(b[key], non_blocking=True)
+                    graph.replay()
+                    if debug:
+                        torch.cuda.synchronize()
+                        print(f"# bench-debug: replay {i} ok", flush=True)
+
+                step = graph_step
+            except Exception as e:  # pragma: no cover - capture unsupported
+                print(f"# graph capture failed ({e}); running the same "
+                      f"step eagerly", flush=True)
+                use_graph = False
+    else:  # CPU path (driver's no-GPU contract check)
         opt = torch.optim.AdamW(model.parameters(), lr=1e-4,
                                 weight_decay=0.035)
         reducer = GradReducer(model)
