@@ -9,7 +9,7 @@ separated history/target (cobra_trainer.py:25-88).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 import torch

@@ -13,7 +13,7 @@ Amazon-Beauty reference stats: ~22.3k users, ~12.1k items, ~198k events.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 import torch

@@ -20,7 +20,6 @@ MI355X redesign:
 
 from __future__ import annotations
 
-import math
 import os
 from typing import Callable, Dict, List, Optional, Tuple
 

@@ -21,7 +21,6 @@ from __future__ import annotations
 from typing import NamedTuple, Optional
 
 import torch
-import torch.nn.functional as F
 from torch import Tensor, nn
 
 from genrec_amd import ops

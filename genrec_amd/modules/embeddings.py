@@ -6,7 +6,6 @@ checkpoints interoperate.
 
 from __future__ import annotations
 
-import torch
 from torch import Tensor, nn
 
 from genrec_amd import ops

@@ -9,7 +9,6 @@ only.
 
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional
 
 import torch

@@ -9,7 +9,6 @@ from __future__ import annotations
 
 from typing import List
 
-import torch
 from torch import Tensor, nn
 
 from genrec_amd.modules.norms import L2Norm

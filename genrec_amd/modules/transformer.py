@@ -21,14 +21,13 @@ MI355X redesign notes:
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
-import torch.nn.functional as F
 from torch import Tensor, nn
 
 from genrec_amd import ops
-from genrec_amd.modules.norms import RMSNorm, T5RMSNorm
+from genrec_amd.modules.norms import RMSNorm
 
 
 def relative_position_bucket(relative_positions: Tensor, num_buckets: int = 32,

@@ -9,8 +9,6 @@ carries the decorators (utils.py:12-60) and the seq-length debug metrics
 from __future__ import annotations
 
 import functools
-from typing import Optional
-
 import torch
 
 from genrec_amd.config.ginlite import parse_config  # re-export  # noqa: F401

@@ -7,7 +7,6 @@ the reference implementations exactly; citations per function.
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

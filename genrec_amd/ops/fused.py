@@ -6,8 +6,6 @@ no-extension: eager composition with identical semantics.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn.functional as F
 from torch import Tensor

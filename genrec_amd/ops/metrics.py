@@ -8,7 +8,6 @@ metrics.py:68-74 — we reduce once per epoch instead).
 
 from __future__ import annotations
 
-import torch
 from torch import Tensor
 
 from genrec_amd.ops import eager

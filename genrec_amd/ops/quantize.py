@@ -8,7 +8,6 @@ the Gumbel/Sinkhorn quantize modes consume it (rqvae.py:202-241).
 
 from __future__ import annotations
 
-import torch
 from torch import Tensor
 
 from genrec_amd.ops import eager

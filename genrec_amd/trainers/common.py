@@ -13,13 +13,13 @@ import logging
 import os
 import random
 import time
-from typing import Any, Dict, Iterable, Optional
+from typing import Any, Dict, Optional
 
 import numpy as np
 import torch
 from torch.utils.data import DataLoader, Dataset, DistributedSampler
 
-from genrec_amd.parallel import DistributedContext, init_distributed
+from genrec_amd.parallel import DistributedContext
 
 logger = logging.getLogger("genrec_amd")
 
