@@ -94,9 +94,13 @@ class SASRecBlock(nn.Module):
 class SASRec(nn.Module):
     def __init__(self, num_items: int, max_seq_len: int = 50,
                  embed_dim: int = 64, num_heads: int = 2, num_blocks: int = 2,
-                 ffn_dim: Optional[int] = None, dropout: float = 0.2) -> None:
+                 ffn_dim: Optional[int] = None, dropout: float = 0.2,
+                 loss_type: str = "full",
+                 num_negatives: int = 1024) -> None:
         super().__init__()
         self.num_items = num_items
+        self.loss_type = loss_type
+        self.num_negatives = num_negatives
         self.max_seq_len = max_seq_len
         self.embed_dim = embed_dim
         ffn_dim = ffn_dim or embed_dim
@@ -137,10 +141,18 @@ class SASRec(nn.Module):
 
         loss = None
         if targets is not None:
-            loss = ops.tied_softmax_ce(
-                x.reshape(-1, self.embed_dim), self.item_embedding.weight,
-                targets.reshape(-1), ignore_index=0,
-            )
+            if self.loss_type == "sampled" and self.training:
+                from genrec_amd.ops.losses import sampled_tied_softmax_ce
+
+                loss = sampled_tied_softmax_ce(
+                    x.reshape(-1, self.embed_dim),
+                    self.item_embedding.weight, targets.reshape(-1),
+                    num_negatives=self.num_negatives, ignore_index=0)
+            else:
+                loss = ops.tied_softmax_ce(
+                    x.reshape(-1, self.embed_dim), self.item_embedding.weight,
+                    targets.reshape(-1), ignore_index=0,
+                )
             logits = None  # not materialized on the training path
             if not self.training:
                 logits = x @ self.item_embedding.weight.t()

@@ -138,6 +138,10 @@ def init_wandb(project: str, config: dict, enabled: bool, is_main: bool):
         import wandb
 
         wandb.init(project=project, config=config)
+        # step-keyed train metrics vs epoch-keyed eval metrics
+        # (ref tiger_trainer.py:140-141)
+        wandb.define_metric("train/*", step_metric="train/step")
+        wandb.define_metric("eval/*", step_metric="eval/epoch")
         return wandb
     except Exception:
         logger.warning("wandb unavailable; logging disabled")
