@@ -73,12 +73,14 @@ def hstu_eval_collate_fn(batch: List[Dict], max_seq_len: int = 50) -> Dict:
 
 def tiger_pad_collate(batch: List[SeqData], pad_id: int = 0,
                       padding_side: str = "left",
-                      sem_id_dim: int = 3) -> Dict[str, torch.Tensor]:
+                      sem_id_dim: int = 3,
+                      fixed_length: int = 0) -> Dict[str, torch.Tensor]:
     """Flattened sem-ID history with token_type = pos % sem_id_dim
     (ref tiger_trainer.py:27-80; note the reference 'left' branch actually
-    places ids at the sequence start — reproduced)."""
+    places ids at the sequence start — reproduced). fixed_length > 0 pads
+    every batch to that length (required for hipGraph-captured steps)."""
     B = len(batch)
-    max_len = max(len(x.item_ids) for x in batch)
+    max_len = fixed_length or max(len(x.item_ids) for x in batch)
     tgt_len = len(batch[0].target_ids)
     user_ids = torch.zeros(B, 1, dtype=torch.long)
     ids = torch.full((B, max_len), pad_id, dtype=torch.long)

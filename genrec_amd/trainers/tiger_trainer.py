@@ -90,6 +90,7 @@ def train(
     eval_max_batches: Optional[int] = None,
     profile_steps: int = 0,
     profile_trace_path: Optional[str] = None,
+    use_hip_graph: bool = False,
 ):
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "tiger")
@@ -115,7 +116,11 @@ def train(
     opt = AdamW(model.parameters(), lr=learning_rate,
                 weight_decay=weight_decay)
 
-    coll = lambda b: tiger_pad_collate(b, sem_id_dim=sem_id_dim)
+    graph_mode = (use_hip_graph and device.type == "cuda"
+                  and gradient_accumulate_every == 1)
+    fixed_len = max_seq_len * sem_id_dim if graph_mode else 0
+    coll = lambda b: tiger_pad_collate(b, sem_id_dim=sem_id_dim,
+                                       fixed_length=fixed_len)
     train_loader = common.make_loader(train_ds, batch_size, ctx, True, coll,
                                       num_workers=num_workers, seed=seed,
                                       drop_last=True)
@@ -124,7 +129,9 @@ def train(
     steps_per_epoch = max(1, len(train_loader) // gradient_accumulate_every)
     sched = get_cosine_schedule_with_warmup(
         opt, num_warmup_steps, steps_per_epoch * epochs)
-    reducer = GradReducer(model)
+    # graph mode reduces via ONE flat all-reduce inside the captured step;
+    # GradReducer's per-bucket hooks must not also fire.
+    reducer = None if graph_mode else GradReducer(model)
 
     start_epoch, step = 0, 0
     if resume_path and os.path.exists(resume_path):
@@ -139,6 +146,30 @@ def train(
                         active=profile_steps,
                         trace_path=profile_trace_path)
 
+    runner = None
+    if graph_mode:
+        # hipGraph-captured full step (fwd+bwd+allreduce+clip+AdamW) at
+        # fixed shapes; falls back to the same step eagerly on capture
+        # failure. The cosine schedule drives a device LR tensor.
+        from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+        example = next(iter(train_loader))
+        example = common.to_device(example, device)
+        runner = GraphedTrainStep(
+            model, example, loss_getter=lambda out: out.loss,
+            lr=learning_rate, weight_decay=weight_decay,
+            clip_norm=1.0, world=ctx.world_size)
+
+        import math as _math
+
+        def _cosine_lr(st: int) -> float:
+            total = steps_per_epoch * epochs
+            if st < num_warmup_steps:
+                return learning_rate * st / max(1, num_warmup_steps)
+            prog = (st - num_warmup_steps) / max(1, total - num_warmup_steps)
+            return learning_rate * max(
+                0.0, 0.5 * (1.0 + _math.cos(_math.pi * prog)))
+
     for epoch in range(start_epoch, epochs):
         model.train()
         if hasattr(train_loader.sampler, "set_epoch"):
@@ -148,6 +179,18 @@ def train(
           with prof.step():
             with roctx_range("data_to_device"):
                 batch = common.to_device(batch, device)
+            if runner is not None:
+                runner.set_lr(_cosine_lr(step))
+                loss_t = runner.step(batch)
+                step += 1
+                if ctx.is_main and step % wandb_log_interval == 0:
+                    logger.info("epoch %d step %d loss %.4f lr %.2e",
+                                epoch, step, loss_t.item(),
+                                float(runner.lr_t))
+                    wb.log({"train/loss": loss_t.item()})
+                if max_steps is not None and step >= max_steps:
+                    break
+                continue
             micro = (it + 1) % gradient_accumulate_every == 0
             reducer.skip_sync = not micro
             with roctx_range("forward"), amp_ctx:
@@ -186,7 +229,8 @@ def train(
         if ctx.is_main and (epoch + 1) % save_every_epoch == 0:
             common.save_checkpoint(
                 os.path.join(save_dir_root, f"checkpoint_epoch_{epoch}.pt"),
-                model, opt, sched, epoch=epoch, is_main=True)
+                model, runner.opt if runner else opt,
+                None if runner else sched, epoch=epoch, is_main=True)
         if max_steps is not None and step >= max_steps:
             break
     prof.report()
@@ -194,7 +238,8 @@ def train(
     if ctx.is_main:
         common.save_checkpoint(
             os.path.join(save_dir_root, "checkpoint_final.pt"),
-            model, opt, sched, epoch=epochs - 1, is_main=True)
+            model, runner.opt if runner else opt,
+            None if runner else sched, epoch=epochs - 1, is_main=True)
     wb.finish()
 
 

@@ -640,3 +640,20 @@ def test_rqvae_gpu_train_steps():
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]  # reconstructing a fixed batch improves
     assert 0.0 <= out.p_unique_ids.item() <= 1.0
+
+
+def test_tiger_trainer_hip_graph_mode(tmp_path):
+    """End-to-end trainer in hipGraph mode: fixed-shape collate, captured
+    step, cosine LR through the device lr tensor, bf16 eval afterwards."""
+    from genrec_amd.trainers import tiger_trainer
+
+    tiger_trainer.train(
+        epochs=1, max_steps=6, batch_size=32, embedding_dim=32, attn_dim=64,
+        num_heads=2, n_layers=2, num_item_embeddings=64, sem_id_dim=3,
+        max_seq_len=8, num_warmup_steps=3, do_eval=True, eval_max_batches=1,
+        amp=False, use_hip_graph=True, num_workers=0,
+        save_dir_root=str(tmp_path), wandb_logging=False,
+        save_every_epoch=100)
+    import os
+
+    assert os.path.exists(os.path.join(str(tmp_path), "checkpoint_final.pt"))
