@@ -1,0 +1,79 @@
+"""Linear layer with a split-K grad-weight backward.
+
+Training GEMMs in this framework are tall-and-skinny: activations are
+[B*L, d] with B*L ~ 15k rows while d is a few hundred (TIGER encoder:
+15616 x 384). The forward/input-grad GEMMs shard fine, but the
+grad-weight GEMM dW = dY^T X has a tiny MxN output (384x384) and a huge
+K — hipBLASLt's heuristic picks a non-split-K kernel whose grid (~150
+workgroups) cannot fill 256 CUs, and each of the ~25 such GEMMs per
+TIGER step costs 75-105us (measured, tools/bench_splitk.py). Splitting K
+into chunks and running one batched GEMM + fp32 partial-sum reduce fills
+the chip and is ~2.7x faster (87 -> 31us at 384x384, K=15616).
+
+`SplitKLinear` is a drop-in `nn.Linear` (bias supported) whose backward
+uses the chunked-bmm path whenever K is large and divisible; otherwise it
+falls back to the plain GEMM. Forward is a stock hipBLASLt GEMM either
+way. hipGraph-capture safe (static shapes, no host branches on data).
+
+Reference parity: replaces the implicit nn.Linear grad GEMMs behind
+transformer.py:72-124 and tiger.py:161-207 of the reference.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import nn
+
+_SPLITK_MIN_K = 4096
+
+
+def _pick_chunks(k: int) -> int:
+    for nc in (16, 8, 4):
+        if k % nc == 0 and k // nc >= 256:
+            return nc
+    return 1
+
+
+class _SplitKLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor,
+                bias: Optional[torch.Tensor]):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        out = x.matmul(weight.t())
+        if bias is not None:
+            out = out + bias
+        return out
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, weight = ctx.saved_tensors
+        dx = dy.matmul(weight) if ctx.needs_input_grad[0] else None
+        dw = None
+        if ctx.needs_input_grad[1]:
+            x2 = x.reshape(-1, x.shape[-1])
+            dy2 = dy.reshape(-1, dy.shape[-1])
+            k = x2.shape[0]
+            nc = _pick_chunks(k) if k >= _SPLITK_MIN_K else 1
+            if nc > 1 and x2.is_contiguous() and dy2.is_contiguous():
+                part = torch.bmm(
+                    dy2.view(nc, k // nc, -1).transpose(1, 2),
+                    x2.view(nc, k // nc, -1))
+                # fp32 partial sum preserves the single-GEMM fp32-accum
+                # rounding behavior
+                dw = part.sum(0, dtype=torch.float32).to(weight.dtype)
+            else:
+                dw = dy2.t().matmul(x2)
+        db = None
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.reshape(-1, dy.shape[-1]).sum(0)
+        return dx, dw, db
+
+
+class SplitKLinear(nn.Linear):
+    """nn.Linear with split-K grad-weight GEMMs (see module docstring)."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return _SplitKLinearFn.apply(x, self.weight, self.bias)

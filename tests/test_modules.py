@@ -173,3 +173,33 @@ def test_sasrec_sampled_loss_mode():
     _, loss = m(ids, ids)
     assert torch.isfinite(loss)
     loss.backward()
+
+
+def test_splitk_linear_matches_nn_linear():
+    from genrec_amd.ops.linear import SplitKLinear, _pick_chunks
+
+    torch.manual_seed(0)
+    assert _pick_chunks(15616) > 1 and _pick_chunks(100) == 1
+    for bias in (False, True):
+        ref = torch.nn.Linear(32, 48, bias=bias)
+        m = SplitKLinear(32, 48, bias=bias)
+        with torch.no_grad():
+            m.weight.copy_(ref.weight)
+            if bias:
+                m.bias.copy_(ref.bias)
+        # K=4096 triggers the chunked-bmm grad-weight path
+        x1 = torch.randn(4096, 32, requires_grad=True)
+        x2 = x1.detach().clone().requires_grad_(True)
+        dy = torch.randn(4096, 48)
+        m(x1).backward(dy)
+        ref(x2).backward(dy)
+        assert torch.allclose(x1.grad, x2.grad, atol=1e-5)
+        assert torch.allclose(m.weight.grad, ref.weight.grad,
+                              atol=1e-3, rtol=1e-4)
+        if bias:
+            assert torch.allclose(m.bias.grad, ref.bias.grad, atol=1e-4)
+    # 3-D input + small-K fallback
+    m3 = SplitKLinear(16, 8, bias=False)
+    x3 = torch.randn(4, 10, 16, requires_grad=True)
+    m3(x3).sum().backward()
+    assert x3.grad.shape == x3.shape
