@@ -50,10 +50,13 @@ class _SplitKLinearFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
         x, weight = ctx.saved_tensors
-        dx = dy.matmul(weight) if ctx.needs_input_grad[0] else None
+        # under autocast the saved weight/x may be fp32 while dy is bf16;
+        # match dtypes explicitly (no-op casts on the pure-bf16 path)
+        dx = (dy.matmul(weight.to(dy.dtype))
+              if ctx.needs_input_grad[0] else None)
         dw = None
         if ctx.needs_input_grad[1]:
-            x2 = x.reshape(-1, x.shape[-1])
+            x2 = x.reshape(-1, x.shape[-1]).to(dy.dtype)
             dy2 = dy.reshape(-1, dy.shape[-1])
             k = x2.shape[0]
             nc = _pick_chunks(k) if k >= _SPLITK_MIN_K else 1
@@ -65,7 +68,7 @@ class _SplitKLinearFn(torch.autograd.Function):
                 # rounding behavior
                 dw = part.sum(0, dtype=torch.float32).to(weight.dtype)
             else:
-                dw = dy2.t().matmul(x2)
+                dw = dy2.t().matmul(x2).to(weight.dtype)
         db = None
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.reshape(-1, dy.shape[-1]).sum(0)

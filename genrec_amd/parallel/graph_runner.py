@@ -23,9 +23,11 @@ from __future__ import annotations
 
 from typing import Callable, Dict, Optional
 
+import logging
+
 import torch
 
-from genrec_amd.trainers.common import logger
+logger = logging.getLogger("genrec_amd")
 
 
 class GraphedTrainStep:

@@ -203,3 +203,16 @@ def test_splitk_linear_matches_nn_linear():
     x3 = torch.randn(4, 10, 16, requires_grad=True)
     m3(x3).sum().backward()
     assert x3.grad.shape == x3.shape
+
+
+def test_splitk_linear_autocast_mixed_dtypes():
+    from genrec_amd.ops.linear import SplitKLinear
+
+    m = SplitKLinear(32, 16, bias=False)  # fp32 params
+    x = torch.randn(4096, 32, requires_grad=True)
+    with torch.amp.autocast("cpu", dtype=torch.bfloat16):
+        y = m(x)
+    assert y.dtype == torch.bfloat16
+    y.float().sum().backward()
+    assert m.weight.grad is not None and torch.isfinite(m.weight.grad).all()
+    assert x.grad is not None
