@@ -33,6 +33,8 @@ import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
 
+from genrec_amd.ops.linear import SplitKLinear
+
 from genrec_amd.config import ginlite
 from genrec_amd.modules.encoders import LightT5Encoder
 
@@ -177,7 +179,7 @@ class Cobra(nn.Module):
                                     n_heads=decoder_num_heads,
                                     dropout=decoder_dropout)
         self.sparse_head = nn.ModuleList([
-            nn.Linear(d_model, id_vocab_size) for _ in range(n_codebooks)])
+            SplitKLinear(d_model, id_vocab_size) for _ in range(n_codebooks)])
         self.temperature = temperature
         # MoCo queue: present for parity but unused (ref cobra.py:497-508)
         self.register_buffer("feat_queue",

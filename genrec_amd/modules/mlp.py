@@ -11,6 +11,8 @@ from typing import List
 
 from torch import Tensor, nn
 
+from genrec_amd.ops.linear import SplitKLinear
+
 from genrec_amd.modules.norms import L2Norm
 
 
@@ -24,7 +26,7 @@ class MLP(nn.Module):
         dims = [input_dim] + self.hidden_dims + [out_dim]
         layers: list[nn.Module] = []
         for i, (d_in, d_out) in enumerate(zip(dims[:-1], dims[1:])):
-            layers.append(nn.Linear(d_in, d_out, bias=False))
+            layers.append(SplitKLinear(d_in, d_out, bias=False))
             if i != len(dims) - 2:
                 layers.append(nn.SiLU())
                 if dropout != 0:
