@@ -248,11 +248,16 @@ hstu_attn_bwd_kernel(
 
   const int col_base = lane & 15;
   const int row_grp = (lane >> 4) << 2;
+  // transposed-tile stores pack a lane's 4 r-values (consecutive columns
+  // of row j) into ONE aligned 8-byte ds_write — see attention_mfma.hip
+  typedef __attribute__((ext_vector_type(4))) short short4vh;
+  const int i0 = strip + row_grp;
 #pragma unroll
   for (int f = 0; f < 4; ++f) {
+    short4vh dpack, apack;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int i = strip + row_grp + r;
+      int i = i0 + r;
       int j = f * 16 + col_base;
       float dsv = 0.f, adv = 0.f;
       if (i < L && j < L) {
@@ -271,13 +276,15 @@ hstu_attn_bwd_kernel(
           }
         }
       }
-      *reinterpret_cast<__hip_bfloat16*>(dsn + hswz(i, j * 2)) =
-          __float2bfloat16(dsv);
-      *reinterpret_cast<__hip_bfloat16*>(dst + hswz(j, i * 2)) =
-          __float2bfloat16(dsv);
-      *reinterpret_cast<__hip_bfloat16*>(adt + hswz(j, i * 2)) =
-          __float2bfloat16(adv);
+      __hip_bfloat16 dh = __float2bfloat16(dsv);
+      __hip_bfloat16 ah = __float2bfloat16(adv);
+      dpack[r] = *reinterpret_cast<short*>(&dh);
+      apack[r] = *reinterpret_cast<short*>(&ah);
+      *reinterpret_cast<__hip_bfloat16*>(dsn + hswz(i, j * 2)) = dh;
     }
+    int j = f * 16 + col_base;
+    *reinterpret_cast<short4vh*>(dst + hswz(j, i0 * 2)) = dpack;
+    *reinterpret_cast<short4vh*>(adt + hswz(j, i0 * 2)) = apack;
   }
   __builtin_amdgcn_wave_barrier();
 
