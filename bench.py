@@ -73,7 +73,7 @@ def main() -> None:
     world = int(os.environ.get("WORLD_SIZE", "1"))
 
     from genrec_amd.models.tiger import Tiger
-    from genrec_amd.parallel import GradReducer, init_distributed
+    from genrec_amd.parallel import init_distributed
     from genrec_amd.parallel.ddp import broadcast_parameters
 
     ctx = init_distributed()
@@ -109,104 +109,28 @@ def main() -> None:
     use_graph = use_gpu and os.environ.get("GENREC_BENCH_GRAPH", "1") == "1"
 
     if use_gpu:
-        # Pure-bf16 compute with fp32 master weights (standard bf16 mixed
-        # precision a la Megatron): the model runs natively in bf16 — no
-        # autocast, so no per-layer weight-cast kernels (~2900/step) — the
-        # optimizer steps fp32 masters, and one fused foreach-copy refreshes
-        # the bf16 params after each step. All capturable.
-        model = model.to(torch.bfloat16)
-        model.train()
-        params = [p for p in model.parameters() if p.requires_grad]
-        masters = [p.detach().float().clone() for p in params]
-        opt = torch.optim.AdamW(masters, lr=1e-4, weight_decay=0.035,
-                                capturable=True, foreach=True)
+        # GraphedTrainStep: pure-bf16 model + ONE flat fp32 master/moment
+        # buffer stepped by the genrec fused_adamw HIP kernel, bf16 flat
+        # gradient all-reduced in a single RCCL message, whole step
+        # hipGraph-captured (eager fallback on capture failure keeps N=1
+        # and N>1 numbers mode-comparable).
+        from genrec_amd.parallel.graph_runner import GraphedTrainStep
 
-        static = {k: v.clone() for k, v in batches[0].items()}
+        runner = GraphedTrainStep(
+            model, batches[0], lambda out: out.loss, lr=1e-4,
+            weight_decay=0.035, clip_norm=1.0, world=world,
+            use_graph=use_graph)
+        use_graph = runner.captured
+        nocopy = os.environ.get("GENREC_BENCH_NOCOPY", "0") == "1"
 
-        # bf16 grads live as views into ONE flat buffer: a single RCCL
-        # all-reduce moves the whole gradient (xGMI likes few large
-        # messages; bf16 halves the bytes), zeroing is one fill, the
-        # global-norm clip is a norm+scale on the flat fp32 copy.
-        model(**static).loss.backward()
-        flat_grads = torch.zeros(sum(p.numel() for p in params),
-                                 device=device, dtype=torch.bfloat16)
-        off = 0
-        for p in params:
-            p.grad = flat_grads[off:off + p.numel()].view_as(p)
-            off += p.numel()
-        flat_master_grad = torch.zeros(flat_grads.numel(), device=device)
-        moff = 0
-        for m in masters:
-            m.grad = flat_master_grad[moff:moff + m.numel()].view_as(m)
-            moff += m.numel()
-
-        def inner_step():
-            flat_grads.zero_()
-            out = model(**static)
-            out.loss.backward()
-            if world > 1:
-                import torch.distributed as dist
-
-                dist.all_reduce(flat_grads)
-                flat_grads.mul_(1.0 / world)
-            flat_master_grad.copy_(flat_grads)
-            norm = flat_master_grad.norm()
-            flat_master_grad.mul_(
-                torch.clamp(1.0 / (norm + 1e-6), max=1.0))
-            opt.step()
-            with torch.no_grad():
-                torch._foreach_copy_(params, masters)
-            return out.loss
-
-        debug = os.environ.get("GENREC_BENCH_DEBUG", "0") == "1"
-
-        def dbg(msg):
-            if debug:
-                torch.cuda.synchronize()
-                print(f"# bench-debug: {msg}", flush=True)
-
-        def eager_step(i: int) -> None:
-            b = batches[i % len(batches)]
-            for key in static:
-                static[key].copy_(b[key], non_blocking=True)
-            inner_step()
-
-        step = eager_step
-        if use_graph:
-            try:
-                # warmup on a side stream (required before capture)
-                s = torch.cuda.Stream()
-                s.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(s):
-                    for _ in range(3):
-                        inner_step()
-                torch.cuda.current_stream().wait_stream(s)
-                dbg("warmup done")
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
-                    static_loss = inner_step()
-                dbg("capture done")
-                graph.replay()
-                dbg("first replay done")
-
-                nocopy = os.environ.get("GENREC_BENCH_NOCOPY", "0") == "1"
-
-                def graph_step(i: int) -> None:
-                    if not nocopy:
-                        b = batches[i % len(batches)]
-                        for key in static:
-                            static[key].copy_(b[key], non_blocking=True)
-                    graph.replay()
-                    if debug:
-                        torch.cuda.synchronize()
-                        print(f"# bench-debug: replay {i} ok", flush=True)
-
-                step = graph_step
-            except Exception as e:  # pragma: no cover - capture unsupported
-                print(f"# graph capture failed ({e}); running the same "
-                      f"step eagerly", flush=True)
-                use_graph = False
+        def step(i: int) -> None:
+            if nocopy and runner.captured:
+                runner._graph.replay()
+                return
+            runner.step(batches[i % len(batches)])
     else:  # CPU path (driver's no-GPU contract check)
+        from genrec_amd.parallel import GradReducer
+
         opt = torch.optim.AdamW(model.parameters(), lr=1e-4,
                                 weight_decay=0.035)
         reducer = GradReducer(model)

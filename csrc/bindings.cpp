@@ -73,6 +73,10 @@ std::vector<torch::Tensor> relu_dropout_fwd(
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
+void fused_adamw(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
+                 torch::Tensor v, torch::Tensor out_p, torch::Tensor lr,
+                 torch::Tensor scale, torch::Tensor step, double beta1,
+                 double beta2, double eps, double weight_decay);
 
 }  // namespace genrec
 
@@ -96,4 +100,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_dropout_fwd", &genrec::relu_dropout_fwd, "dropout(relu) fwd");
   m.def("dropout_fuse_bwd", &genrec::dropout_fuse_bwd, "fused dropout bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
+  m.def("fused_adamw", &genrec::fused_adamw,
+        "fused flat AdamW step (device lr/scale/step scalars)");
 }

@@ -1,17 +1,22 @@
 """hipGraph-captured training step (shared by bench.py and the trainers).
 
 Captures one full training step — forward, backward, flat-gradient RCCL
-all-reduce, global-norm clip, fp32-master AdamW, bf16 parameter refresh —
-and replays it per batch. Requirements/properties:
+all-reduce, global-norm clip, fused flat AdamW — and replays it per
+batch. Requirements/properties:
 
-  * the model runs natively in bf16 (fp32 master weights live in the
-    optimizer); no autocast, so no per-layer weight-cast kernels
+  * the model runs natively in bf16 (fp32 master weights live in ONE flat
+    buffer); no autocast, so no per-layer weight-cast kernels
   * every batch must have the SAME shapes (fixed-size collate + drop_last)
   * gradients are views into one flat bf16 buffer -> N>1 all-reduce is a
     single large RCCL message
-  * the learning rate is a device tensor read by the captured capturable
-    AdamW, so LR schedules keep working across replays (set_lr / an
-    lr_lambda evaluated on the host each step)
+  * on CUDA the optimizer is the genrec fused_adamw kernel
+    (csrc/kernels/adamw.hip): masters/moments/params/grads are flat
+    tensors and the whole AdamW step is ONE bandwidth-bound kernel.
+    torch's capturable foreach AdamW with a tensor lr falls off the
+    foreach fast route and launches ~4 tiny kernels per parameter per
+    step (~1.5 ms/step measured on TIGER)
+  * lr / clip-scale / step count are DEVICE scalars read by the kernel,
+    so LR schedules (set_lr) and clipping keep working across replays
   * dropout stays live across replays (torch philox is graph-safe; the
     genrec_amd fused kernels read a captured device seed counter)
   * on ANY capture failure the same step runs eagerly — numerics are
@@ -21,9 +26,8 @@ and replays it per batch. Requirements/properties:
 
 from __future__ import annotations
 
-from typing import Callable, Dict, Optional
-
 import logging
+from typing import Callable, Dict, Optional
 
 import torch
 
@@ -35,7 +39,8 @@ class GraphedTrainStep:
                  example_batch: Dict[str, torch.Tensor],
                  loss_getter: Callable,
                  lr: float = 1e-4, weight_decay: float = 0.0,
-                 betas=(0.9, 0.999), clip_norm: Optional[float] = 1.0,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 clip_norm: Optional[float] = 1.0,
                  world: int = 1, use_graph: bool = True,
                  warmup_iters: int = 3):
         self.model = model.to(torch.bfloat16)
@@ -43,36 +48,61 @@ class GraphedTrainStep:
         self.world = world
         self.loss_getter = loss_getter
         self.clip_norm = clip_norm
+        self.betas = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
         device = next(model.parameters()).device
         self.device = device
 
         self.params = [p for p in model.parameters() if p.requires_grad]
-        self.masters = [p.detach().float().clone() for p in self.params]
-        # capturable AdamW (CUDA only) reads a DEVICE lr tensor, so LR
-        # schedules survive graph replay; the CPU fallback uses a plain
-        # float lr updated through param_groups.
-        self._capturable = device.type == "cuda"
-        self.lr_t = torch.tensor(lr, device=device)
-        self.opt = torch.optim.AdamW(
-            self.masters, lr=self.lr_t if self._capturable else lr,
-            betas=betas, weight_decay=weight_decay,
-            capturable=self._capturable, foreach=True)
+        n_total = sum(p.numel() for p in self.params)
+        self.lr_t = torch.tensor(float(lr), device=device,
+                                 dtype=torch.float32)
+
+        from genrec_amd import ops
+
+        self.fused = device.type == "cuda" and ops.has_ext()
+        if self.fused:
+            # flat bf16 params (p.data re-pointed into views) + flat fp32
+            # masters/moments: the fused AdamW kernel updates everything
+            # in one pass and writes the bf16 params directly.
+            self.flat_params = torch.empty(n_total, device=device,
+                                           dtype=torch.bfloat16)
+            off = 0
+            with torch.no_grad():
+                for p in self.params:
+                    view = self.flat_params[off:off + p.numel()].view_as(p)
+                    view.copy_(p.data)
+                    p.data = view
+                    off += p.numel()
+            self.flat_master = self.flat_params.float()
+            self.m = torch.zeros(n_total, device=device)
+            self.v = torch.zeros(n_total, device=device)
+            self.step_t = torch.zeros(1, device=device, dtype=torch.int32)
+            self._no_scale = torch.empty(0, device=device)
+            self.opt = None
+        else:
+            self.masters = [p.detach().float().clone() for p in self.params]
+            self.opt = torch.optim.AdamW(self.masters, lr=lr, betas=betas,
+                                         eps=eps,
+                                         weight_decay=weight_decay)
 
         self.static = {k: v.clone() for k, v in example_batch.items()}
-        # materialize grads once, then re-point them into flat buffers
+        # materialize grads once, then re-point them into the flat buffer
         self.loss_getter(self.model(**self.static)).backward()
-        n_total = sum(p.numel() for p in self.params)
         self.flat_grads = torch.zeros(n_total, device=device,
                                       dtype=torch.bfloat16)
         off = 0
         for p in self.params:
             p.grad = self.flat_grads[off:off + p.numel()].view_as(p)
             off += p.numel()
-        self.flat_master_grad = torch.zeros(n_total, device=device)
-        off = 0
-        for m in self.masters:
-            m.grad = self.flat_master_grad[off:off + m.numel()].view_as(m)
-            off += m.numel()
+        if not self.fused:
+            self.flat_master_grad = torch.zeros(n_total, device=device)
+            off = 0
+            for mt in self.masters:
+                mt.grad = self.flat_master_grad[off:off + mt.numel()] \
+                    .view_as(mt)
+                off += mt.numel()
 
         self._graph = None
         self._loss = None
@@ -107,6 +137,20 @@ class GraphedTrainStep:
 
             dist.all_reduce(self.flat_grads)
             self.flat_grads.mul_(1.0 / self.world)
+        if self.fused:
+            from genrec_amd import ops
+
+            if self.clip_norm is not None:
+                norm = torch.linalg.vector_norm(self.flat_grads, 2,
+                                                dtype=torch.float32)
+                scale = (self.clip_norm / (norm + 1e-6)).clamp(max=1.0)
+            else:
+                scale = self._no_scale
+            ops.ext().fused_adamw(
+                self.flat_master, self.flat_grads, self.m, self.v,
+                self.flat_params, self.lr_t, scale, self.step_t,
+                self.betas[0], self.betas[1], self.eps, self.weight_decay)
+            return loss
         self.flat_master_grad.copy_(self.flat_grads)
         if self.clip_norm is not None:
             norm = self.flat_master_grad.norm()
@@ -123,7 +167,7 @@ class GraphedTrainStep:
 
     def set_lr(self, lr: float) -> None:
         self.lr_t.fill_(lr)
-        if not self._capturable:
+        if self.opt is not None:
             for g in self.opt.param_groups:
                 g["lr"] = lr
 
@@ -138,14 +182,35 @@ class GraphedTrainStep:
         return self._inner()
 
     def state_dict(self) -> dict:
+        if self.fused:
+            return {"flat_master": self.flat_master.cpu(),
+                    "m": self.m.cpu(), "v": self.v.cpu(),
+                    "step": self.step_t.cpu(), "lr": float(self.lr_t)}
         return {"masters": [m.detach().cpu() for m in self.masters],
                 "optimizer": self.opt.state_dict(),
                 "lr": float(self.lr_t)}
 
     def load_state_dict(self, state: dict) -> None:
         with torch.no_grad():
-            for m, s in zip(self.masters, state["masters"]):
-                m.copy_(s.to(m.device))
-            torch._foreach_copy_(self.params, self.masters)
-        self.opt.load_state_dict(state["optimizer"])
+            if self.fused:
+                if "flat_master" in state:
+                    self.flat_master.copy_(state["flat_master"].to(
+                        self.device))
+                    self.m.copy_(state["m"].to(self.device))
+                    self.v.copy_(state["v"].to(self.device))
+                    self.step_t.copy_(state["step"].to(self.device))
+                else:  # checkpoint written by the non-fused path
+                    off = 0
+                    for ms in state["masters"]:
+                        n = ms.numel()
+                        self.flat_master[off:off + n].copy_(
+                            ms.reshape(-1).to(self.device))
+                        off += n
+                self.flat_params.copy_(
+                    self.flat_master.to(torch.bfloat16))
+            else:
+                for mt, s in zip(self.masters, state["masters"]):
+                    mt.copy_(s.to(mt.device))
+                torch._foreach_copy_(self.params, self.masters)
+                self.opt.load_state_dict(state["optimizer"])
         self.set_lr(state.get("lr", float(self.lr_t)))

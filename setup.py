@@ -26,6 +26,7 @@ sources = [
     "csrc/kernels/metrics.hip",
     "csrc/kernels/embedding.hip",
     "csrc/kernels/fused_elementwise.hip",
+    "csrc/kernels/adamw.hip",
 ]
 
 cxx_flags = ["-O3", "-std=c++17"]
