@@ -63,6 +63,10 @@ attn_fwd_mfma_kernel(
     unsigned char* __restrict__ drop_mask,
     const unsigned int* __restrict__ seed_dev,
     int B, int H, int Lq, int Lk, int D,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl,
+    int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl,
+    int64_t o_sb, int64_t o_sh, int64_t o_sl,
     float scale, int bias_dim, bool bias_bf16, bool causal,
     float dropout_p, unsigned int seed, int q_tile) {
   if (seed_dev) seed += *seed_dev;
@@ -94,14 +98,14 @@ attn_fwd_mfma_kernel(
     int qi = q0 + row;
     if (qi < Lq && d0 < D) {
       val = *reinterpret_cast<const short8v*>(
-          &q[IDX4M(b, h, qi, d0, H, Lq, D)]);
+          &q[(int64_t)b * q_sb + h * q_sh + qi * q_sl + d0]);
     }
     *reinterpret_cast<short8v*>(qs + swz(row, d0 * 2)) = val;
     // K tile
     val = z;
     if (row < Lk && d0 < D) {
       val = *reinterpret_cast<const short8v*>(
-          &k[IDX4M(b, h, row, d0, H, Lk, D)]);
+          &k[(int64_t)b * k_sb + h * k_sh + row * k_sl + d0]);
     }
     *reinterpret_cast<short8v*>(ks + swz(row, d0 * 2)) = val;
     // V^T tile: vt[d][j] = V[j][d]; this thread owns (d=row, j=d0..d0+7)
@@ -109,7 +113,8 @@ attn_fwd_mfma_kernel(
     for (int jj = 0; jj < 8; ++jj) {
       int j = d0 + jj;
       tmp[jj] = (j < Lk && row < D)
-          ? v[IDX4M(b, h, j, row, H, Lk, D)] : __hip_bfloat16(0.f);
+          ? v[(int64_t)b * v_sb + h * v_sh + j * v_sl + row]
+          : __hip_bfloat16(0.f);
     }
     *reinterpret_cast<short8v*>(vt + swz(row, d0 * 2)) =
         *reinterpret_cast<short8v*>(tmp);
@@ -247,7 +252,8 @@ attn_fwd_mfma_kernel(
       int i = q0 + strip + row_grp + r;
       int d = f * 16 + col_base;
       if (i < Lq && d < D) {
-        out[IDX4M(b, h, i, d, H, Lq, D)] = __float2bfloat16(acc2[f][r]);
+        out[(int64_t)b * o_sb + h * o_sh + i * o_sl + d] =
+            __float2bfloat16(acc2[f][r]);
       }
     }
   }
@@ -271,6 +277,13 @@ attn_bwd_ds_kernel(
     __hip_bfloat16* __restrict__ dv_out,      // [B,H,Lk,D]
     float* __restrict__ ds_saved,             // null | [B,H,Lq,Lk] bias grad
     int B, int H, int Lq, int Lk, int D,
+    int64_t do_sb, int64_t do_sh, int64_t do_sl,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl,
+    int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl,
+    int64_t dq_sb, int64_t dq_sh, int64_t dq_sl,
+    int64_t dk_sb, int64_t dk_sh, int64_t dk_sl,
+    int64_t dv_sb, int64_t dv_sh, int64_t dv_sl,
     float scale, float dropout_p) {
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh % H;
@@ -295,13 +308,13 @@ attn_bwd_ds_kernel(
     short8v val = {};
     if (row < Lq && d0 < D) {
       val = *reinterpret_cast<const short8v*>(
-          &dout[IDX4M(b, h, row, d0, H, Lq, D)]);
+          &dout[(int64_t)b * do_sb + h * do_sh + row * do_sl + d0]);
     }
     *reinterpret_cast<short8v*>(dos + swz(row, d0 * 2)) = val;
     short8v val2 = {};
     if (row < Lk && d0 < D) {
       val2 = *reinterpret_cast<const short8v*>(
-          &v[IDX4M(b, h, row, d0, H, Lk, D)]);
+          &v[(int64_t)b * v_sb + h * v_sh + row * v_sl + d0]);
     }
     *reinterpret_cast<short8v*>(vs + swz(row, d0 * 2)) = val2;
     // transposed images: row = d, cols = sequence positions
@@ -309,11 +322,14 @@ attn_bwd_ds_kernel(
     for (int jj = 0; jj < 8; ++jj) {
       int p = d0 + jj;
       tk[jj] = (p < Lk && row < D)
-          ? k[IDX4M(b, h, p, row, H, Lk, D)] : __hip_bfloat16(0.f);
+          ? k[(int64_t)b * k_sb + h * k_sh + p * k_sl + row]
+          : __hip_bfloat16(0.f);
       tq[jj] = (p < Lq && row < D)
-          ? q[IDX4M(b, h, p, row, H, Lq, D)] : __hip_bfloat16(0.f);
+          ? q[(int64_t)b * q_sb + h * q_sh + p * q_sl + row]
+          : __hip_bfloat16(0.f);
       td[jj] = (p < Lq && row < D)
-          ? dout[IDX4M(b, h, p, row, H, Lq, D)] : __hip_bfloat16(0.f);
+          ? dout[(int64_t)b * do_sb + h * do_sh + p * do_sl + row]
+          : __hip_bfloat16(0.f);
     }
     *reinterpret_cast<short8v*>(kt + swz(row, d0 * 2)) =
         *reinterpret_cast<short8v*>(tk);
@@ -442,7 +458,8 @@ attn_bwd_ds_kernel(
         int i = strip + row_grp + r;
         int d = f * 16 + col_base;
         if (i < Lq && d < D) {
-          dq_out[IDX4M(b, h, i, d, H, Lq, D)] = __float2bfloat16(accq[f][r]);
+          dq_out[(int64_t)b * dq_sb + h * dq_sh + i * dq_sl + d] =
+              __float2bfloat16(accq[f][r]);
         }
       }
     }
@@ -476,8 +493,10 @@ attn_bwd_ds_kernel(
         int j = strip + row_grp + r;
         int d = f * 16 + col_base;
         if (j < Lk && d < D) {
-          dk_out[IDX4M(b, h, j, d, H, Lk, D)] = __float2bfloat16(acck[f][r]);
-          dv_out[IDX4M(b, h, j, d, H, Lk, D)] = __float2bfloat16(accv[f][r]);
+          dk_out[(int64_t)b * dk_sb + h * dk_sh + j * dk_sl + d] =
+              __float2bfloat16(acck[f][r]);
+          dv_out[(int64_t)b * dv_sb + h * dv_sh + j * dv_sl + d] =
+              __float2bfloat16(accv[f][r]);
         }
       }
     }
@@ -497,6 +516,11 @@ std::vector<torch::Tensor> attn_fwd_mfma(
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
   TORCH_CHECK(Lk <= TILE && D <= TILE && D % 32 == 0);
+  // q/k/v may be transpose views (e.g. [B,L,H,D] -> [B,H,L,D]); only the
+  // D axis must be unit-stride. Saves the .contiguous() copies around
+  // every attention call.
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "attn_fwd_mfma: innermost (D) stride must be 1");
   auto out = torch::empty_like(q);
   auto p_saved = torch::empty({B, H, Lq, Lk},
                               q.options().dtype(torch::kFloat32));
@@ -542,7 +566,12 @@ std::vector<torch::Tensor> attn_fwd_mfma(
       seed_dev.has_value()                                                     \
           ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())        \
           : nullptr,                                                           \
-      B, H, Lq, Lk, D, (float)scale, bias_dim, bias_bf16, causal,              \
+      B, H, Lq, Lk, D,                                                         \
+      q.stride(0), q.stride(1), q.stride(2),                                   \
+      k.stride(0), k.stride(1), k.stride(2),                                   \
+      v.stride(0), v.stride(1), v.stride(2),                                   \
+      out.stride(0), out.stride(1), out.stride(2),                             \
+      (float)scale, bias_dim, bias_bf16, causal,                               \
       (float)dropout_p, (unsigned int)seed, q_tile)
 
   if (act == 0) LAUNCH_FWD_MFMA(false);
@@ -561,6 +590,9 @@ std::vector<torch::Tensor> attn_bwd_mfma(
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
   TORCH_CHECK(Lq <= TILE && Lk <= TILE && D % 32 == 0);
+  TORCH_CHECK(dout.stride(3) == 1 && q.stride(3) == 1 && k.stride(3) == 1 &&
+                  v.stride(3) == 1,
+              "attn_bwd_mfma: innermost (D) stride must be 1");
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
@@ -591,7 +623,15 @@ std::vector<torch::Tensor> attn_bwd_mfma(
       reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),                        \
       reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),                        \
       reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()), ds_ptr,                \
-      B, H, Lq, Lk, D, (float)scale, (float)dropout_p)
+      B, H, Lq, Lk, D,                                                         \
+      dout.stride(0), dout.stride(1), dout.stride(2),                          \
+      q.stride(0), q.stride(1), q.stride(2),                                   \
+      k.stride(0), k.stride(1), k.stride(2),                                   \
+      v.stride(0), v.stride(1), v.stride(2),                                   \
+      dq.stride(0), dq.stride(1), dq.stride(2),                                \
+      dk.stride(0), dk.stride(1), dk.stride(2),                                \
+      dv.stride(0), dv.stride(1), dv.stride(2),                                \
+      (float)scale, (float)dropout_p)
 
   if (act == 0) LAUNCH_BWD_DS(false);
   else LAUNCH_BWD_DS(true);

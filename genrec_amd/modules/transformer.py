@@ -142,7 +142,7 @@ class T5Attention(nn.Module):
             self.dropout_p, self.training,
         )
         b = out.size(0)
-        out = out.transpose(1, 2).contiguous().view(b, -1, self.d_model)
+        out = out.transpose(1, 2).reshape(b, -1, self.d_model)
         return self.o(out)
 
 

@@ -57,7 +57,17 @@ class _FusedAttnFn(torch.autograd.Function):
                 scale, causal, act, dropout_p, training):
         from genrec_amd import ops
 
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        use_mfma = (q.dtype == torch.bfloat16 and q.size(3) % 32 == 0
+                    and os.environ.get("GENREC_DISABLE_MFMA", "0") != "1")
+        if use_mfma:
+            # the MFMA kernels are stride-aware: [B,L,H,D] transpose views
+            # (the layout GEMM outputs naturally produce) go in directly —
+            # no .contiguous() copies on q/k/v, out, or the grads.
+            q = q if q.stride(-1) == 1 else q.contiguous()
+            k = k if k.stride(-1) == 1 else k.contiguous()
+            v = v if v.stride(-1) == 1 else v.contiguous()
+        else:
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
         seed_dev = None
         seed = 0
         if dropout_p > 0 and training:
@@ -65,8 +75,6 @@ class _FusedAttnFn(torch.autograd.Function):
                 if not torch.cuda.is_current_stream_capturing() else 12345
             seed_dev = _seed_counter(q.device)
             seed_dev.add_(1)
-        use_mfma = (q.dtype == torch.bfloat16 and q.size(3) % 32 == 0
-                    and os.environ.get("GENREC_DISABLE_MFMA", "0") != "1")
         fwd = ops.ext().attn_fwd_mfma if use_mfma else ops.ext().attn_fwd
         out, probs, dmask = fwd(
             q, k, v, bias, key_pad_mask, additive_mask, query_mask,
@@ -90,8 +98,10 @@ class _FusedAttnFn(torch.autograd.Function):
         (scale, causal, act, dropout_p, seed, bias_grad, bias_dim,
          bias_dtype) = ctx.meta
         bwd = ops.ext().attn_bwd_mfma if ctx.use_mfma else ops.ext().attn_bwd
+        if not (ctx.use_mfma and dout.stride(-1) == 1):
+            dout = dout.contiguous()
         dq, dk, dv, dbias = bwd(
-            dout.contiguous(), q, k, v, probs, dmask,
+            dout, q, k, v, probs, dmask,
             query_mask if query_mask.numel() else None,
             scale, act, dropout_p, seed, bias_grad, bias_dim,
         )

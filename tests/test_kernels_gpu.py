@@ -708,3 +708,31 @@ def test_fused_adamw_clip_scale():
                           0.9, 0.999, 1e-8, 0.0)
     # effective grad = 1.0 -> first Adam step is -lr * g/|g| = -1e-2
     assert torch.allclose(master, torch.full_like(master, -1e-2), atol=1e-5)
+
+
+def test_mfma_attention_strided_views_match_contiguous():
+    """[B,L,H,D] transpose views (zero-copy path) vs contiguous inputs."""
+    from genrec_amd.ops.attention import fused_attention
+
+    torch.manual_seed(0)
+    B, L, H, D = 8, 61, 6, 64
+    blhd = torch.randn(B, L, H, 3 * D, device=DEV, dtype=torch.bfloat16)
+    qv = blhd[..., :D].view(B, L, H, D).transpose(1, 2)  # strided everywhere
+    kv = blhd[..., D:2 * D].view(B, L, H, D).transpose(1, 2)
+    vv = blhd[..., 2 * D:].view(B, L, H, D).transpose(1, 2)
+    bias = torch.randn(H, L, L, device=DEV)
+
+    def run(q, k, v):
+        q = q.detach().requires_grad_(True)
+        k = k.detach().requires_grad_(True)
+        v = v.detach().requires_grad_(True)
+        out = fused_attention(q, k, v, scale=0.125, bias=bias, causal=False)
+        out.float().pow(2).sum().backward()
+        return out, q.grad, k.grad, v.grad
+
+    o1, dq1, dk1, dv1 = run(qv, kv, vv)
+    o2, dq2, dk2, dv2 = run(qv.contiguous(), kv.contiguous(),
+                            vv.contiguous())
+    for a, b in [(o1, o2), (dq1, dq2), (dk1, dk2), (dv1, dv2)]:
+        assert torch.allclose(a.float(), b.float(), atol=1e-3, rtol=1e-3), \
+            (a.float() - b.float()).abs().max()
