@@ -37,16 +37,38 @@ def _kernel_available(name: str, *tensors: Tensor) -> bool:
 
 
 _seed_counters = {}
+_call_salt = [0]
 
 
 def _seed_counter(device) -> torch.Tensor:
-    """Per-device dropout seed counter; incremented per call so dropout
-    masks vary across hipGraph replays (the increment is captured too)."""
+    """Per-device dropout seed counter read by the fused kernels. Under
+    hipGraph capture each CALL SITE bakes a distinct python-side salt into
+    its seed, and advance_dropout_seeds() — captured ONCE per step by
+    GraphedTrainStep — bumps this counter so masks vary across replays.
+    (Per-call device increments cost ~40 tiny int-add kernels per TIGER
+    step; one per step is enough for uniqueness.)"""
     t = _seed_counters.get(device)
     if t is None:
         t = torch.zeros(1, dtype=torch.int32, device=device)
         _seed_counters[device] = t
     return t
+
+
+def advance_dropout_seeds(device) -> None:
+    """One captured increment per training step (graph runners call this);
+    eager paths get fresh python-random seeds per call and don't need it."""
+    _seed_counter(device).add_(1)
+
+
+def _call_seed() -> int:
+    """Per-call seed: python-random when eager; a distinct per-call-site
+    salt under capture (the capture-time constant differentiates sites,
+    the device counter differentiates replays)."""
+    _call_salt[0] = (_call_salt[0] + 1) & 0x7FFFFFFF
+    if torch.cuda.is_available() and \
+            torch.cuda.is_current_stream_capturing():
+        return (12345 + _call_salt[0] * 2654435761) & 0x7FFFFFFF
+    return int(torch.randint(0, 2**31 - 1, (1,)).item())
 
 
 class _FusedAttnFn(torch.autograd.Function):
@@ -71,10 +93,8 @@ class _FusedAttnFn(torch.autograd.Function):
         seed_dev = None
         seed = 0
         if dropout_p > 0 and training:
-            seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) \
-                if not torch.cuda.is_current_stream_capturing() else 12345
+            seed = _call_seed()
             seed_dev = _seed_counter(q.device)
-            seed_dev.add_(1)
         fwd = ops.ext().attn_fwd_mfma if use_mfma else ops.ext().attn_fwd
         out, probs, dmask = fwd(
             q, k, v, bias, key_pad_mask, additive_mask, query_mask,

@@ -12,15 +12,11 @@ from torch import Tensor
 
 
 def _seed_args(device, active: bool):
-    from genrec_amd.ops.attention import _seed_counter
+    from genrec_amd.ops.attention import _call_seed, _seed_counter
 
     if not active:
         return 0, None
-    seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) \
-        if not torch.cuda.is_current_stream_capturing() else 54321
-    c = _seed_counter(device)
-    c.add_(1)
-    return seed, c
+    return _call_seed(), _seed_counter(device)
 
 
 class _DropoutAddFn(torch.autograd.Function):

@@ -26,7 +26,9 @@ from typing import Optional
 import torch
 from torch import nn
 
-_SPLITK_MIN_K = 4096
+# below ~8k rows the single hipBLASLt GEMM wins (the chunked path adds a
+# bmm + reduce launch per backward; SASRec/HSTU at K=6400 measured slower)
+_SPLITK_MIN_K = 8192
 
 
 def _pick_chunks(k: int) -> int:
@@ -64,9 +66,10 @@ class _SplitKLinearFn(torch.autograd.Function):
                 part = torch.bmm(
                     dy2.view(nc, k // nc, -1).transpose(1, 2),
                     x2.view(nc, k // nc, -1))
-                # fp32 partial sum preserves the single-GEMM fp32-accum
-                # rounding behavior
-                dw = part.sum(0, dtype=torch.float32).to(weight.dtype)
+                # ATen reduces bf16 sums in fp32 (acc_type), so one
+                # bf16-out sum keeps single-rounding numerics without a
+                # separate cast kernel
+                dw = part.sum(0).to(weight.dtype)
             else:
                 dw = dy2.t().matmul(x2).to(weight.dtype)
         db = None

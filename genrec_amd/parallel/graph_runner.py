@@ -128,6 +128,10 @@ class GraphedTrainStep:
                 self._graph = None
 
     def _inner(self):
+        if self.device.type == "cuda":
+            from genrec_amd.ops.attention import advance_dropout_seeds
+
+            advance_dropout_seeds(self.device)
         self.flat_grads.zero_()
         out = self.model(**self.static)
         loss = self.loss_getter(out)

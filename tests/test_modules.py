@@ -187,10 +187,10 @@ def test_splitk_linear_matches_nn_linear():
             m.weight.copy_(ref.weight)
             if bias:
                 m.bias.copy_(ref.bias)
-        # K=4096 triggers the chunked-bmm grad-weight path
-        x1 = torch.randn(4096, 32, requires_grad=True)
+        # K=8192 triggers the chunked-bmm grad-weight path
+        x1 = torch.randn(8192, 32, requires_grad=True)
         x2 = x1.detach().clone().requires_grad_(True)
-        dy = torch.randn(4096, 48)
+        dy = torch.randn(8192, 48)
         m(x1).backward(dy)
         ref(x2).backward(dy)
         assert torch.allclose(x1.grad, x2.grad, atol=1e-5)
@@ -209,7 +209,7 @@ def test_splitk_linear_autocast_mixed_dtypes():
     from genrec_amd.ops.linear import SplitKLinear
 
     m = SplitKLinear(32, 16, bias=False)  # fp32 params
-    x = torch.randn(4096, 32, requires_grad=True)
+    x = torch.randn(8192, 32, requires_grad=True)
     with torch.amp.autocast("cpu", dtype=torch.bfloat16):
         y = m(x)
     assert y.dtype == torch.bfloat16
