@@ -24,6 +24,7 @@ from __future__ import annotations
 from typing import Optional
 
 import torch
+import torch.nn.functional as F
 from torch import nn
 
 # below ~8k rows the single hipBLASLt GEMM wins (the chunked path adds a
@@ -82,4 +83,8 @@ class SplitKLinear(nn.Linear):
     """nn.Linear with split-K grad-weight GEMMs (see module docstring)."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        # small-K layers take the stock C++ fast path: the custom Function
+        # only pays for itself when the split-K backward will be used
+        if x.numel() // x.shape[-1] < _SPLITK_MIN_K:
+            return F.linear(x, self.weight, self.bias)
         return _SplitKLinearFn.apply(x, self.weight, self.bias)
