@@ -108,16 +108,18 @@ attn_fwd_mfma_kernel(
           &k[(int64_t)b * k_sb + h * k_sh + row * k_sl + d0]);
     }
     *reinterpret_cast<short8v*>(ks + swz(row, d0 * 2)) = val;
-    // V^T tile: vt[d][j] = V[j][d]; this thread owns (d=row, j=d0..d0+7)
-    __hip_bfloat16 tmp[8];
-    for (int jj = 0; jj < 8; ++jj) {
-      int j = d0 + jj;
-      tmp[jj] = (j < Lk && row < D)
-          ? v[(int64_t)b * v_sb + h * v_sh + j * v_sl + row]
-          : __hip_bfloat16(0.f);
+    // V^T tile: vt[d][j] = V[j][d]. Read V row-natural (ONE b128
+    // coalesced load) and scatter the 8 elements into LDS columns —
+    // global-memory access stays coalesced, the transpose happens in LDS.
+    short8v vv = z;
+    if (row < Lk && d0 < D) {
+      vv = *reinterpret_cast<const short8v*>(
+          &v[(int64_t)b * v_sb + h * v_sh + row * v_sl + d0]);
     }
-    *reinterpret_cast<short8v*>(vt + swz(row, d0 * 2)) =
-        *reinterpret_cast<short8v*>(tmp);
+    const __hip_bfloat16* ve = reinterpret_cast<const __hip_bfloat16*>(&vv);
+    for (int t = 0; t < 8; ++t) {
+      *reinterpret_cast<__hip_bfloat16*>(vt + swz(d0 + t, row * 2)) = ve[t];
+    }
   }
   __syncthreads();
 
@@ -317,26 +319,29 @@ attn_bwd_ds_kernel(
           &v[(int64_t)b * v_sb + h * v_sh + row * v_sl + d0]);
     }
     *reinterpret_cast<short8v*>(vs + swz(row, d0 * 2)) = val2;
-    // transposed images: row = d, cols = sequence positions
-    __hip_bfloat16 tk[8], tq[8], td[8];
-    for (int jj = 0; jj < 8; ++jj) {
-      int p = d0 + jj;
-      tk[jj] = (p < Lk && row < D)
-          ? k[(int64_t)b * k_sb + h * k_sh + p * k_sl + row]
-          : __hip_bfloat16(0.f);
-      tq[jj] = (p < Lq && row < D)
-          ? q[(int64_t)b * q_sb + h * q_sh + p * q_sl + row]
-          : __hip_bfloat16(0.f);
-      td[jj] = (p < Lq && row < D)
-          ? dout[(int64_t)b * do_sb + h * do_sh + p * do_sl + row]
-          : __hip_bfloat16(0.f);
+    // transposed images (row = d, cols = sequence positions): read each
+    // source row-natural — ONE b128 coalesced load per tensor — and
+    // scatter the 8 elements down LDS columns; the transpose happens in
+    // LDS, global traffic stays coalesced.
+    short8v kk8 = {}, qq8 = {}, dd8 = {};
+    if (row < Lk && d0 < D) {
+      kk8 = *reinterpret_cast<const short8v*>(
+          &k[(int64_t)b * k_sb + h * k_sh + row * k_sl + d0]);
     }
-    *reinterpret_cast<short8v*>(kt + swz(row, d0 * 2)) =
-        *reinterpret_cast<short8v*>(tk);
-    *reinterpret_cast<short8v*>(qt + swz(row, d0 * 2)) =
-        *reinterpret_cast<short8v*>(tq);
-    *reinterpret_cast<short8v*>(dot + swz(row, d0 * 2)) =
-        *reinterpret_cast<short8v*>(td);
+    if (row < Lq && d0 < D) {
+      qq8 = *reinterpret_cast<const short8v*>(
+          &q[(int64_t)b * q_sb + h * q_sh + row * q_sl + d0]);
+      dd8 = *reinterpret_cast<const short8v*>(
+          &dout[(int64_t)b * do_sb + h * do_sh + row * do_sl + d0]);
+    }
+    const __hip_bfloat16* ke = reinterpret_cast<const __hip_bfloat16*>(&kk8);
+    const __hip_bfloat16* qe = reinterpret_cast<const __hip_bfloat16*>(&qq8);
+    const __hip_bfloat16* de = reinterpret_cast<const __hip_bfloat16*>(&dd8);
+    for (int t = 0; t < 8; ++t) {
+      *reinterpret_cast<__hip_bfloat16*>(kt + swz(d0 + t, row * 2)) = ke[t];
+      *reinterpret_cast<__hip_bfloat16*>(qt + swz(d0 + t, row * 2)) = qe[t];
+      *reinterpret_cast<__hip_bfloat16*>(dot + swz(d0 + t, row * 2)) = de[t];
+    }
   }
   __syncthreads();
 

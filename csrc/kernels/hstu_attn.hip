@@ -92,14 +92,16 @@ hstu_attn_fwd_kernel(
           &k[HIDX4(b, h, row, d0, H, L, D)]);
     }
     *reinterpret_cast<short8vh*>(ks + hswz(row, d0 * 2)) = val2;
-    __hip_bfloat16 tmp[8];
-    for (int jj = 0; jj < 8; ++jj) {
-      int j = d0 + jj;
-      tmp[jj] = (j < L && row < D)
-          ? v[HIDX4(b, h, j, row, H, L, D)] : __hip_bfloat16(0.f);
+    // V^T: ONE b128 coalesced natural-row load, transpose via LDS scatter
+    short8vh vv = {};
+    if (row < L && d0 < D) {
+      vv = *reinterpret_cast<const short8vh*>(
+          &v[HIDX4(b, h, row, d0, H, L, D)]);
     }
-    *reinterpret_cast<short8vh*>(vt + hswz(row, d0 * 2)) =
-        *reinterpret_cast<short8vh*>(tmp);
+    const __hip_bfloat16* ve = reinterpret_cast<const __hip_bfloat16*>(&vv);
+    for (int t = 0; t < 8; ++t) {
+      *reinterpret_cast<__hip_bfloat16*>(vt + hswz(d0 + t, row * 2)) = ve[t];
+    }
   }
   __syncthreads();
 
@@ -218,20 +220,26 @@ hstu_attn_bwd_kernel(
           &v[HIDX4(b, h, row, d0, H, L, D)]);
     }
     *reinterpret_cast<short8vh*>(vs + hswz(row, d0 * 2)) = val2;
-    __hip_bfloat16 tk[8], tq[8], td[8];
-    for (int jj = 0; jj < 8; ++jj) {
-      int p = d0 + jj;
-      bool ok = p < L && row < D;
-      tk[jj] = ok ? k[HIDX4(b, h, p, row, H, L, D)] : __hip_bfloat16(0.f);
-      tq[jj] = ok ? q[HIDX4(b, h, p, row, H, L, D)] : __hip_bfloat16(0.f);
-      td[jj] = ok ? dout[HIDX4(b, h, p, row, H, L, D)] : __hip_bfloat16(0.f);
+    // K^T/Q^T/dO^T: coalesced natural-row b128 loads + LDS-scatter
+    // transpose (dout natural already in `dos`, but re-reading the global
+    // row keeps the loop uniform — it is L2-hot)
+    short8vh kk8 = {}, qq8 = {}, dd8 = {};
+    if (row < L && d0 < D) {
+      kk8 = *reinterpret_cast<const short8vh*>(
+          &k[HIDX4(b, h, row, d0, H, L, D)]);
+      qq8 = *reinterpret_cast<const short8vh*>(
+          &q[HIDX4(b, h, row, d0, H, L, D)]);
+      dd8 = *reinterpret_cast<const short8vh*>(
+          &dout[HIDX4(b, h, row, d0, H, L, D)]);
     }
-    *reinterpret_cast<short8vh*>(kt + hswz(row, d0 * 2)) =
-        *reinterpret_cast<short8vh*>(tk);
-    *reinterpret_cast<short8vh*>(qt + hswz(row, d0 * 2)) =
-        *reinterpret_cast<short8vh*>(tq);
-    *reinterpret_cast<short8vh*>(dot + hswz(row, d0 * 2)) =
-        *reinterpret_cast<short8vh*>(td);
+    const __hip_bfloat16* ke = reinterpret_cast<const __hip_bfloat16*>(&kk8);
+    const __hip_bfloat16* qe = reinterpret_cast<const __hip_bfloat16*>(&qq8);
+    const __hip_bfloat16* de = reinterpret_cast<const __hip_bfloat16*>(&dd8);
+    for (int t = 0; t < 8; ++t) {
+      *reinterpret_cast<__hip_bfloat16*>(kt + hswz(d0 + t, row * 2)) = ke[t];
+      *reinterpret_cast<__hip_bfloat16*>(qt + hswz(d0 + t, row * 2)) = qe[t];
+      *reinterpret_cast<__hip_bfloat16*>(dot + hswz(d0 + t, row * 2)) = de[t];
+    }
   }
   __syncthreads();
 
