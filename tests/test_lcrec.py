@@ -99,3 +99,23 @@ def test_extract_sem_ids():
 
     assert extract_sem_ids("<C0_5><C1_9><C2_0>", 3) == [5, 9, 0]
     assert extract_sem_ids("junk <C1_3>", 3) == [-1, 3, -1]
+
+
+def test_lcrec_trainer_smoke(tmp_path):
+    """End-to-end SFT trainer on a tiny random-init backbone (CPU)."""
+    from genrec_amd.trainers import lcrec_trainer
+
+    lcrec_trainer.train(
+        epochs=1, max_steps=2, batch_size=2, max_seq_len=64,
+        n_codebooks=3, codebook_size=8, backbone_config=dict(TINY),
+        gradient_checkpointing=False, use_lora=False,
+        max_train_samples=8, max_eval_samples=2,
+        do_eval=True, eval_max_batches=1, eval_beam_width=2,
+        save_dir_root=str(tmp_path), save_every_epoch=1,
+        wandb_logging=False, num_workers=0)
+    import os
+
+    # HF-format epoch directory (save_pretrained), reference parity
+    assert os.path.isdir(os.path.join(tmp_path, "epoch_0"))
+    assert os.path.exists(os.path.join(tmp_path, "epoch_0",
+                                       "tokenizer_config.json"))
