@@ -29,8 +29,6 @@ class PrefixKVCache:
     (ref notellm.py:20-41)."""
 
     def __init__(self, cache):
-        from transformers import cache_utils
-
         self.fixed_len = cache.get_seq_length()
         self._base = cache
         self._per_device: Dict[torch.device, object] = {}
