@@ -40,6 +40,26 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ ((row & 7) << 4));
 }
 
+
+// In-register 8x8 bf16 block transpose across the 8 lanes {8g+c : g=0..7}
+// (3 butterfly stages of shfl_xor; validated against a host simulation).
+// Turns a lane's natural row-chunk into a transposed row-chunk so the
+// LDS transpose staging becomes ONE b128 store instead of 8 scattered
+// u16 stores.
+__device__ __forceinline__ void xpose8x8(short (&vals)[8], int g) {
+#pragma unroll
+  for (int m = 1; m < 8; m <<= 1) {
+    short nv[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int t = __shfl_xor((int)vals[e ^ m], 8 * m, 64);
+      nv[e] = ((e & m) != (g & m)) ? (short)t : vals[e];
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) vals[e] = nv[e];
+  }
+}
+
 // load one 16x32 A/B fragment (8 bf16 = 16 B per lane) from a swizzled tile
 __device__ __forceinline__ short8v frag_load(const char* base, int row0,
                                              int k0, int lane) {
@@ -109,16 +129,24 @@ attn_fwd_mfma_kernel(
     }
     *reinterpret_cast<short8v*>(ks + swz(row, d0 * 2)) = val;
     // V^T tile: vt[d][j] = V[j][d]. Read V row-natural (ONE b128
-    // coalesced load) and scatter the 8 elements into LDS columns —
-    // global-memory access stays coalesced, the transpose happens in LDS.
+    // coalesced load), transpose the 8x8 block in-register across the
+    // lane group, store ONE b128 transposed row-chunk.
     short8v vv = z;
     if (row < Lk && d0 < D) {
       vv = *reinterpret_cast<const short8v*>(
           &v[(int64_t)b * v_sb + h * v_sh + row * v_sl + d0]);
     }
-    const __hip_bfloat16* ve = reinterpret_cast<const __hip_bfloat16*>(&vv);
-    for (int t = 0; t < 8; ++t) {
-      *reinterpret_cast<__hip_bfloat16*>(vt + swz(d0 + t, row * 2)) = ve[t];
+    {
+      const int g = (lane >> 3) & 7;
+      short tv[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) tv[e] = vv[e];
+      xpose8x8(tv, g);
+      short8v pack;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) pack[e] = tv[e];
+      // this lane now holds V[j0..j0+7][d0+g] -> vt row d0+g
+      *reinterpret_cast<short8v*>(vt + swz(d0 + g, (row & ~7) * 2)) = pack;
     }
   }
   __syncthreads();
@@ -320,9 +348,8 @@ attn_bwd_ds_kernel(
     }
     *reinterpret_cast<short8v*>(vs + swz(row, d0 * 2)) = val2;
     // transposed images (row = d, cols = sequence positions): read each
-    // source row-natural — ONE b128 coalesced load per tensor — and
-    // scatter the 8 elements down LDS columns; the transpose happens in
-    // LDS, global traffic stays coalesced.
+    // source row-natural — ONE b128 coalesced load per tensor —
+    // transpose 8x8 in-register, store ONE b128 per tensor.
     short8v kk8 = {}, qq8 = {}, dd8 = {};
     if (row < Lk && d0 < D) {
       kk8 = *reinterpret_cast<const short8v*>(
@@ -334,13 +361,23 @@ attn_bwd_ds_kernel(
       dd8 = *reinterpret_cast<const short8v*>(
           &dout[(int64_t)b * do_sb + h * do_sh + row * do_sl + d0]);
     }
-    const __hip_bfloat16* ke = reinterpret_cast<const __hip_bfloat16*>(&kk8);
-    const __hip_bfloat16* qe = reinterpret_cast<const __hip_bfloat16*>(&qq8);
-    const __hip_bfloat16* de = reinterpret_cast<const __hip_bfloat16*>(&dd8);
-    for (int t = 0; t < 8; ++t) {
-      *reinterpret_cast<__hip_bfloat16*>(kt + swz(d0 + t, row * 2)) = ke[t];
-      *reinterpret_cast<__hip_bfloat16*>(qt + swz(d0 + t, row * 2)) = qe[t];
-      *reinterpret_cast<__hip_bfloat16*>(dot + swz(d0 + t, row * 2)) = de[t];
+    {
+      const int g = (lane >> 3) & 7;
+      const int jb = (row & ~7) * 2;
+      short tk[8], tq2[8], td2[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        tk[e] = kk8[e]; tq2[e] = qq8[e]; td2[e] = dd8[e];
+      }
+      xpose8x8(tk, g);
+      xpose8x8(tq2, g);
+      xpose8x8(td2, g);
+      short8v pk, pq, pd;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) { pk[e] = tk[e]; pq[e] = tq2[e]; pd[e] = td2[e]; }
+      *reinterpret_cast<short8v*>(kt + swz(d0 + g, jb)) = pk;
+      *reinterpret_cast<short8v*>(qt + swz(d0 + g, jb)) = pq;
+      *reinterpret_cast<short8v*>(dot + swz(d0 + g, jb)) = pd;
     }
   }
   __syncthreads();

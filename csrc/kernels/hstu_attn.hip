@@ -34,6 +34,21 @@ __device__ __forceinline__ int hswz(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ ((row & 7) << 4));
 }
 
+// 8x8 in-register block transpose (see attention_mfma.hip::xpose8x8)
+__device__ __forceinline__ void hxpose8x8(short (&vals)[8], int g) {
+#pragma unroll
+  for (int m = 1; m < 8; m <<= 1) {
+    short nv[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int t = __shfl_xor((int)vals[e ^ m], 8 * m, 64);
+      nv[e] = ((e & m) != (g & m)) ? (short)t : vals[e];
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) vals[e] = nv[e];
+  }
+}
+
 __device__ __forceinline__ short8vh hfrag(const char* base, int row0, int k0,
                                           int lane) {
   int row = row0 + (lane & 15);
@@ -92,15 +107,22 @@ hstu_attn_fwd_kernel(
           &k[HIDX4(b, h, row, d0, H, L, D)]);
     }
     *reinterpret_cast<short8vh*>(ks + hswz(row, d0 * 2)) = val2;
-    // V^T: ONE b128 coalesced natural-row load, transpose via LDS scatter
+    // V^T: coalesced natural-row load + in-register 8x8 transpose
     short8vh vv = {};
     if (row < L && d0 < D) {
       vv = *reinterpret_cast<const short8vh*>(
           &v[HIDX4(b, h, row, d0, H, L, D)]);
     }
-    const __hip_bfloat16* ve = reinterpret_cast<const __hip_bfloat16*>(&vv);
-    for (int t = 0; t < 8; ++t) {
-      *reinterpret_cast<__hip_bfloat16*>(vt + hswz(d0 + t, row * 2)) = ve[t];
+    {
+      const int g = (lane >> 3) & 7;
+      short tv[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) tv[e] = vv[e];
+      hxpose8x8(tv, g);
+      short8vh pack;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) pack[e] = tv[e];
+      *reinterpret_cast<short8vh*>(vt + hswz(d0 + g, (row & ~7) * 2)) = pack;
     }
   }
   __syncthreads();
@@ -220,9 +242,8 @@ hstu_attn_bwd_kernel(
           &v[HIDX4(b, h, row, d0, H, L, D)]);
     }
     *reinterpret_cast<short8vh*>(vs + hswz(row, d0 * 2)) = val2;
-    // K^T/Q^T/dO^T: coalesced natural-row b128 loads + LDS-scatter
-    // transpose (dout natural already in `dos`, but re-reading the global
-    // row keeps the loop uniform — it is L2-hot)
+    // K^T/Q^T/dO^T: coalesced natural-row b128 loads + in-register 8x8
+    // transpose, one b128 LDS store per tensor
     short8vh kk8 = {}, qq8 = {}, dd8 = {};
     if (row < L && d0 < D) {
       kk8 = *reinterpret_cast<const short8vh*>(
@@ -232,13 +253,23 @@ hstu_attn_bwd_kernel(
       dd8 = *reinterpret_cast<const short8vh*>(
           &dout[HIDX4(b, h, row, d0, H, L, D)]);
     }
-    const __hip_bfloat16* ke = reinterpret_cast<const __hip_bfloat16*>(&kk8);
-    const __hip_bfloat16* qe = reinterpret_cast<const __hip_bfloat16*>(&qq8);
-    const __hip_bfloat16* de = reinterpret_cast<const __hip_bfloat16*>(&dd8);
-    for (int t = 0; t < 8; ++t) {
-      *reinterpret_cast<__hip_bfloat16*>(kt + hswz(d0 + t, row * 2)) = ke[t];
-      *reinterpret_cast<__hip_bfloat16*>(qt + hswz(d0 + t, row * 2)) = qe[t];
-      *reinterpret_cast<__hip_bfloat16*>(dot + hswz(d0 + t, row * 2)) = de[t];
+    {
+      const int g = (lane >> 3) & 7;
+      const int jb = (row & ~7) * 2;
+      short tk[8], tq2[8], td2[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        tk[e] = kk8[e]; tq2[e] = qq8[e]; td2[e] = dd8[e];
+      }
+      hxpose8x8(tk, g);
+      hxpose8x8(tq2, g);
+      hxpose8x8(td2, g);
+      short8vh pk, pq, pd;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) { pk[e] = tk[e]; pq[e] = tq2[e]; pd[e] = td2[e]; }
+      *reinterpret_cast<short8vh*>(kt + hswz(d0 + g, jb)) = pk;
+      *reinterpret_cast<short8vh*>(qt + hswz(d0 + g, jb)) = pq;
+      *reinterpret_cast<short8vh*>(dot + hswz(d0 + g, jb)) = pd;
     }
   }
   __syncthreads();
