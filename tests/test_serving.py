@@ -66,3 +66,32 @@ def test_checkpoint_roundtrip(service, tmp_path):
         device=torch.device("cpu"))
     recs = svc2.recommend_batch([0], [[1, 2, 3]])
     assert len(recs) == 1
+
+
+def test_micro_batcher_fuses_concurrent_requests(service):
+    """Concurrent submits inside the window run as ONE generate call."""
+    import asyncio
+
+    from genrec_amd.serving.server import _MicroBatcher
+
+    calls = []
+    orig = service.recommend_batch
+
+    def counting(users, hists, k):
+        calls.append(len(users))
+        return orig(users, hists, k)
+
+    service.recommend_batch = counting
+    try:
+        batcher = _MicroBatcher(service, max_batch=64, window_ms=50.0)
+
+        async def drive():
+            return await asyncio.gather(*[
+                batcher.submit(u, [1, 2, 3], 5) for u in range(6)])
+
+        results = asyncio.run(drive())
+    finally:
+        service.recommend_batch = orig
+    assert len(results) == 6
+    assert all(len(r) == 5 for r in results)
+    assert calls == [6], calls  # one fused call, not six
