@@ -110,3 +110,55 @@ def test_tiger_resume(tmp_path):
     ck = os.path.join(str(tmp_path), "checkpoint_final.pt")
     assert os.path.exists(ck)
     tiger_trainer.train(resume_path=ck, **kw)
+
+
+def test_rqvae_trainer_iteration_mode(tmp_path):
+    """iterations= (epochs=None) drives the iteration-keyed loop and
+    checkpoint naming (ref rqvae_trainer.py:91-96 mutual exclusion)."""
+    from genrec_amd.data.synthetic import SyntheticItemDataset
+    from genrec_amd.trainers import rqvae_trainer
+
+    class Tiny(SyntheticItemDataset):
+        def __init__(self, **kw):
+            kw.update(num_items=200, dim=16, n_cat_features=0)
+            super().__init__(**kw)
+
+    rqvae_trainer.train(
+        dataset=Tiny, epochs=None, iterations=6, batch_size=64,
+        num_workers=0, vae_input_dim=16, vae_hidden_dims=[8],
+        vae_embed_dim=4, vae_codebook_size=8, kmeans_warmup_samples=100,
+        save_dir_root=str(tmp_path), eval_every=100, save_model_every=100,
+        warmup_epochs=1)
+    assert os.path.exists(os.path.join(str(tmp_path), "checkpoint_final.pt"))
+    import pytest
+
+    with pytest.raises(AssertionError):
+        rqvae_trainer.train(dataset=Tiny, epochs=2, iterations=6)
+
+
+def test_cobra_trainer_resume(tmp_path):
+    """Save -> resume continues from the stored epoch with state intact."""
+    from genrec_amd.data.cobra_synthetic import SyntheticCobraDataset
+    from genrec_amd.trainers import cobra_trainer
+
+    class Tiny(SyntheticCobraDataset):
+        def __init__(self, **kw):
+            kw.update(num_users=30, num_items=40, id_vocab_size=16,
+                      max_text_len=8)
+            super().__init__(**kw)
+
+    kw = dict(dataset=Tiny, batch_size=8, num_workers=0, n_codebooks=3,
+              id_vocab_size=16, d_model=32, decoder_n_layers=1,
+              decoder_num_heads=2, encoder_n_layers=1,
+              encoder_type="light_t5", do_eval=False, amp=False,
+              save_dir_root=str(tmp_path), save_every_epoch=1,
+              wandb_logging=False)
+    cobra_trainer.train(epochs=1, **kw)
+    ck = os.path.join(str(tmp_path), "checkpoint_final.pt")
+    assert os.path.exists(ck)
+    state = torch.load(ck, weights_only=False)
+    assert state["epoch"] == 0
+    # resume: runs epoch 1 only
+    cobra_trainer.train(epochs=2, resume_path=ck, **kw)
+    state2 = torch.load(ck, weights_only=False)
+    assert state2["epoch"] == 1
