@@ -37,8 +37,15 @@ if os.environ.get("GENREC_DEBUG_BUILD", "0") == "1":
     cxx_flags = ["-O1", "-g", "-std=c++17", "-fno-omit-frame-pointer"]
     hip_flags = ["-O1", "-g", "-std=c++17", "-fno-omit-frame-pointer"]
 
+from setuptools import find_packages
+
 setup(
-    name="genrec_amd_ext",
+    name="genrec_amd",
+    version="0.1.0",
+    description=("MI355X-native generative recommendation framework "
+                 "(CDNA4 HIP kernels + PyTorch-ROCm + RCCL)"),
+    packages=find_packages(include=["genrec_amd", "genrec_amd.*"]),
+    python_requires=">=3.10",
     ext_modules=[
         CUDAExtension(
             name="genrec_amd._C",
