@@ -88,14 +88,30 @@ class GraphedTrainStep:
                                          weight_decay=weight_decay)
 
         self.static = {k: v.clone() for k, v in example_batch.items()}
-        # materialize grads once, then re-point them into the flat buffer
+        # materialize grads once to learn which params receive them
         self.loss_getter(self.model(**self.static)).backward()
         self.flat_grads = torch.zeros(n_total, device=device,
                                       dtype=torch.bfloat16)
+        self._views = []
         off = 0
         for p in self.params:
-            p.grad = self.flat_grads[off:off + p.numel()].view_as(p)
+            self._views.append(self.flat_grads[off:off + p.numel()]
+                               .view_as(p))
             off += p.numel()
+        if self.fused:
+            # Leave p.grad unset each step so autograd STEALS the computed
+            # gradient (no AccumulateGrad add kernel per param — ~126 tiny
+            # adds/step on TIGER); one foreach copy then moves the used
+            # grads into the flat buffer. Unused params' flat regions stay
+            # zero from init.
+            self._used = [i for i, p in enumerate(self.params)
+                          if p.grad is not None]
+            self._used_views = [self._views[i] for i in self._used]
+            for p in self.params:
+                p.grad = None
+        else:
+            for p, view in zip(self.params, self._views):
+                p.grad = view
         if not self.fused:
             self.flat_master_grad = torch.zeros(n_total, device=device)
             off = 0
@@ -132,10 +148,18 @@ class GraphedTrainStep:
             from genrec_amd.ops.attention import advance_dropout_seeds
 
             advance_dropout_seeds(self.device)
-        self.flat_grads.zero_()
+        if self.fused:
+            for p in self.params:
+                p.grad = None
+        else:
+            self.flat_grads.zero_()
         out = self.model(**self.static)
         loss = self.loss_getter(out)
         loss.backward()
+        if self.fused:
+            torch._foreach_copy_(
+                self._used_views,
+                [self.params[i].grad for i in self._used])
         if self.world > 1:
             import torch.distributed as dist
 
