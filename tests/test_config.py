@@ -153,3 +153,65 @@ def test_all_shipped_gin_configs_parse():
         ginlite.parse_file(f, substitutions={"split": "beauty"})
         if os.path.basename(f) != "base.gin":
             assert ginlite.get_bindings("train"), f
+
+
+def test_call_reference_and_nested_refs(tmp_path):
+    """@name() call-refs resolve at injection time; refs nested in lists."""
+    from genrec_amd.config import ginlite
+
+    ginlite.clear_config()
+
+    @ginlite.configurable(name="make_thing")
+    def make_thing(val: int = 3):
+        return {"val": val}
+
+    @ginlite.configurable(name="consumer")
+    def consumer(thing=None, factories=None):
+        return thing, factories
+
+    cfg = tmp_path / "c.gin"
+    cfg.write_text(
+        "make_thing.val = 9\n"
+        "consumer.thing = @make_thing()\n"
+        "consumer.factories = [@make_thing, @consumer]\n")
+    ginlite.parse_file(str(cfg))
+    thing, factories = consumer()
+    assert thing == {"val": 9}  # called at injection, bindings applied
+    assert callable(factories[0]) and callable(factories[1])
+
+
+def test_utils_decorators_and_debug_metrics():
+    import torch
+
+    from genrec_amd.modules.utils import (
+        compute_debug_metrics, eval_mode, reset_kv_cache,
+    )
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.lin = torch.nn.Linear(2, 2)
+            self.kv_cache = {"k": 1}
+            self.was_training = None
+
+        def reset_kv_cache(self):
+            self.kv_cache = {}
+
+        @eval_mode
+        def infer(self):
+            self.was_training = self.training
+            return 1
+
+        @reset_kv_cache
+        def gen(self):
+            return dict(self.kv_cache)
+
+    m = M()
+    m.train()
+    assert m.infer() == 1
+    assert m.was_training is False and m.training is True  # restored
+    assert m.gen() == {}  # cache cleared before the call
+
+    mask = torch.tensor([[1, 1, 0], [1, 1, 1]])
+    d = compute_debug_metrics(mask)
+    assert any("seq_length" in k for k in d)
