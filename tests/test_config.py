@@ -183,9 +183,8 @@ def test_call_reference_and_nested_refs(tmp_path):
 def test_utils_decorators_and_debug_metrics():
     import torch
 
-    from genrec_amd.modules.utils import (
-        compute_debug_metrics, eval_mode, reset_kv_cache,
-    )
+    from genrec_amd.modules.utils import compute_debug_metrics, eval_mode
+    from genrec_amd.modules.utils import reset_kv_cache as reset_kv_deco
 
     class M(torch.nn.Module):
         def __init__(self):
@@ -202,7 +201,7 @@ def test_utils_decorators_and_debug_metrics():
             self.was_training = self.training
             return 1
 
-        @reset_kv_cache
+        @reset_kv_deco
         def gen(self):
             return dict(self.kv_cache)
 
