@@ -10,11 +10,14 @@ import torch
 from genrec_amd.data.schemas import SeqData
 
 
-def sasrec_collate_fn(batch: List[Dict], max_seq_len: int = 50) -> Dict:
-    """Left-pad; input = seq[:-1], target = seq[1:] (shifted next-item)."""
+def sasrec_collate_fn(batch: List[Dict], max_seq_len: int = 50,
+                      fixed_length: bool = False) -> Dict:
+    """Left-pad; input = seq[:-1], target = seq[1:] (shifted next-item).
+    fixed_length pads every batch to max_seq_len (hipGraph capture)."""
     histories = [b["history"] for b in batch]
     targets = [b["target"] for b in batch]
-    max_len = min(max(len(h) for h in histories), max_seq_len)
+    max_len = max_seq_len if fixed_length else \
+        min(max(len(h) for h in histories), max_seq_len)
     input_ids, target_ids = [], []
     for history, target in zip(histories, targets):
         if len(history) > max_len:
@@ -45,9 +48,10 @@ def sasrec_eval_collate_fn(batch: List[Dict], max_seq_len: int = 50) -> Dict:
     }
 
 
-def hstu_collate_fn(batch: List[Dict], max_seq_len: int = 50) -> Dict:
+def hstu_collate_fn(batch: List[Dict], max_seq_len: int = 50,
+                    fixed_length: bool = False) -> Dict:
     """SASRec collate + per-position unix timestamps carried through."""
-    out = sasrec_collate_fn(batch, max_seq_len)
+    out = sasrec_collate_fn(batch, max_seq_len, fixed_length=fixed_length)
     L = out["input_ids"].size(1)
     ts = []
     for b in batch:

@@ -70,3 +70,24 @@ def test_trainer_use_hip_graph_flag_is_cpu_safe(tmp_path):
         use_hip_graph=True, num_workers=0,
         save_dir_root=str(tmp_path), wandb_logging=False,
         dataset=None, save_every_epoch=100)
+
+
+def test_sasrec_hstu_use_hip_graph_flag_cpu_safe(tmp_path):
+    # on CPU the flag is a no-op (graph mode requires CUDA); the fixed-
+    # length collate path still produces correct shapes
+    from genrec_amd.data.collate import hstu_collate_fn, sasrec_collate_fn
+    from genrec_amd.trainers import hstu_trainer, sasrec_trainer
+
+    b = [{"history": [1, 2], "target": 3, "timestamps": [10, 20]}]
+    out = sasrec_collate_fn(b, max_seq_len=6, fixed_length=True)
+    assert out["input_ids"].shape == (1, 6)
+    out = hstu_collate_fn(b, max_seq_len=6, fixed_length=True)
+    assert out["timestamps"].shape == (1, 6)
+
+    common = dict(epochs=1, max_steps=2, batch_size=8, max_seq_len=6,
+                  embed_dim=16, num_heads=2, num_blocks=1, do_eval=False,
+                  amp=False, use_hip_graph=True, num_workers=0,
+                  save_dir_root=str(tmp_path), wandb_logging=False,
+                  save_every_epoch=100)
+    sasrec_trainer.train(ffn_dim=16, **common)
+    hstu_trainer.train(**common)
