@@ -29,3 +29,23 @@ def test_bench_json_contract():
     assert d["config"]["global_batch"] == 8
     assert d["config"]["parallelism"] == "dp1"
     assert isinstance(d["ms_per_step"], float)
+
+
+def test_bench_world2_gloo_contract():
+    """The driver's torchrun launch shape: 2 ranks over gloo on CPU; ONE
+    JSON line from rank 0 with n_gpus=2 and the global batch."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29531", "bench.py", "--gpus", "2",
+         "--steps", "1", "--warmup", "0", "--batch-size", "4"],
+        cwd=repo, capture_output=True, text=True, timeout=900, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 8
