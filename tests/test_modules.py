@@ -216,3 +216,55 @@ def test_splitk_linear_autocast_mixed_dtypes():
     y.float().sum().backward()
     assert m.weight.grad is not None and torch.isfinite(m.weight.grad).all()
     assert x.grad is not None
+
+
+def _naive_attention(q, k, v, scale, bias=None, key_pad=None, add_mask=None,
+                     causal=False, query_mask=None, act="softmax"):
+    """Independent hand-rolled formula (reference semantics written from
+    scratch) — anchors ops.eager, which in turn anchors the GPU kernels."""
+    s = torch.einsum("bhid,bhjd->bhij", q.float(), k.float()) * scale
+    if bias is not None:
+        s = s + (bias.unsqueeze(0) if bias.dim() == 3 else bias).float()
+    if causal:
+        i = torch.arange(s.size(2)).unsqueeze(1)
+        j = torch.arange(s.size(3)).unsqueeze(0)
+        s = s.masked_fill((j > i).view(1, 1, *s.shape[2:]), -1e9)
+    if key_pad is not None:
+        s = s.masked_fill(key_pad.view(key_pad.size(0), 1, 1, -1), -1e9)
+    if add_mask is not None:
+        s = s + add_mask.view(1, 1, *add_mask.shape).float()
+    if act == "silu":
+        p = torch.nn.functional.silu(s)
+    else:
+        p = torch.softmax(s, dim=-1)
+    if query_mask is not None:
+        p = p * query_mask.view(query_mask.size(0), 1, -1, 1).float()
+    return torch.einsum("bhij,bhjd->bhid", p, v.float()).to(q.dtype)
+
+
+@pytest.mark.parametrize("case", ["plain", "sasrec", "t5", "silu"])
+def test_eager_attention_matches_naive_formula(case):
+    from genrec_amd.ops import eager
+
+    torch.manual_seed(3)
+    B, H, L, D = 3, 2, 9, 8
+    q, k, v = (torch.randn(B, H, L, D) for _ in range(3))
+    kw, nkw = {}, {}
+    if case == "sasrec":
+        vm = (torch.rand(B, L) > 0.3).float()
+        kw = dict(key_pad_mask=vm == 0, causal=True, query_mask=vm)
+        nkw = dict(key_pad=vm == 0, causal=True, query_mask=vm)
+    elif case == "t5":
+        bias = torch.randn(H, L, L)
+        am = torch.randn(L, L)
+        kp = torch.zeros(B, L, dtype=torch.bool)
+        kp[:, -2:] = True
+        kw = dict(bias=bias, additive_mask=am, key_pad_mask=kp)
+        nkw = dict(bias=bias, add_mask=am, key_pad=kp)
+    elif case == "silu":
+        bias = torch.randn(1, H, L, L)
+        kw = dict(bias=bias, causal=True, score_act="silu")
+        nkw = dict(bias=bias, causal=True, act="silu")
+    out = eager.fused_attention(q, k, v, scale=0.25, **kw)
+    ref = _naive_attention(q, k, v, 0.25, **nkw)
+    assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
