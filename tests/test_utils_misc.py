@@ -1,0 +1,46 @@
+"""Utility coverage: offline tokenizer, light text encoder, profiler."""
+
+import torch
+
+from genrec_amd.modules.encoders import LightT5Encoder
+from genrec_amd.utils.profiling import StepProfiler, roctx_range
+from genrec_amd.utils.tokenizer import build_offline_tokenizer
+
+
+def test_offline_tokenizer_round_trips_and_special_tokens():
+    tok = build_offline_tokenizer()
+    s = "history: item 42, item 7; recommend next"
+    assert tok.decode(tok(s).input_ids) == s
+    n0 = len(tok)
+    added = tok.add_tokens(["<C0_1>", "<C1_2>"], special_tokens=True)
+    assert added == 2 and len(tok) == n0 + 2
+    ids = tok("<C0_1><C1_2>").input_ids
+    assert len(ids) == 2  # special tokens tokenize atomically
+    assert tok.pad_token == tok.eos_token
+
+
+def test_light_t5_encoder_shapes_and_norm():
+    torch.manual_seed(0)
+    enc = LightT5Encoder(n_layers=1, hidden_dim=32, output_dim=16,
+                        num_heads=4, ff_dim=64, vocab_size=100,
+                        max_seq_len=24)
+    enc.eval()
+    x2 = torch.randint(1, 100, (3, 7))
+    out2 = enc(x2)
+    assert out2.shape == (3, 16)
+    assert torch.allclose(out2.norm(dim=-1), torch.ones(3), atol=1e-5)
+    x3 = torch.randint(1, 100, (2, 4, 7))
+    assert enc(x3).shape == (2, 4, 16)
+    # pad tokens must not change the pooled embedding
+    xp = torch.cat([x2, torch.zeros(3, 5, dtype=torch.long)], dim=1)
+    assert torch.allclose(enc(xp), out2, atol=1e-5)
+
+
+def test_step_profiler_and_roctx_noop_on_cpu():
+    prof = StepProfiler(enabled=True, active=2)  # disabled without CUDA
+    for _ in range(3):
+        with prof.step():
+            pass
+    prof.report()
+    with roctx_range("cpu-noop"):
+        pass
