@@ -77,6 +77,18 @@ void fused_adamw(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
                  torch::Tensor v, torch::Tensor out_p, torch::Tensor lr,
                  torch::Tensor scale, torch::Tensor step, double beta1,
                  double beta2, double eps, double weight_decay);
+std::vector<torch::Tensor> attn_fwd_flash(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
+    c10::optional<torch::Tensor> add_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, bool causal, double dropout_p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev);
+std::vector<torch::Tensor> attn_bwd_flash(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor s_saved, torch::Tensor ml,
+    torch::Tensor drop_mask, c10::optional<torch::Tensor> query_mask,
+    double scale, double dropout_p, bool bias_grad, int64_t bias_dim);
 
 }  // namespace genrec
 
@@ -102,4 +114,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
   m.def("fused_adamw", &genrec::fused_adamw,
         "fused flat AdamW step (device lr/scale/step scalars)");
+  m.def("attn_fwd_flash", &genrec::attn_fwd_flash,
+        "flash-tiled attention fwd (staged; GENREC_ATTN_FLASH=1)");
+  m.def("attn_bwd_flash", &genrec::attn_bwd_flash,
+        "flash-tiled attention bwd (staged; GENREC_ATTN_FLASH=1)");
 }

@@ -1,0 +1,593 @@
+// Flash-tiled MFMA attention (bf16, softmax) — Lq/Lk beyond one 64-tile.
+//
+// Same fragment geometry as attention_mfma.hip, plus a running softmax
+// over 64-wide key tiles (forward) and a k-tile-owning backward that
+// loops q-tiles, accumulating dK/dV in registers and dQ via fp32 global
+// atomics. The tile math (running m/l rescale; single-pass backward via
+// the identity dot_i = rowsum(dO * O)) is validated at fragment
+// granularity against autograd in tools/sim_flash_tiles.py (~1e-15).
+//
+// STAGED FOR ROUND 2: compiled and bound, but dispatched only with
+// GENREC_ATTN_FLASH=1 (ops/attention.py) until GPU-validated — the
+// <=64-tile kernels in attention_mfma.hip remain the production path.
+//
+// Backward saves scores S (post-mask) + per-row (m, l) instead of
+// normalized P; P is reconstructed as exp(S - m)/l in the epilogue.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../core/common.h"
+
+namespace genrec {
+
+constexpr int FTILE = 64;
+constexpr float FNEG = -1e9f;
+
+typedef __attribute__((ext_vector_type(8))) short short8f;
+typedef __attribute__((ext_vector_type(4))) float float4f;
+
+#define FIDX(b, h, i, j, H, I, J) \
+  ((((int64_t)(b) * (H) + (h)) * (I) + (i)) * (J) + (j))
+
+__device__ __forceinline__ int fswz(int row, int byte_in_row) {
+  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
+}
+
+__device__ __forceinline__ short8f ffrag(const char* base, int row0, int k0,
+                                         int lane) {
+  int row = row0 + (lane & 15);
+  int byte = (k0 + ((lane >> 4) << 3)) * 2;
+  return *reinterpret_cast<const short8f*>(base + fswz(row, byte));
+}
+
+__device__ __forceinline__ void fxpose8x8(short (&vals)[8], int g) {
+#pragma unroll
+  for (int m = 1; m < 8; m <<= 1) {
+    short nv[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int t = __shfl_xor((int)vals[e ^ m], 8 * m, 64);
+      nv[e] = ((e & m) != (g & m)) ? (short)t : vals[e];
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) vals[e] = nv[e];
+  }
+}
+
+__global__ void __launch_bounds__(256)
+attn_fwd_flash_kernel(
+    const __hip_bfloat16* __restrict__ q,   // [B,H,Lq,D] contiguous
+    const __hip_bfloat16* __restrict__ k,   // [B,H,Lk,D]
+    const __hip_bfloat16* __restrict__ v,   // [B,H,Lk,D]
+    const void* __restrict__ bias,          // null | [H,Lq,Lk] | [B,H,Lq,Lk]
+    const bool* __restrict__ key_pad,       // null | [B,Lk]
+    const float* __restrict__ add_mask,     // null | [Lq,Lk]
+    const float* __restrict__ query_mask,   // null | [B,Lq]
+    __hip_bfloat16* __restrict__ out,       // [B,H,Lq,D]
+    float* __restrict__ s_saved,            // [B,H,Lq,Lk] post-mask scores
+    float* __restrict__ ml_saved,           // [B,H,Lq,2] (m, l)
+    unsigned char* __restrict__ drop_mask,  // null | [B,H,Lq,Lk]
+    const unsigned int* __restrict__ seed_dev,
+    int B, int H, int Lq, int Lk, int D,
+    float scale, int bias_dim, bool bias_bf16, bool causal,
+    float dropout_p, unsigned int seed) {
+  if (seed_dev) seed += *seed_dev;
+  const float* bias_f = reinterpret_cast<const float*>(bias);
+  const __hip_bfloat16* bias_b = reinterpret_cast<const __hip_bfloat16*>(bias);
+  const int bh = blockIdx.x;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.y * FTILE;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* qs = smem;                  // [64][128B] Q rows
+  char* ks = qs + FTILE * 128;      // [64][128B] K rows (per tile)
+  char* vt = ks + FTILE * 128;      // [64(d)][128B(j)] V^T (per tile)
+  char* ps = vt + FTILE * 128;      // [64][128B] P (per tile)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int col_base = lane & 15;
+  const int row_grp = (lane >> 4) << 2;
+  const int strip = wid * 16;
+  const float inv_keep = dropout_p > 0.f ? 1.0f / (1.0f - dropout_p) : 1.0f;
+
+  // ---- stage Q once
+  for (int idx = tid; idx < FTILE * (FTILE / 8); idx += blockDim.x) {
+    int row = idx / (FTILE / 8);
+    int d0 = (idx % (FTILE / 8)) * 8;
+    short8f val = {};
+    int qi = q0 + row;
+    if (qi < Lq && d0 < D) {
+      val = *reinterpret_cast<const short8f*>(
+          &q[FIDX(b, h, qi, d0, H, Lq, D)]);
+    }
+    *reinterpret_cast<short8f*>(qs + fswz(row, d0 * 2)) = val;
+  }
+
+  float m_row[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_row[4] = {0.f, 0.f, 0.f, 0.f};
+  float4f accO[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+  const int nfrag_d = (D + 15) / 16;
+
+  for (int kt0 = 0; kt0 < Lk; kt0 += FTILE) {
+    __syncthreads();  // previous iteration's ps/ks/vt fully consumed
+    // ---- stage K tile + V^T tile
+    for (int idx = tid; idx < FTILE * (FTILE / 8); idx += blockDim.x) {
+      int row = idx / (FTILE / 8);
+      int d0 = (idx % (FTILE / 8)) * 8;
+      int kj = kt0 + row;
+      short8f val = {};
+      if (kj < Lk && d0 < D) {
+        val = *reinterpret_cast<const short8f*>(
+            &k[FIDX(b, h, kj, d0, H, Lk, D)]);
+      }
+      *reinterpret_cast<short8f*>(ks + fswz(row, d0 * 2)) = val;
+      short8f vv = {};
+      if (kj < Lk && d0 < D) {
+        vv = *reinterpret_cast<const short8f*>(
+            &v[FIDX(b, h, kj, d0, H, Lk, D)]);
+      }
+      const int g = (lane >> 3) & 7;
+      short tv[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) tv[e] = vv[e];
+      fxpose8x8(tv, g);
+      short8f pack;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) pack[e] = tv[e];
+      *reinterpret_cast<short8f*>(vt + fswz(d0 + g, (row & ~7) * 2)) = pack;
+    }
+    __syncthreads();
+
+    // ---- S = Q K_t^T
+    float4f acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      for (int kk = 0; kk < D; kk += 32) {
+        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ffrag(qs, strip, kk, lane), ffrag(ks, f * 16, kk, lane), acc[f],
+            0, 0, 0);
+      }
+    }
+
+    // ---- epilogue: masks + running softmax state
+    float s_val[4][4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int i = q0 + strip + row_grp + r;
+        int j = kt0 + f * 16 + col_base;
+        float s = acc[f][r];
+        if (i < Lq && j < Lk) {
+          s *= scale;
+          if (bias_dim) {
+            int64_t bi = (bias_dim == 3) ? ((int64_t)h * Lq + i) * Lk + j
+                                         : FIDX(b, h, i, j, H, Lq, Lk);
+            s += bias_bf16 ? to_f32(bias_b[bi]) : bias_f[bi];
+          }
+          if (causal && j > i) s = FNEG;
+          if (key_pad && key_pad[(int64_t)b * Lk + j]) s = FNEG;
+          if (add_mask) s += add_mask[(int64_t)i * Lk + j];
+          s_saved[FIDX(b, h, i, j, H, Lq, Lk)] = s;
+        } else {
+          s = -INFINITY;
+        }
+        s_val[f][r] = s;
+      }
+    }
+
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float tmax = fmaxf(fmaxf(s_val[0][r], s_val[1][r]),
+                         fmaxf(s_val[2][r], s_val[3][r]));
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
+      float m_new = fmaxf(m_row[r], fmaxf(tmax, -1e30f));
+      float a = __expf(m_row[r] - m_new);
+      float sum = 0.f;
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        float p = (s_val[f][r] == -INFINITY) ? 0.f
+                                             : __expf(s_val[f][r] - m_new);
+        s_val[f][r] = p;  // reuse as unnormalized p
+        sum += p;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        sum += __shfl_xor(sum, off, 64);
+      l_row[r] = l_row[r] * a + sum;
+      m_row[r] = m_new;
+      // rescale accumulated O for this row
+#pragma unroll
+      for (int f = 0; f < 4; ++f) accO[f][r] *= a;
+    }
+
+    // ---- dropout (on the accumulated contribution only) + stash P
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int i = q0 + strip + row_grp + r;
+        int j = kt0 + f * 16 + col_base;
+        float p = s_val[f][r];
+        if (i < Lq && j < Lk && dropout_p > 0.f) {
+          unsigned long long gidx = FIDX(b, h, i, j, H, Lq, Lk);
+          bool keep = (hash_rng(seed, gidx) & 0xFFFFFF) >=
+                      (unsigned int)(dropout_p * 16777216.0f);
+          drop_mask[gidx] = keep;
+          p = keep ? p * inv_keep : 0.f;
+        }
+        int row = strip + row_grp + r;
+        *reinterpret_cast<__hip_bfloat16*>(ps + fswz(row, (j - kt0) * 2)) =
+            __float2bfloat16(p);
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- accO += P V_t
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      if (f >= nfrag_d) break;
+      for (int kk = 0; kk < FTILE; kk += 32) {
+        accO[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ffrag(ps, strip, kk, lane), ffrag(vt, f * 16, kk, lane), accO[f],
+            0, 0, 0);
+      }
+    }
+  }
+
+  // ---- finalize: normalize, query mask, write out + (m, l)
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    if (f >= nfrag_d) break;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = q0 + strip + row_grp + r;
+      int d = f * 16 + col_base;
+      if (i < Lq && d < D) {
+        float inv = (l_row[r] > 0.f) ? 1.0f / l_row[r] : 0.f;
+        float o = accO[f][r] * inv;
+        if (query_mask) o *= query_mask[(int64_t)b * Lq + i];
+        out[FIDX(b, h, i, d, H, Lq, D)] = __float2bfloat16(o);
+      }
+    }
+  }
+  if (col_base == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int i = q0 + strip + row_grp + r;
+      if (i < Lq) {
+        ml_saved[((int64_t)bh * Lq + i) * 2 + 0] = m_row[r];
+        ml_saved[((int64_t)bh * Lq + i) * 2 + 1] = l_row[r];
+      }
+    }
+  }
+}
+
+// Backward: one block per (b, h, k-tile); loops q-tiles. dK/dV accumulate
+// in registers across q-tiles; dQ partials go to an fp32 buffer via
+// atomicAdd (each (i,d) is touched by n_k_tiles blocks).
+__global__ void __launch_bounds__(256)
+attn_bwd_flash_kernel(
+    const __hip_bfloat16* __restrict__ dout,  // [B,H,Lq,D]
+    const __hip_bfloat16* __restrict__ q,
+    const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v,
+    const float* __restrict__ s_saved,        // [B,H,Lq,Lk]
+    const float* __restrict__ ml_saved,       // [B,H,Lq,2]
+    const float* __restrict__ dot_row,        // [B,H,Lq] rowsum(dO*O)
+    const float* __restrict__ query_mask,     // null | [B,Lq]
+    const unsigned char* __restrict__ drop_mask,
+    float* __restrict__ dq_f32,               // [B,H,Lq,D] zero-init
+    __hip_bfloat16* __restrict__ dk_out,
+    __hip_bfloat16* __restrict__ dv_out,
+    float* __restrict__ ds_saved,             // null | [B,H,Lq,Lk]
+    int B, int H, int Lq, int Lk, int D,
+    float scale, float dropout_p) {
+  const int bh = blockIdx.x;
+  const int b = bh / H, h = bh % H;
+  const int kt0 = blockIdx.y * FTILE;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* vs = smem;                  // [64][128B] V_t rows (persistent)
+  char* kt = vs + FTILE * 128;      // [64(d)][128B(j)] K_t^T (persistent)
+  char* dos = kt + FTILE * 128;     // [64][128B] dO rows (per q-tile)
+  char* qt = dos + FTILE * 128;     // [64(d)][128B(i)] Q^T (per q-tile)
+  char* dot = qt + FTILE * 128;     // [64(d)][128B(i)] dO^T (per q-tile)
+  char* dsn = dot + FTILE * 128;    // [64(i)][128B(j)] dS
+  char* dst = dsn + FTILE * 128;    // [64(j)][128B(i)] dS^T
+  char* adt = dst + FTILE * 128;    // [64(j)][128B(i)] A_d^T
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int col_base = lane & 15;
+  const int row_grp = (lane >> 4) << 2;
+  const int strip = wid * 16;
+  const int i0g = strip + row_grp;  // 8B-aligned store base (see mfma bwd)
+  const float inv_keep = dropout_p > 0.f ? 1.0f / (1.0f - dropout_p) : 1.0f;
+  typedef __attribute__((ext_vector_type(4))) short short4f;
+
+  // ---- stage V_t natural + K_t^T once
+  for (int idx = tid; idx < FTILE * (FTILE / 8); idx += blockDim.x) {
+    int row = idx / (FTILE / 8);
+    int d0 = (idx % (FTILE / 8)) * 8;
+    int kj = kt0 + row;
+    short8f val = {};
+    short8f kk8 = {};
+    if (kj < Lk && d0 < D) {
+      val = *reinterpret_cast<const short8f*>(
+          &v[FIDX(b, h, kj, d0, H, Lk, D)]);
+      kk8 = *reinterpret_cast<const short8f*>(
+          &k[FIDX(b, h, kj, d0, H, Lk, D)]);
+    }
+    *reinterpret_cast<short8f*>(vs + fswz(row, d0 * 2)) = val;
+    const int g = (lane >> 3) & 7;
+    short tk[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) tk[e] = kk8[e];
+    fxpose8x8(tk, g);
+    short8f pk;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) pk[e] = tk[e];
+    *reinterpret_cast<short8f*>(kt + fswz(d0 + g, (row & ~7) * 2)) = pk;
+  }
+
+  const int nfrag_d = (D + 15) / 16;
+  float4f acck[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+  float4f accv[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+
+  for (int q0 = 0; q0 < Lq; q0 += FTILE) {
+    __syncthreads();
+    // ---- stage dO natural + dO^T + Q^T for this q-tile
+    for (int idx = tid; idx < FTILE * (FTILE / 8); idx += blockDim.x) {
+      int row = idx / (FTILE / 8);
+      int d0 = (idx % (FTILE / 8)) * 8;
+      int qi = q0 + row;
+      short8f dd8 = {}, qq8 = {};
+      if (qi < Lq && d0 < D) {
+        dd8 = *reinterpret_cast<const short8f*>(
+            &dout[FIDX(b, h, qi, d0, H, Lq, D)]);
+        qq8 = *reinterpret_cast<const short8f*>(
+            &q[FIDX(b, h, qi, d0, H, Lq, D)]);
+      }
+      *reinterpret_cast<short8f*>(dos + fswz(row, d0 * 2)) = dd8;
+      const int g = (lane >> 3) & 7;
+      short td[8], tq[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) { td[e] = dd8[e]; tq[e] = qq8[e]; }
+      fxpose8x8(td, g);
+      fxpose8x8(tq, g);
+      short8f pd, pq;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) { pd[e] = td[e]; pq[e] = tq[e]; }
+      *reinterpret_cast<short8f*>(dot + fswz(d0 + g, (row & ~7) * 2)) = pd;
+      *reinterpret_cast<short8f*>(qt + fswz(d0 + g, (row & ~7) * 2)) = pq;
+    }
+    __syncthreads();
+
+    // ---- dA = dO V_t^T : C rows = q rows (this wave's strip)
+    float4f acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      for (int kk = 0; kk < D; kk += 32) {
+        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ffrag(dos, strip, kk, lane), ffrag(vs, f * 16, kk, lane), acc[f],
+            0, 0, 0);
+      }
+    }
+
+    // ---- epilogue: reconstruct P, dS = P (M*dA - dot_i); stash tiles
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      short4f dpack, apack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int i = q0 + strip + row_grp + r;
+        int j = kt0 + f * 16 + col_base;
+        float dval = 0.f, aval = 0.f;
+        if (i < Lq && j < Lk) {
+          float m = ml_saved[((int64_t)bh * Lq + i) * 2 + 0];
+          float l = ml_saved[((int64_t)bh * Lq + i) * 2 + 1];
+          float s = s_saved[FIDX(b, h, i, j, H, Lq, Lk)];
+          float p = (l > 0.f) ? __expf(s - m) / l : 0.f;
+          float mult = 1.f;
+          if (query_mask) mult *= query_mask[(int64_t)b * Lq + i];
+          if (dropout_p > 0.f) {
+            mult *= drop_mask[FIDX(b, h, i, j, H, Lq, Lk)] ? inv_keep : 0.f;
+          }
+          float dp = acc[f][r] * mult;
+          dval = p * (dp - dot_row[(int64_t)bh * Lq + i]);
+          aval = p * mult;
+          if (ds_saved) ds_saved[FIDX(b, h, i, j, H, Lq, Lk)] = dval;
+        }
+        __hip_bfloat16 dh = __float2bfloat16(dval * scale);
+        __hip_bfloat16 ah = __float2bfloat16(aval);
+        dpack[r] = *reinterpret_cast<short*>(&dh);
+        apack[r] = *reinterpret_cast<short*>(&ah);
+        int row = strip + row_grp + r;
+        *reinterpret_cast<__hip_bfloat16*>(
+            dsn + fswz(row, (f * 16 + col_base) * 2)) = dh;
+      }
+      int jl = f * 16 + col_base;  // tile-local j row of the transposed tiles
+      *reinterpret_cast<short4f*>(dst + fswz(jl, i0g * 2)) = dpack;
+      *reinterpret_cast<short4f*>(adt + fswz(jl, i0g * 2)) = apack;
+    }
+    // dQ uses only this wave's dS rows
+    __builtin_amdgcn_wave_barrier();
+
+    {  // dQ[strip rows of this q-tile] += (scale*dS) @ K_t  (atomic fp32)
+      float4f accq[4] = {{0, 0, 0, 0}, {0, 0, 0, 0},
+                         {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        if (f >= nfrag_d) break;
+        for (int kk = 0; kk < FTILE; kk += 32) {
+          accq[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ffrag(dsn, strip, kk, lane), ffrag(kt, f * 16, kk, lane),
+              accq[f], 0, 0, 0);
+        }
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        if (f >= nfrag_d) break;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int i = q0 + strip + row_grp + r;
+          int d = f * 16 + col_base;
+          if (i < Lq && d < D) {
+            atomicAdd(&dq_f32[FIDX(b, h, i, d, H, Lq, D)], accq[f][r]);
+          }
+        }
+      }
+    }
+    __syncthreads();  // dst/adt complete across waves
+
+    {  // dK_t += (scale*dS)^T Q ; dV_t += A_d^T dO  (accumulate over q0)
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        if (f >= nfrag_d) break;
+        for (int kk = 0; kk < FTILE; kk += 32) {
+          acck[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ffrag(dst, strip, kk, lane), ffrag(qt, f * 16, kk, lane),
+              acck[f], 0, 0, 0);
+          accv[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ffrag(adt, strip, kk, lane), ffrag(dot, f * 16, kk, lane),
+              accv[f], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // ---- write dK/dV rows of this k-tile
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    if (f >= nfrag_d) break;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int j = kt0 + strip + row_grp + r;
+      int d = f * 16 + col_base;
+      if (j < Lk && d < D) {
+        dk_out[FIDX(b, h, j, d, H, Lk, D)] = __float2bfloat16(acck[f][r]);
+        dv_out[FIDX(b, h, j, d, H, Lk, D)] = __float2bfloat16(accv[f][r]);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------ hosts
+
+std::vector<torch::Tensor> attn_fwd_flash(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
+    c10::optional<torch::Tensor> add_mask,
+    c10::optional<torch::Tensor> query_mask,
+    double scale, bool causal, double dropout_p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev) {
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 && q.is_contiguous() &&
+              k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
+  const int Lk = k.size(2);
+  TORCH_CHECK(D <= FTILE && D % 32 == 0);
+  auto out = torch::empty_like(q);
+  auto opts_f = q.options().dtype(torch::kFloat32);
+  auto s_saved = torch::empty({B, H, Lq, Lk}, opts_f);
+  auto ml = torch::empty({B, H, Lq, 2}, opts_f);
+  torch::Tensor dmask;
+  if (dropout_p > 0) {
+    dmask = torch::empty({B, H, Lq, Lk}, q.options().dtype(torch::kUInt8));
+  } else {
+    dmask = torch::empty({0}, q.options().dtype(torch::kUInt8));
+  }
+  torch::Tensor bias_c;
+  int bias_dim = 0;
+  bool bias_bf16 = false;
+  if (bias.has_value()) {
+    bias_c = bias->contiguous();
+    bias_dim = bias_c.dim();
+    bias_bf16 = bias_c.scalar_type() == torch::kBFloat16;
+  }
+  torch::Tensor am_f, qm_f;
+  if (add_mask.has_value()) am_f = add_mask->to(torch::kFloat32).contiguous();
+  if (query_mask.has_value())
+    qm_f = query_mask->to(torch::kFloat32).contiguous();
+  dim3 block(256);
+  dim3 grid(B * H, (Lq + FTILE - 1) / FTILE);
+  size_t smem = 4 * FTILE * 128;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(attn_fwd_flash_kernel, grid, block, smem, stream,
+      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+      bias_dim ? bias_c.data_ptr() : nullptr,
+      key_pad.has_value() ? key_pad->data_ptr<bool>() : nullptr,
+      add_mask.has_value() ? am_f.data_ptr<float>() : nullptr,
+      query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,
+      reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+      s_saved.data_ptr<float>(), ml.data_ptr<float>(),
+      dropout_p > 0 ? dmask.data_ptr<unsigned char>() : nullptr,
+      seed_dev.has_value()
+          ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())
+          : nullptr,
+      B, H, Lq, Lk, D, (float)scale, bias_dim, bias_bf16, causal,
+      (float)dropout_p, (unsigned int)seed);
+  return {out, s_saved, ml, dmask};
+}
+
+std::vector<torch::Tensor> attn_bwd_flash(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor s_saved, torch::Tensor ml,
+    torch::Tensor drop_mask, c10::optional<torch::Tensor> query_mask,
+    double scale, double dropout_p, bool bias_grad, int64_t bias_dim) {
+  const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
+  const int Lk = k.size(2);
+  dout = dout.contiguous();
+  // flash identity: dot_i = rowsum(dO * O) (tools/sim_flash_tiles.py)
+  auto dot_row = (dout.to(torch::kFloat32) * out.to(torch::kFloat32))
+                     .sum(-1).contiguous();
+  auto dq_f32 = torch::zeros({B, H, Lq, D},
+                             q.options().dtype(torch::kFloat32));
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  torch::Tensor ds_saved;
+  float* ds_ptr = nullptr;
+  if (bias_grad) {
+    ds_saved = torch::empty({B, H, Lq, Lk},
+                            q.options().dtype(torch::kFloat32));
+    ds_ptr = ds_saved.data_ptr<float>();
+  }
+  torch::Tensor qm_f;
+  if (query_mask.has_value())
+    qm_f = query_mask->to(torch::kFloat32).contiguous();
+  dim3 block(256);
+  dim3 grid(B * H, (Lk + FTILE - 1) / FTILE);
+  size_t smem = 8 * FTILE * 128;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(attn_bwd_flash_kernel, grid, block, smem, stream,
+      reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+      s_saved.data_ptr<float>(), ml.data_ptr<float>(),
+      dot_row.data_ptr<float>(),
+      query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,
+      dropout_p > 0 ? drop_mask.data_ptr<unsigned char>() : nullptr,
+      dq_f32.data_ptr<float>(),
+      reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),
+      reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()), ds_ptr,
+      B, H, Lq, Lk, D, (float)scale, (float)dropout_p);
+  auto dq = dq_f32.to(torch::kBFloat16);
+  torch::Tensor dbias;
+  if (bias_grad) {
+    dbias = (bias_dim == 3) ? ds_saved.sum(0) : ds_saved;
+  } else {
+    dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
+  }
+  return {dq, dk, dv, dbias};
+}
+
+}  // namespace genrec

@@ -131,6 +131,50 @@ class _FusedAttnFn(torch.autograd.Function):
                 None, None, None, None, None, None, None, None)
 
 
+class _FlashAttnFn(torch.autograd.Function):
+    """Flash-tiled attention (Lk/Lq beyond one 64-tile). STAGED: compiled
+    + bound but dispatched only with GENREC_ATTN_FLASH=1 until
+    GPU-validated (tile math proven in tools/sim_flash_tiles.py)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, key_pad_mask, additive_mask, query_mask,
+                scale, causal, dropout_p, training):
+        from genrec_amd import ops
+
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        seed_dev = None
+        seed = 0
+        if dropout_p > 0 and training:
+            seed = _call_seed()
+            seed_dev = _seed_counter(q.device)
+        out, s_saved, ml, dmask = ops.ext().attn_fwd_flash(
+            q, k, v, bias, key_pad_mask, additive_mask, query_mask,
+            scale, causal, dropout_p if training else 0.0, seed, seed_dev)
+        ctx.save_for_backward(
+            q, k, v, out, s_saved, ml, dmask,
+            query_mask if query_mask is not None else torch.empty(0))
+        ctx.meta = (scale, dropout_p if training else 0.0,
+                    bias is not None and bias.requires_grad,
+                    bias.dim() if bias is not None else 0,
+                    bias.dtype if bias is not None else None)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from genrec_amd import ops
+
+        q, k, v, out, s_saved, ml, dmask, query_mask = ctx.saved_tensors
+        scale, dropout_p, bias_grad, bias_dim, bias_dtype = ctx.meta
+        dq, dk, dv, dbias = ops.ext().attn_bwd_flash(
+            dout.contiguous(), q, k, v, out, s_saved, ml, dmask,
+            query_mask if query_mask.numel() else None,
+            scale, dropout_p, bias_grad, bias_dim)
+        if bias_grad and dbias.dtype != bias_dtype:
+            dbias = dbias.to(bias_dtype)
+        return (dq, dk, dv, dbias if bias_grad else None,
+                None, None, None, None, None, None, None)
+
+
 def fused_attention(
     q: Tensor,
     k: Tensor,
@@ -157,6 +201,18 @@ def fused_attention(
         qm = query_mask.contiguous() if query_mask is not None else None
         return _FusedAttnFn.apply(q, k, v, b, kp, am, qm,
                                   scale, causal, act, dropout_p, training)
+    flash_ok = (os.environ.get("GENREC_ATTN_FLASH", "0") == "1"
+                and act == _ACT_SOFTMAX and q.dtype == torch.bfloat16
+                and q.size(3) % 32 == 0 and q.size(3) <= 64
+                and (additive_mask is None or additive_mask.dim() == 2)
+                and _kernel_available("attn_fwd_flash", q, k, v))
+    if flash_ok:
+        b = bias.contiguous() if bias is not None else None
+        kp = key_pad_mask.contiguous() if key_pad_mask is not None else None
+        am = additive_mask.contiguous() if additive_mask is not None else None
+        qm = query_mask.contiguous() if query_mask is not None else None
+        return _FlashAttnFn.apply(q, k, v, b, kp, am, qm,
+                                  scale, causal, dropout_p, training)
     return eager.fused_attention(
         q, k, v, scale=scale, bias=bias, key_pad_mask=key_pad_mask,
         additive_mask=additive_mask, causal=causal, query_mask=query_mask,

@@ -27,6 +27,7 @@ sources = [
     "csrc/kernels/embedding.hip",
     "csrc/kernels/fused_elementwise.hip",
     "csrc/kernels/adamw.hip",
+    "csrc/kernels/attention_flash.hip",
 ]
 
 cxx_flags = ["-O3", "-std=c++17"]

@@ -749,3 +749,70 @@ def test_mfma_attention_strided_views_match_contiguous():
     for a, b in [(o1, o2), (dq1, dq2), (dk1, dk2), (dv1, dv2)]:
         assert torch.allclose(a.float(), b.float(), atol=1e-3, rtol=1e-3), \
             (a.float() - b.float()).abs().max()
+
+
+@pytest.mark.skipif(os.environ.get("GENREC_ATTN_FLASH") != "1",
+                    reason="staged flash path; enable GENREC_ATTN_FLASH=1")
+@pytest.mark.parametrize("case", ["plain128", "bias80", "causal128",
+                                  "padmask96", "dropout128"])
+def test_flash_attention_vs_eager(case):
+    """Flash-tiled kernels (Lk>64) vs the eager fp32 reference. Skipped
+    until the staged path is enabled (BACKLOG.md item 3)."""
+    from genrec_amd.ops.attention import fused_attention
+
+    torch.manual_seed(0)
+    B, H, D = 4, 3, 64
+    Lq = {"plain128": 100, "bias80": 80, "causal128": 128,
+          "padmask96": 61, "dropout128": 90}[case]
+    Lk = {"plain128": 128, "bias80": 80, "causal128": 128,
+          "padmask96": 96, "dropout128": 128}[case]
+    q = torch.randn(B, H, Lq, D, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, H, Lk, D, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, H, Lk, D, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    kw = {}
+    if case == "bias80":
+        kw["bias"] = torch.randn(H, Lq, Lk, device=DEV, requires_grad=True)
+    if case == "causal128":
+        kw["causal"] = True
+    if case == "padmask96":
+        kp = torch.zeros(B, Lk, dtype=torch.bool, device=DEV)
+        kp[:, -20:] = True
+        kw["key_pad_mask"] = kp
+    drop = 0.3 if case == "dropout128" else 0.0
+
+    out = fused_attention(q, k, v, scale=0.125, dropout_p=drop,
+                          training=drop > 0, **kw)
+    if drop > 0:
+        assert torch.isfinite(out.float()).all()
+        out.float().sum().backward()
+        assert torch.isfinite(q.grad.float()).all()
+        return
+    # fp32 eager reference on the same inputs
+    from genrec_amd.ops import eager
+
+    ref = eager.fused_attention(
+        q.detach().float(), k.detach().float(), v.detach().float(),
+        scale=0.125,
+        **{kk: (vv.detach().float() if torch.is_tensor(vv) and
+                vv.dtype.is_floating_point else vv)
+           for kk, vv in kw.items()})
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2), \
+        (out.float() - ref).abs().max()
+    g = torch.randn_like(ref)
+    out.backward(g.to(torch.bfloat16))
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    kw2 = {kk: (vv.detach().float().requires_grad_(True)
+                if torch.is_tensor(vv) and vv.dtype.is_floating_point
+                else vv) for kk, vv in kw.items()}
+    eager.fused_attention(q2, k2, v2, scale=0.125, **kw2).backward(g)
+    for a, b2 in ((q.grad, q2.grad), (k.grad, k2.grad), (v.grad, v2.grad)):
+        assert torch.allclose(a.float(), b2, atol=8e-2, rtol=8e-2), \
+            (a.float() - b2).abs().max()
+    if case == "bias80":
+        assert torch.allclose(kw["bias"].grad, kw2["bias"].grad,
+                              atol=8e-2, rtol=8e-2)
