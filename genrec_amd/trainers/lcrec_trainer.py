@@ -149,7 +149,8 @@ def evaluate_aux_tasks(model: LCRec, metas, device,
 def evaluate(model: LCRec, loader, device, helper: ConstrainedDecodingHelper,
              n_codebooks: int, ks=(1, 5, 10), beam_width: int = 10,
              max_batches: Optional[int] = None,
-             aux_tasks: bool = True) -> Dict[str, float]:
+             aux_tasks: bool = True,
+             debug_logging: bool = False) -> Dict[str, float]:
     model.eval()
     acc = TopKAccumulator(ks=list(ks))
     allowed = helper.allowed_token_ids()
@@ -177,6 +178,11 @@ def evaluate(model: LCRec, loader, device, helper: ConstrainedDecodingHelper,
             while len(beams) < max(ks):
                 beams.append([-1] * n_codebooks)
             preds.append(beams)
+        if debug_logging and bi == 0:
+            # target-vs-pred sample dump (ref lcrec_trainer.py:176-178)
+            for b in range(min(3, len(tgts))):
+                logger.info("debug sample %d: target=%s top1=%s", b,
+                            tgts[b], preds[b][0])
         acc.accumulate(torch.tensor(tgts, device=device),
                        torch.tensor(preds, device=device))
     metrics = acc.reduce(all_reduce=True)
@@ -226,7 +232,22 @@ def train(
     max_steps: Optional[int] = None,
     num_workers: int = 0,
     eval_max_batches: Optional[int] = None,
+    # ---- reference-config drop-in aliases (config/lcrec/amazon/lcrec.gin
+    # of the reference binds these names)
+    max_length: Optional[int] = None,          # == max_seq_len
+    num_codebooks: Optional[int] = None,       # == n_codebooks
+    warmup_ratio: Optional[float] = None,      # fraction -> num_warmup_steps
+    eval_batch_size: Optional[int] = None,
+    amp: bool = True,                          # False -> fp32 eval/train
+    debug_logging: bool = False,               # print target-vs-pred samples
+    max_text_len: Optional[int] = None,        # forwarded to the dataset
 ):
+    if max_length is not None:
+        max_seq_len = max_length
+    if num_codebooks is not None:
+        n_codebooks = num_codebooks
+    if not amp:
+        mixed_precision_type = None
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "lcrec")
     common.set_seed(seed, ctx.rank)
@@ -256,7 +277,7 @@ def train(
     ds_cls = dataset or SyntheticLCRecDataset
     mk = lambda mode, cap: _make_dataset(
         ds_cls, mode, dataset_folder, split, n_codebooks, codebook_size,
-        pretrained_rqvae_path, cap)
+        pretrained_rqvae_path, cap, max_text_len=max_text_len)
     train_ds = mk("train", max_train_samples)
     valid_ds = mk("valid", max_eval_samples)
 
@@ -266,8 +287,10 @@ def train(
     train_loader = common.make_loader(train_ds, batch_size, ctx, True, coll,
                                       num_workers=num_workers, seed=seed,
                                       drop_last=True)
-    valid_loader = common.make_loader(valid_ds, batch_size, ctx, False,
-                                      gcoll, num_workers=num_workers)
+    valid_loader = common.make_loader(valid_ds,
+                                      eval_batch_size or batch_size,
+                                      ctx, False, gcoll,
+                                      num_workers=num_workers)
 
     wb = common.init_wandb(wandb_project, {"model": "lcrec"},
                            wandb_logging, ctx.is_main)
@@ -276,7 +299,8 @@ def train(
     if eval_only:
         metrics = evaluate(model, valid_loader, device, helper, n_codebooks,
                            beam_width=eval_beam_width,
-                           max_batches=eval_max_batches)
+                           max_batches=eval_max_batches,
+                           debug_logging=debug_logging)
         if ctx.is_main:
             logger.info("eval-only %s", metrics)
         wb.finish()
@@ -285,6 +309,9 @@ def train(
     opt = torch.optim.AdamW(model.parameters(), lr=learning_rate,
                             weight_decay=weight_decay)
     steps_per_epoch = max(1, len(train_loader) // gradient_accumulate_every)
+    if warmup_ratio is not None:
+        num_warmup_steps = max(1, int(warmup_ratio * steps_per_epoch
+                                      * epochs))
     sched = get_cosine_schedule_with_warmup(opt, num_warmup_steps,
                                             steps_per_epoch * epochs)
     # LCRec-scale grads are bandwidth-bound: >=100 MB buckets
@@ -319,7 +346,8 @@ def train(
         if do_eval and (epoch + 1) % eval_every_epoch == 0:
             metrics = evaluate(model, valid_loader, device, helper,
                                n_codebooks, beam_width=eval_beam_width,
-                               max_batches=eval_max_batches)
+                               max_batches=eval_max_batches,
+                               debug_logging=debug_logging)
             if ctx.is_main:
                 logger.info("epoch %d eval %s", epoch, metrics)
                 wb.log({f"eval/{k}": v for k, v in metrics.items()})
@@ -333,7 +361,7 @@ def train(
 
 
 def _make_dataset(ds_cls, mode, folder, split, n_codebooks, codebook_size,
-                  rqvae_path, max_samples):
+                  rqvae_path, max_samples, max_text_len=None):
     import inspect
 
     sig = inspect.signature(ds_cls.__init__)
@@ -342,10 +370,14 @@ def _make_dataset(ds_cls, mode, folder, split, n_codebooks, codebook_size,
                   max_samples=max_samples)
         if rqvae_path:
             kw["pretrained_rqvae_path"] = rqvae_path
+        if max_text_len is not None and "max_text_len" in sig.parameters:
+            kw["max_text_len"] = max_text_len
         return ds_cls(**kw)
-    return ds_cls(**common.dataset_kwargs(ds_cls, {
-        "split": mode, "sem_id_dim": n_codebooks,
-        "codebook_size": codebook_size, "max_samples": max_samples}))
+    extra = {"split": mode, "sem_id_dim": n_codebooks,
+             "codebook_size": codebook_size, "max_samples": max_samples}
+    if max_text_len is not None:
+        extra["max_text_len"] = max_text_len
+    return ds_cls(**common.dataset_kwargs(ds_cls, extra))
 
 
 if __name__ == "__main__":

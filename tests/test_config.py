@@ -214,3 +214,36 @@ def test_utils_decorators_and_debug_metrics():
     mask = torch.tensor([[1, 1, 0], [1, 1, 1]])
     d = compute_debug_metrics(mask)
     assert any("seq_length" in k for k in d)
+
+
+def test_reference_gin_binding_names_are_drop_in():
+    """Every `train.X` binding in the REFERENCE's shipped configs must be
+    a parameter of the corresponding trainer here (config drop-in)."""
+    import inspect
+    import re
+
+    from genrec_amd.trainers import (hstu_trainer, lcrec_trainer,
+                                     rqvae_trainer, sasrec_trainer,
+                                     tiger_trainer)
+
+    ref = "/root/reference/config"
+    if not os.path.isdir(ref):
+        import pytest
+
+        pytest.skip("reference configs not mounted")
+    cases = {
+        sasrec_trainer: [f"{ref}/sasrec/amazon.gin"],
+        hstu_trainer: [f"{ref}/hstu/amazon.gin"],
+        rqvae_trainer: [f"{ref}/tiger/amazon/rqvae.gin",
+                        f"{ref}/lcrec/amazon/rqvae.gin"],
+        tiger_trainer: [f"{ref}/tiger/amazon/tiger.gin"],
+        lcrec_trainer: [f"{ref}/lcrec/amazon/lcrec.gin",
+                        f"{ref}/lcrec/amazon/lcrec_debug.gin"],
+    }
+    for mod, files in cases.items():
+        params = set(inspect.signature(mod.train.__wrapped__).parameters)
+        for f in files:
+            for line in open(f):
+                m = re.match(r"\s*train\.(\w+)\s*=", line)
+                assert not m or m.group(1) in params, \
+                    f"{mod.__name__} missing param {m.group(1)} ({f})"
