@@ -162,3 +162,32 @@ def test_cobra_trainer_resume(tmp_path):
     cobra_trainer.train(epochs=2, resume_path=ck, **kw)
     state2 = torch.load(ck, weights_only=False)
     assert state2["epoch"] == 1
+
+
+def test_trainer_cli_end_to_end(tmp_path):
+    """The documented CLI: python -m genrec_amd.trainers.tiger_trainer
+    <cfg.gin> --gin overrides (ref modules/utils.py:85-117 UX)."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "-m", "genrec_amd.trainers.tiger_trainer",
+         "config/tiger/synthetic/tiger.gin",
+         "--gin", "train.max_steps=2",
+         "--gin", "train.epochs=1",
+         "--gin", "train.batch_size=8",
+         "--gin", "train.n_layers=1",
+         "--gin", "train.num_heads=2",
+         "--gin", "train.attn_dim=32",
+         "--gin", "train.embedding_dim=16",
+         "--gin", "train.do_eval=False",
+         "--gin", "train.num_workers=0",
+         "--gin", "train.amp=False",
+         "--gin", f"train.save_dir_root=\"{tmp_path}\"",
+         "--gin", "SyntheticSemIdSeqDataset.num_users=40",
+         "--gin", "SyntheticSemIdSeqDataset.num_items=60"],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert os.path.exists(os.path.join(str(tmp_path),
+                                       "checkpoint_final.pt"))
