@@ -96,11 +96,6 @@ def main() -> None:
         for i in range(8)
     ]
 
-    # cache_enabled=False: required under hipGraph capture — autocast's
-    # weight-cast cache frees graph-pool blocks the graph still references.
-    amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
-                         cache_enabled=False) if use_gpu else _null_ctx()
-
     # hipGraph capture of the whole train step (fwd+bwd+allreduce+clip+AdamW)
     # removes per-launch host overhead — the step is ~1000 tiny dispatches.
     # On capture failure (e.g. an RCCL build without graph support) the SAME
@@ -138,8 +133,7 @@ def main() -> None:
         def step(i: int) -> None:
             b = batches[i % len(batches)]
             opt.zero_grad(set_to_none=False)
-            with amp:
-                out = model(**b)
+            out = model(**b)
             out.loss.backward()
             reducer.finalize()
             torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
@@ -195,14 +189,6 @@ def main() -> None:
                 "hip_graph": bool(use_graph),
             },
         }))
-
-
-class _null_ctx:
-    def __enter__(self):
-        return self
-
-    def __exit__(self, *a):
-        return False
 
 
 if __name__ == "__main__":
