@@ -128,5 +128,65 @@ def main():
           "bwd via dot=rowsum(dO*O))")
 
 
+def check_multipliers():
+    """query-mask + dropout multipliers: out = qm * (drop ∘ softmax(S)) @ V.
+    Verifies (a) dropout applied to the accumulated contribution only while
+    l sums UNDROPPED p, and (b) the identity dot_i = rowsum(dO*O) still
+    equals sum_j P dP with dP = M ∘ dA — the exact epilogue math of
+    csrc/kernels/attention_flash.hip."""
+    import torch
+
+    np.random.seed(1)
+    Lq, Lk, D, scale, tile = 61, 128, 32, 0.2, 64
+    q, k, v = (np.random.randn(*sh) for sh in
+               ((Lq, D), (Lk, D), (Lk, D)))
+    qm = (np.random.rand(Lq) > 0.2).astype(float)
+    keep = np.random.rand(Lq, Lk) > 0.3
+    inv_keep = 1 / 0.7
+    M = qm[:, None] * np.where(keep, inv_keep, 0.0)
+    dout = np.random.randn(Lq, D)
+
+    tq = torch.tensor(q, requires_grad=True)
+    tk = torch.tensor(k, requires_grad=True)
+    tv = torch.tensor(v, requires_grad=True)
+    P = torch.softmax(tq @ tk.T * scale, 1)
+    O = (torch.tensor(M) * P) @ tv
+    O.backward(torch.tensor(dout))
+
+    m = np.full(Lq, -1e30)
+    l = np.zeros(Lq)
+    acc = np.zeros((Lq, D))
+    s_saved = np.zeros((Lq, Lk))
+    for k0 in range(0, Lk, tile):
+        s = q @ k[k0:k0 + tile].T * scale
+        s_saved[:, k0:k0 + tile] = s
+        m_new = np.maximum(m, s.max(1))
+        a = np.exp(m - m_new)
+        p = np.exp(s - m_new[:, None])
+        l = l * a + p.sum(1)
+        p_drop = p * np.where(keep[:, k0:k0 + tile], inv_keep, 0.0)
+        acc = acc * a[:, None] + p_drop @ v[k0:k0 + tile]
+        m = m_new
+    out = acc / l[:, None] * qm[:, None]
+    assert np.abs(out - O.detach().numpy()).max() < 1e-12
+
+    dot = (dout * out).sum(1)
+    dq = np.zeros_like(q)
+    dk = np.zeros_like(k)
+    dv = np.zeros_like(v)
+    for k0 in range(0, Lk, tile):
+        p = np.exp(s_saved[:, k0:k0 + tile] - m[:, None]) / l[:, None]
+        Mt = M[:, k0:k0 + tile]
+        dp = (dout @ v[k0:k0 + tile].T) * Mt
+        ds = p * (dp - dot[:, None])
+        dq += scale * ds @ k[k0:k0 + tile]
+        dk[k0:k0 + tile] = scale * ds.T @ q
+        dv[k0:k0 + tile] = (Mt * p).T @ dout
+    for a2, b2 in ((dq, tq.grad), (dk, tk.grad), (dv, tv.grad)):
+        assert np.abs(a2 - b2.numpy()).max() < 1e-12
+    print("multiplier (query-mask + dropout) variant validated")
+
+
 if __name__ == "__main__":
     main()
+    check_multipliers()
