@@ -311,9 +311,17 @@ class Tiger(nn.Module):
         return TigerGenerationOutput(sem_ids=beam_seqs, log_probas=beam_logps)
 
     def load_pretrained(self, path: str) -> None:
+        """Load weights from a safetensors dir/file (ref tiger.py:248-253)
+        or a reference-layout dict checkpoint (.pt)."""
         import os
 
-        from safetensors.torch import load_file
+        if os.path.isdir(path) or path.endswith(".safetensors"):
+            from safetensors.torch import load_file
 
-        state = load_file(os.path.join(path, "model.safetensors"))
+            f = path if path.endswith(".safetensors") else \
+                os.path.join(path, "model.safetensors")
+            state = load_file(f)
+        else:
+            obj = torch.load(path, map_location="cpu", weights_only=False)
+            state = obj.get("model", obj) if isinstance(obj, dict) else obj
         self.load_state_dict(state, strict=True)

@@ -204,3 +204,31 @@ def test_rqvae_reference_checkpoint_layout(tmp_path):
     m2.load_pretrained(path)
     for p1, p2 in zip(m.parameters(), m2.parameters()):
         assert torch.equal(p1, p2)
+
+
+def test_tiger_load_pretrained_both_formats(tmp_path):
+    """safetensors (ref tiger.py:248-253) AND dict checkpoints load."""
+    import os
+
+    from safetensors.torch import save_file
+
+    from genrec_amd.models.tiger import Tiger
+
+    torch.manual_seed(0)
+    kw = dict(embedding_dim=16, attn_dim=32, dropout=0.0, num_heads=2,
+              n_layers=1, num_item_embeddings=8, num_user_embeddings=50,
+              sem_id_dim=3)
+    m1 = Tiger(**kw)
+    st_dir = tmp_path / "hf"
+    os.makedirs(st_dir)
+    save_file({k: v.contiguous() for k, v in m1.state_dict().items()},
+              str(st_dir / "model.safetensors"))
+    torch.save({"model": m1.state_dict(), "epoch": 0},
+               str(tmp_path / "ck.pt"))
+
+    m2 = Tiger(**kw)
+    m2.load_pretrained(str(st_dir))
+    m3 = Tiger(**kw)
+    m3.load_pretrained(str(tmp_path / "ck.pt"))
+    for a, b, c in zip(m1.parameters(), m2.parameters(), m3.parameters()):
+        assert torch.equal(a, b) and torch.equal(a, c)
