@@ -84,7 +84,13 @@ def tiger_pad_collate(batch: List[SeqData], pad_id: int = 0,
     places ids at the sequence start — reproduced). fixed_length > 0 pads
     every batch to that length (required for hipGraph-captured steps)."""
     B = len(batch)
-    max_len = fixed_length or max(len(x.item_ids) for x in batch)
+    natural = max(len(x.item_ids) for x in batch)
+    if fixed_length:
+        assert natural <= fixed_length, (
+            f"tiger_pad_collate: sample length {natural} exceeds "
+            f"fixed_length {fixed_length} (set max_items_per_seq to match "
+            f"max_seq_len for hipGraph training)")
+    max_len = fixed_length or natural
     tgt_len = len(batch[0].target_ids)
     user_ids = torch.zeros(B, 1, dtype=torch.long)
     ids = torch.full((B, max_len), pad_id, dtype=torch.long)

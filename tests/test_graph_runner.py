@@ -91,3 +91,16 @@ def test_sasrec_hstu_use_hip_graph_flag_cpu_safe(tmp_path):
                   save_every_epoch=100)
     sasrec_trainer.train(ffn_dim=16, **common)
     hstu_trainer.train(**common)
+
+
+def test_tiger_fixed_length_collate_guard():
+    import pytest
+
+    from genrec_amd.data.collate import tiger_pad_collate
+    from genrec_amd.data.schemas import SeqData
+
+    b = [SeqData(0, [1] * 12, [1, 2, 3])]
+    out = tiger_pad_collate(b, sem_id_dim=3, fixed_length=15)
+    assert out["item_input_ids"].shape == (1, 15)
+    with pytest.raises(AssertionError, match="exceeds"):
+        tiger_pad_collate(b, sem_id_dim=3, fixed_length=9)
