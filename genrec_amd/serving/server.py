@@ -172,23 +172,55 @@ except ImportError:  # pragma: no cover
 
 def create_app(service: RecommendationService, max_batch: int = 64,
                window_ms: float = 2.0):
-    from fastapi import FastAPI
+    from fastapi import FastAPI, Response
 
     app = FastAPI(title="genrec_amd TIGER serving")
     batcher = _MicroBatcher(service, max_batch=max_batch,
                             window_ms=window_ms)
+
+    # Prometheus observability (the reference's deployment doc only
+    # sketches monitoring; this is a working /metrics endpoint)
+    try:
+        from prometheus_client import (CollectorRegistry, Counter,
+                                       Histogram, generate_latest)
+
+        registry = CollectorRegistry()
+        req_count = Counter("genrec_requests_total",
+                            "recommendation requests", ["endpoint"],
+                            registry=registry)
+        req_lat = Histogram(
+            "genrec_request_latency_seconds", "request latency",
+            ["endpoint"], registry=registry,
+            buckets=(.002, .005, .01, .025, .05, .1, .25, .5, 1., 2.5))
+    except ImportError:  # pragma: no cover
+        registry = None
+
+    def _observe(endpoint: str, t0: float) -> float:
+        dt = time.perf_counter() - t0
+        if registry is not None:
+            req_count.labels(endpoint).inc()
+            req_lat.labels(endpoint).observe(dt)
+        return dt * 1000.0
 
     @app.get("/health")
     def health():
         return {"status": "ok", "device": str(service.device),
                 "num_items": service.item_sem_ids.size(0)}
 
+    @app.get("/metrics")
+    def metrics():
+        if registry is None:
+            return Response("prometheus_client not installed",
+                            status_code=501)
+        return Response(generate_latest(registry),
+                        media_type="text/plain; version=0.0.4")
+
     @app.post("/recommend")
     async def recommend(req: "RecommendRequest"):
         t0 = time.perf_counter()
         recs = await batcher.submit(req.user_id, req.history, req.top_k)
         return {"recommendations": recs,
-                "latency_ms": (time.perf_counter() - t0) * 1000.0}
+                "latency_ms": _observe("recommend", t0)}
 
     @app.post("/batch_recommend")
     def batch_recommend(req: "BatchRecommendRequest"):
@@ -196,7 +228,7 @@ def create_app(service: RecommendationService, max_batch: int = 64,
         recs = service.recommend_batch(req.user_ids, req.histories,
                                        req.top_k)
         return {"recommendations": recs,
-                "latency_ms": (time.perf_counter() - t0) * 1000.0}
+                "latency_ms": _observe("batch_recommend", t0)}
 
     return app
 

@@ -95,3 +95,18 @@ def test_micro_batcher_fuses_concurrent_requests(service):
     assert len(results) == 6
     assert all(len(r) == 5 for r in results)
     assert calls == [6], calls  # one fused call, not six
+
+
+def test_metrics_endpoint(service):
+    from fastapi.testclient import TestClient
+
+    from genrec_amd.serving.server import create_app
+
+    client = TestClient(create_app(service, window_ms=1.0))
+    r = client.post("/recommend",
+                    json={"user_id": 1, "history": [1, 2], "top_k": 3})
+    assert r.status_code == 200
+    m = client.get("/metrics")
+    assert m.status_code == 200
+    assert "genrec_requests_total" in m.text
+    assert 'endpoint="recommend"' in m.text
