@@ -1,0 +1,73 @@
+"""Property-based tests (hypothesis) for foundational pieces: config value
+parsing, metric math, bucketing."""
+
+import math
+
+import torch
+from hypothesis import given, settings, strategies as st
+
+from genrec_amd.config import ginlite
+from genrec_amd.modules import TopKAccumulator, relative_position_bucket
+
+literals = st.recursive(
+    st.one_of(st.integers(-10**6, 10**6),
+              st.floats(allow_nan=False, allow_infinity=False,
+                        width=32),
+              st.booleans(), st.none(),
+              st.text(alphabet=st.characters(
+                  whitelist_categories=("Ll", "Lu", "Nd"),
+                  whitelist_characters=" _-"), max_size=12)),
+    lambda c: st.lists(c, max_size=4),
+    max_leaves=6)
+
+
+@settings(max_examples=150, deadline=None)
+@given(literals)
+def test_ginlite_parses_any_python_literal(value):
+    assert ginlite._parse_value(repr(value)) == value
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.integers(0, 2**31), st.integers(1, 8), st.integers(1, 6),
+       st.integers(2, 5))
+def test_topk_accumulator_matches_bruteforce(seed, b, k, d):
+    g = torch.Generator().manual_seed(seed)
+    actual = torch.randint(0, 4, (b, d), generator=g)
+    topk = torch.randint(0, 4, (b, k, d), generator=g)
+    acc = TopKAccumulator(ks=[1, k])
+    acc.accumulate(actual, topk)
+    m = acc.reduce()
+    # brute force: first exact-match rank per row
+    hits1 = hitsk = ndcg = 0.0
+    for i in range(b):
+        rank = None
+        for r in range(k):
+            if torch.equal(topk[i, r], actual[i]):
+                rank = r
+                break
+        if rank is not None:
+            hitsk += 1
+            ndcg += 1.0 / math.log2(rank + 2)
+            if rank == 0:
+                hits1 += 1
+    assert abs(m["Recall@1"] - hits1 / b) < 1e-6
+    assert abs(m[f"Recall@{k}"] - hitsk / b) < 1e-6 or k == 1
+    assert abs(m[f"NDCG@{k}"] - ndcg / b) < 1e-6
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.integers(-500, 500), st.integers(4, 64).filter(lambda x: x % 2 == 0),
+       st.integers(16, 256))
+def test_relative_position_bucket_bounds(rel, num_buckets, max_distance):
+    b = relative_position_bucket(torch.tensor([[rel]]),
+                                 num_buckets=num_buckets,
+                                 max_distance=max_distance,
+                                 bidirectional=True)
+    assert 0 <= b.item() < num_buckets
+    # monotone in |distance| within each sign
+    if rel > 1:
+        b2 = relative_position_bucket(torch.tensor([[rel - 1]]),
+                                      num_buckets=num_buckets,
+                                      max_distance=max_distance,
+                                      bidirectional=True)
+        assert b2.item() <= b.item()
