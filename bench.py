@@ -66,7 +66,6 @@ def main() -> None:
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--batch-size", type=int, default=256)
-    p.add_argument("--model", type=str, default="tiger")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
