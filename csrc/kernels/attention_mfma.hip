@@ -14,10 +14,11 @@
 //
 // Backward: attn_bwd_ds_kernel computes dP = dO V^T with the same MFMA
 // geometry, applies the dropout/query-mask multiplier and the softmax
-// Jacobian (or SiLU') in-register, and writes dS (bf16) and the masked
-// probabilities A_d (bf16) to global. dQ/dK/dV are then three plain
-// strided-batched GEMMs done by hipBLASLt from the host wrapper — exactly
-// the "library GEMMs stay in the library" split.
+// Jacobian (or SiLU') in-register, stages dS / dS^T / A_d^T in LDS, and
+// finishes dQ = dS K, dK = dS^T Q, dV = A_d^T dO as MFMA in the SAME
+// launch (no extra batched GEMMs, no extra global round-trips). All
+// global tensors are stride-parameterized so [B,L,H,D] transpose views
+// flow in/out without permute copies.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
