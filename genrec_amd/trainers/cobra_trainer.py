@@ -91,6 +91,7 @@ def train(
     num_workers: int = 0,
     eval_max_batches: Optional[int] = None,
 ):
+    common.enable_tuned_gemms()
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "cobra")
     common.set_seed(seed, ctx.rank)

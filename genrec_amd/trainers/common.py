@@ -24,6 +24,23 @@ from genrec_amd.parallel import DistributedContext
 logger = logging.getLogger("genrec_amd")
 
 
+def enable_tuned_gemms() -> None:
+    """Point TunableOp at the shipped hipBLASLt solution tables
+    (benchmarks/tunableop<ordinal>.csv) when present — same mechanism
+    bench.py uses. No-op if the env is already configured or the tables
+    are absent; tuning itself stays off."""
+    repo = os.path.dirname(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    csv0 = os.path.join(repo, "benchmarks", "tunableop0.csv")
+    if os.path.exists(csv0) and \
+            os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") != "1":
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                              os.path.join(repo, "benchmarks",
+                                           "tunableop.csv"))
+
+
 def setup_logging(log_dir: Optional[str] = None, name: str = "train") -> None:
     handlers: list = [logging.StreamHandler()]
     if log_dir:

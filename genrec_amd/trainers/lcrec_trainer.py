@@ -248,6 +248,7 @@ def train(
         n_codebooks = num_codebooks
     if not amp:
         mixed_precision_type = None
+    common.enable_tuned_gemms()
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "lcrec")
     common.set_seed(seed, ctx.rank)

@@ -83,6 +83,7 @@ def train(
 ):
     assert (epochs is None) != (iterations is None), \
         "epochs and iterations are mutually exclusive (ref rqvae_trainer.py:91-96)"
+    common.enable_tuned_gemms()
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "rqvae")
     common.set_seed(seed, ctx.rank)

@@ -86,6 +86,7 @@ def train(
     num_workers: int = 4,
     use_hip_graph: bool = False,
 ):
+    common.enable_tuned_gemms()
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "sasrec")
     common.set_seed(seed, ctx.rank)

@@ -92,6 +92,7 @@ def train(
     profile_trace_path: Optional[str] = None,
     use_hip_graph: bool = False,
 ):
+    common.enable_tuned_gemms()
     ctx = init_distributed()
     common.setup_logging(save_dir_root if ctx.is_main else None, "tiger")
     common.set_seed(seed, ctx.rank)
