@@ -268,3 +268,18 @@ def test_eager_attention_matches_naive_formula(case):
     out = eager.fused_attention(q, k, v, scale=0.25, **kw)
     ref = _naive_attention(q, k, v, 0.25, **nkw)
     assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
+
+
+def test_flash_dispatch_inert_by_default(monkeypatch):
+    """Long-Lk attention falls back to eager on CPU regardless of the
+    staged-flash env flag (the flash kernels need the GPU extension)."""
+    from genrec_amd.ops.attention import fused_attention
+
+    q = torch.randn(2, 2, 80, 32)
+    k = torch.randn(2, 2, 128, 32)
+    v = torch.randn(2, 2, 128, 32)
+    out1 = fused_attention(q, k, v, scale=0.1, causal=True)
+    monkeypatch.setenv("GENREC_ATTN_FLASH", "1")
+    out2 = fused_attention(q, k, v, scale=0.1, causal=True)
+    assert torch.allclose(out1, out2)
+    assert out1.shape == (2, 2, 80, 32)
