@@ -135,7 +135,12 @@ class T5Attention(nn.Module):
 
         bias = None
         if self.rel_bias is not None:
-            bias = self.compute_bias(q.size(2), k.size(2), q.device)
+            # cached incremental decode: queries sit at the LAST q_len
+            # positions of the k_len-long sequence, so take the bottom
+            # rows of the full bias square
+            bias = self.compute_bias(k.size(2), k.size(2), q.device)
+            if q.size(2) != k.size(2):
+                bias = bias[:, k.size(2) - q.size(2):, :]
 
         out = ops.t5_attention(
             q, k, v, bias, key_padding_mask, attn_mask, self.scale,

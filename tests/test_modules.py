@@ -283,3 +283,27 @@ def test_flash_dispatch_inert_by_default(monkeypatch):
     out2 = fused_attention(q, k, v, scale=0.1, causal=True)
     assert torch.allclose(out1, out2)
     assert out1.shape == (2, 2, 80, 32)
+
+
+def test_decoder_kv_cache_incremental_matches_full():
+    """Token-by-token decode with kv_caches == full causal forward,
+    including the relative-position bias rows for cached queries."""
+    from genrec_amd.modules.transformer import TransformerDecoder
+
+    torch.manual_seed(0)
+    dec = TransformerDecoder(dim=32, depth=2, num_heads=4, dropout=0.0,
+                             ff_hidden_dim=48)
+    dec.eval()
+    B, T, Lm = 2, 5, 7
+    tgt = torch.randn(B, T, 32)
+    memory = torch.randn(B, Lm, 32)
+    causal = torch.full((T, T), float("-inf")).triu(1)
+    with torch.no_grad():
+        full = dec(tgt, memory=memory, attn_mask=causal)
+        caches = [dict() for _ in range(2)]
+        steps = []
+        for t in range(T):
+            steps.append(dec(tgt[:, t:t + 1], memory=memory,
+                             kv_caches=caches))
+        inc = torch.cat(steps, dim=1)
+    assert torch.allclose(full, inc, atol=1e-5), (full - inc).abs().max()
