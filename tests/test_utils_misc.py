@@ -44,3 +44,14 @@ def test_step_profiler_and_roctx_noop_on_cpu():
     prof.report()
     with roctx_range("cpu-noop"):
         pass
+
+
+def test_checkpoint_gated_encoders_raise_clearly(tmp_path):
+    import pytest
+
+    from genrec_amd.modules.encoders import (BgeEncoder, ErnieEncoder,
+                                             SentenceT5Encoder)
+
+    for cls in (SentenceT5Encoder, ErnieEncoder, BgeEncoder):
+        with pytest.raises(FileNotFoundError, match="local pretrained"):
+            cls(str(tmp_path / "nonexistent-model"))
