@@ -55,3 +55,10 @@ def test_checkpoint_gated_encoders_raise_clearly(tmp_path):
     for cls in (SentenceT5Encoder, ErnieEncoder, BgeEncoder):
         with pytest.raises(FileNotFoundError, match="local pretrained"):
             cls(str(tmp_path / "nonexistent-model"))
+
+
+def test_data_cycle_wraps_around():
+    from genrec_amd.data.utils import cycle
+
+    it = cycle([1, 2, 3])
+    assert [next(it) for _ in range(7)] == [1, 2, 3, 1, 2, 3, 1]
