@@ -88,6 +88,7 @@ def test_tiger_trainer_smoke(tmp_path):
 
     tiger_trainer.train(
         dataset=Tiny, epochs=1, max_steps=2, num_workers=0, batch_size=16,
+        profile_steps=1,
         save_dir_root=str(tmp_path), amp=False, eval_max_batches=1,
         save_every_epoch=1)
     assert os.path.exists(os.path.join(str(tmp_path), "checkpoint_final.pt"))
