@@ -62,3 +62,24 @@ def test_data_cycle_wraps_around():
 
     it = cycle([1, 2, 3])
     assert [next(it) for _ in range(7)] == [1, 2, 3, 1, 2, 3, 1]
+
+
+def test_enable_tuned_gemms_env(monkeypatch):
+    import os
+
+    from genrec_amd.trainers import common
+
+    for k in ("PYTORCH_TUNABLEOP_ENABLED", "PYTORCH_TUNABLEOP_TUNING",
+              "PYTORCH_TUNABLEOP_FILENAME"):
+        monkeypatch.delenv(k, raising=False)
+    common.enable_tuned_gemms()
+    # the repo ships benchmarks/tunableop0.csv -> env gets pointed at it
+    assert os.environ["PYTORCH_TUNABLEOP_ENABLED"] == "1"
+    assert os.environ["PYTORCH_TUNABLEOP_TUNING"] == "0"
+    assert os.environ["PYTORCH_TUNABLEOP_FILENAME"].endswith(
+        "tunableop.csv")
+    # explicit tuning runs are left alone
+    monkeypatch.setenv("PYTORCH_TUNABLEOP_TUNING", "1")
+    monkeypatch.delenv("PYTORCH_TUNABLEOP_ENABLED", raising=False)
+    common.enable_tuned_gemms()
+    assert "PYTORCH_TUNABLEOP_ENABLED" not in os.environ
