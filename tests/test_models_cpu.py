@@ -232,3 +232,18 @@ def test_tiger_load_pretrained_both_formats(tmp_path):
     m3.load_pretrained(str(tmp_path / "ck.pt"))
     for a, b, c in zip(m1.parameters(), m2.parameters(), m3.parameters()):
         assert torch.equal(a, b) and torch.equal(a, c)
+
+
+def test_rotation_trick_identity_invariant():
+    """When the input already equals the codeword, the rotation-trick
+    transform is the identity (Householder reflection of u onto q with
+    u == q) — and gradients flow through x (unlike STE)."""
+    from genrec_amd.models.rqvae import efficient_rotation_trick_transform
+
+    torch.manual_seed(0)
+    x = torch.randn(6, 8, requires_grad=True)
+    xn = x / x.norm(dim=-1, keepdim=True)
+    out = efficient_rotation_trick_transform(xn.detach(), xn.detach(), x)
+    assert torch.allclose(out, x, atol=1e-5)
+    out.sum().backward()
+    assert x.grad is not None and x.grad.abs().sum() > 0
