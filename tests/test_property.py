@@ -71,3 +71,24 @@ def test_relative_position_bucket_bounds(rel, num_buckets, max_distance):
                                       max_distance=max_distance,
                                       bidirectional=True)
         assert b2.item() <= b.item()
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(2, 50), st.integers(1, 6), st.integers(0, 2**31))
+def test_sem_id_flat_index_bijective(vocab, dim, seed):
+    """SemIdEmbedding's type*V+id flat index is a bijection over
+    (type, id) pairs — no two pairs share a row, padding row is last."""
+    from genrec_amd.modules import SemIdEmbedding
+
+    emb = SemIdEmbedding(num_embeddings=vocab, sem_ids_dim=dim,
+                         embeddings_dim=4)
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, vocab, (1, dim), generator=g)
+    types = torch.arange(dim).unsqueeze(0)
+    flat = types * vocab + ids
+    assert flat.max() < vocab * dim  # always below the padding row
+    assert emb.padding_idx == vocab * dim
+    # bijection: enumerate all pairs
+    allp = torch.cartesian_prod(torch.arange(dim), torch.arange(vocab))
+    rows = allp[:, 0] * vocab + allp[:, 1]
+    assert rows.unique().numel() == dim * vocab
