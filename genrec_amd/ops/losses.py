@@ -102,10 +102,11 @@ def sampled_tied_softmax_ce(hidden: Tensor, emb_weight: Tensor,
     device = hidden.device
     V = emb_weight.size(0)
     valid = targets != ignore_index
+    if not bool(valid.any()):
+        # all-padding batch: grad-connected zero instead of CE's 0/0 NaN
+        return hidden.sum() * 0.0
     neg = torch.randint(1 if ignore_index == 0 else 0, V,
                         (num_negatives,), device=device)
-    cand = torch.cat([targets.clamp_min(0).unsqueeze(0), 
-                      neg.unsqueeze(0).expand(n, -1).reshape(n, -1).t()])         if False else None  # (kept simple below)
     # logits: [N, 1+M] = [h . e_target, h @ E_neg^T]
     tgt_emb = emb_weight[targets.clamp(0, V - 1)]            # [N, D]
     pos_logit = (hidden * tgt_emb).sum(-1, keepdim=True)     # [N, 1]

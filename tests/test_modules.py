@@ -307,3 +307,22 @@ def test_decoder_kv_cache_incremental_matches_full():
                              kv_caches=caches))
         inc = torch.cat(steps, dim=1)
     assert torch.allclose(full, inc, atol=1e-5), (full - inc).abs().max()
+
+
+def test_sampled_ce_masks_accidental_hits():
+    """A negative equal to the target must not contribute (masked -1e9) —
+    loss with a colliding negative pool stays finite and close to the
+    collision-free value."""
+    from genrec_amd.ops.losses import sampled_tied_softmax_ce
+
+    torch.manual_seed(0)
+    h = torch.randn(16, 8)
+    E = torch.randn(4, 8)  # tiny vocab -> frequent collisions
+    t = torch.randint(1, 4, (16,))
+    loss = sampled_tied_softmax_ce(h, E, t, num_negatives=64,
+                                   ignore_index=0)
+    assert torch.isfinite(loss)
+    # all rows ignored -> zero-ish loss, no NaN
+    t0 = torch.zeros(16, dtype=torch.long)
+    l0 = sampled_tied_softmax_ce(h, E, t0, num_negatives=8, ignore_index=0)
+    assert torch.isfinite(l0)
