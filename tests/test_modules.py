@@ -326,3 +326,17 @@ def test_sampled_ce_masks_accidental_hits():
     t0 = torch.zeros(16, dtype=torch.long)
     l0 = sampled_tied_softmax_ce(h, E, t0, num_negatives=8, ignore_index=0)
     assert torch.isfinite(l0)
+
+
+def test_summed_ce_matches_manual_formula():
+    """TIGER loss: CE(reduction=none) summed over T, mean over B
+    (ref tiger.py:232-240) — against a from-scratch formula."""
+    from genrec_amd.ops.losses import summed_ce
+
+    torch.manual_seed(0)
+    B, T, V = 5, 3, 11
+    logits = torch.randn(B, T, V)
+    targets = torch.randint(0, V, (B, T))
+    lp = torch.log_softmax(logits, dim=-1)
+    manual = -lp.gather(-1, targets.unsqueeze(-1)).squeeze(-1).sum(1).mean()
+    assert torch.allclose(summed_ce(logits, targets), manual, atol=1e-6)
