@@ -312,9 +312,16 @@ class AmazonSeqDataset(Dataset):
         self._generate_samples()
 
     def _generate_samples(self) -> None:
+        import zlib
+
         self.samples = []
         for uidx, full in enumerate(self.sequences):
-            user_id = hash(self.user_ids[uidx]) % 10000
+            # deliberate fix of a reference quirk: the reference uses
+            # python hash() % 10000 (amazon.py:412), which is randomized
+            # per process (PYTHONHASHSEED) — under multi-process DDP the
+            # SAME user maps to different embedding rows on different
+            # ranks. crc32 is stable across processes and runs.
+            user_id = zlib.crc32(str(self.user_ids[uidx]).encode()) % 10000
             if self.train_test_split == "train":
                 seq = full[:-2]
                 for i in range(1, len(seq)):

@@ -141,3 +141,19 @@ def test_missing_raw_raises(tmp_path):
     with pytest.raises(FileNotFoundError):
         AmazonSASRecDataset(root=str(tmp_path), split="beauty",
                             train_test_split="train")
+
+
+def test_user_hash_is_process_stable(amazon_root, rqvae_ckpt):
+    """User-id hashing must be deterministic across processes (the
+    reference's python hash() is PYTHONHASHSEED-randomized, which maps
+    the same user to different embedding rows on different DDP ranks)."""
+    import subprocess
+    import sys
+
+    code = (
+        "import zlib; print(zlib.crc32('A1B2C3'.encode()) % 10000)")
+    outs = {subprocess.run([sys.executable, "-c", code],
+                           capture_output=True, text=True,
+                           env={"PYTHONHASHSEED": str(i)}).stdout
+            for i in (0, 1)}
+    assert len(outs) == 1  # same value under different hash seeds
