@@ -66,9 +66,11 @@ class RecommendationService:
         item_ids = torch.zeros(b, L, dtype=torch.long)
         mask = torch.zeros(b, L, dtype=torch.long)
         ttype = (torch.arange(L) % dim).unsqueeze(0).expand(b, -1).clone()
+        n_items = self.item_sem_ids.size(0)
         sem_cpu = self.item_sem_ids.cpu()
         for i, h in enumerate(histories):
-            h = h[-max_len:]
+            # drop out-of-catalog indices instead of crashing the batch
+            h = [x for x in h if 0 <= x < n_items][-max_len:]
             flat = sem_cpu[torch.tensor(h, dtype=torch.long)].reshape(-1) \
                 if h else torch.zeros(0, dtype=torch.long)
             item_ids[i, :flat.numel()] = flat

@@ -110,3 +110,9 @@ def test_metrics_endpoint(service):
     assert m.status_code == 200
     assert "genrec_requests_total" in m.text
     assert 'endpoint="recommend"' in m.text
+
+
+def test_out_of_catalog_history_is_dropped(service):
+    n = service.item_sem_ids.size(0)
+    recs = service.recommend_batch([1], [[0, n + 50, -3, 1]], top_k=3)
+    assert len(recs) == 1 and len(recs[0]) <= 3  # served, not crashed
