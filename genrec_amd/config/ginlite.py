@@ -174,14 +174,25 @@ def bind(key: str, value_text_or_value: Any, *, raw: bool = False) -> None:
 _LINE_RE = re.compile(r"^\s*([\w./%-]+(?:/[\w.]+)?)\s*=\s*(.+?)\s*$")
 
 
+_PARSE_STACK: list = []
+
+
 def parse_file(path: str, substitutions: Optional[Dict[str, str]] = None) -> None:
+    ap = os.path.abspath(path)
+    if ap in _PARSE_STACK:
+        raise GinError(
+            "circular include: " + " -> ".join(_PARSE_STACK + [ap]))
     with open(path, "r") as f:
         text = f.read()
     if substitutions:
         for k, v in substitutions.items():
             text = text.replace("{" + k + "}", v)
     base_dir = os.path.dirname(os.path.abspath(path))
-    _parse_text(text, base_dir, substitutions)
+    _PARSE_STACK.append(ap)
+    try:
+        _parse_text(text, base_dir, substitutions)
+    finally:
+        _PARSE_STACK.pop()
 
 
 def _strip_comment(line: str) -> str:

@@ -247,3 +247,16 @@ def test_reference_gin_binding_names_are_drop_in():
                 m = re.match(r"\s*train\.(\w+)\s*=", line)
                 assert not m or m.group(1) in params, \
                     f"{mod.__name__} missing param {m.group(1)} ({f})"
+
+
+def test_circular_include_raises(tmp_path):
+    from genrec_amd.config import ginlite
+
+    a = tmp_path / "a.gin"
+    b = tmp_path / "b.gin"
+    a.write_text(f'include "{b}"\n')
+    b.write_text(f'include "{a}"\n')
+    import pytest
+
+    with pytest.raises(ginlite.GinError, match="circular"):
+        ginlite.parse_file(str(a))
