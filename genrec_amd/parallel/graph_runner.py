@@ -237,8 +237,16 @@ class GraphedTrainStep:
                 self.flat_params.copy_(
                     self.flat_master.to(torch.bfloat16))
             else:
-                for mt, s in zip(self.masters, state["masters"]):
-                    mt.copy_(s.to(mt.device))
+                if "flat_master" in state:  # written by the fused path
+                    flat = state["flat_master"]
+                    off = 0
+                    for mt in self.masters:
+                        mt.copy_(flat[off:off + mt.numel()]
+                                 .view_as(mt).to(mt.device))
+                        off += mt.numel()
+                else:
+                    for mt, s in zip(self.masters, state["masters"]):
+                        mt.copy_(s.to(mt.device))
+                    self.opt.load_state_dict(state["optimizer"])
                 torch._foreach_copy_(self.params, self.masters)
-                self.opt.load_state_dict(state["optimizer"])
         self.set_lr(state.get("lr", float(self.lr_t)))

@@ -104,3 +104,20 @@ def test_tiger_fixed_length_collate_guard():
     assert out["item_input_ids"].shape == (1, 15)
     with pytest.raises(AssertionError, match="exceeds"):
         tiger_pad_collate(b, sem_id_dim=3, fixed_length=9)
+
+
+def test_fused_checkpoint_loads_into_cpu_runner():
+    """A fused-path (GPU) runner checkpoint restores into the CPU
+    fallback runner (cross-device resume)."""
+    import torch
+
+    model = _tiny_model()
+    r = GraphedTrainStep(model, _batch(), lambda o: o.loss, use_graph=False)
+    n = sum(p.numel() for p in r.params)
+    fake = {"flat_master": torch.arange(n, dtype=torch.float32) * 1e-4,
+            "m": torch.zeros(n), "v": torch.zeros(n),
+            "step": torch.zeros(1, dtype=torch.int32), "lr": 5e-4}
+    r.load_state_dict(fake)
+    flat = torch.cat([m.reshape(-1) for m in r.masters])
+    assert torch.allclose(flat, fake["flat_master"])
+    assert r.opt.param_groups[0]["lr"] == 5e-4
