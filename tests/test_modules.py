@@ -340,3 +340,24 @@ def test_summed_ce_matches_manual_formula():
     lp = torch.log_softmax(logits, dim=-1)
     manual = -lp.gather(-1, targets.unsqueeze(-1)).squeeze(-1).sum(1).mean()
     assert torch.allclose(summed_ce(logits, targets), manual, atol=1e-6)
+
+
+def test_cosine_warmup_schedule_matches_closed_form():
+    """Same lambda as HF's get_cosine_schedule_with_warmup (the reference
+    trainers import it from transformers; tiger_trainer.py:222-227) AND
+    the graph-mode trainer's host-side _cosine_lr must agree with it."""
+    from genrec_amd.modules.schedulers import get_cosine_schedule_with_warmup
+
+    warm, total, base = 10, 50, 2.0
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=base)
+    sched = get_cosine_schedule_with_warmup(opt, warm, total)
+    for step in range(total):
+        lr = opt.param_groups[0]["lr"]
+        if step < warm:
+            expect = base * step / warm
+        else:
+            prog = (step - warm) / (total - warm)
+            expect = base * max(0.0, 0.5 * (1 + math.cos(math.pi * prog)))
+        assert abs(lr - expect) < 1e-9, (step, lr, expect)
+        opt.step()
+        sched.step()
