@@ -218,3 +218,47 @@ def test_sasrec_trainer_world2_with_eval(tmp_path):
     mp.spawn(_run_sasrec_trainer, args=(2, 29931, results, str(tmp_path)),
              nprocs=2, join=True)
     assert results[0] and results[1]
+
+
+def _run_accum(rank, world, port, results):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(7)
+    from genrec_amd.parallel import GradReducer
+
+    model = torch.nn.Linear(4, 3)
+    reducer = GradReducer(model, bucket_cap_mb=0.0001)
+    torch.manual_seed(100 + rank)
+    micro = [torch.randn(5, 4) for _ in range(2)]
+    # micro-step 1: no sync
+    reducer.skip_sync = True
+    model(micro[0]).sum().backward()
+    # micro-step 2: sync + finalize
+    reducer.skip_sync = False
+    model(micro[1]).sum().backward()
+    reducer.finalize()
+    got = torch.cat([p.grad.flatten() for p in model.parameters()])
+    # expected: rank-local sum of micro grads, averaged across ranks
+    ref = torch.nn.Linear(4, 3)
+    torch.manual_seed(7)
+    ref = torch.nn.Linear(4, 3)
+    for m in micro:
+        ref(m).sum().backward()
+    local = torch.cat([p.grad.flatten() for p in ref.parameters()])
+    results[f"got{rank}"] = got.clone()
+    results[f"local{rank}"] = local.clone()
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_grad_reducer_accumulation_micro_steps(tmp_path):
+    """skip_sync micro-steps accumulate locally; the final sync averages
+    the ACCUMULATED gradients across ranks."""
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_run_accum, args=(2, 29937, results), nprocs=2, join=True)
+    expect = (results["local0"] + results["local1"]) / 2
+    assert torch.allclose(results["got0"], expect, atol=1e-6)
+    assert torch.allclose(results["got1"], expect, atol=1e-6)
