@@ -247,3 +247,22 @@ def test_rotation_trick_identity_invariant():
     assert torch.allclose(out, x, atol=1e-5)
     out.sum().backward()
     assert x.grad is not None and x.grad.abs().sum() > 0
+
+
+def test_hstu_temporal_bucket_formula():
+    """log2-bucketed |ts_i - ts_j| with clamp (ref hstu.py:368-409):
+    bucket = clamp(floor(ln(max(|dt|,1)) / ln 2), 0, n-1)."""
+    import math
+
+    from genrec_amd.models.hstu import TemporalBias
+
+    tb = TemporalBias(num_heads=2, num_buckets=8)
+    ts = torch.tensor([[0, 1, 10, 1000, 10**9]], dtype=torch.long)
+    bias = tb(ts)  # [B, H, L, L]
+    assert bias.shape == (1, 2, 5, 5)
+    diff = (ts.unsqueeze(2) - ts.unsqueeze(1)).abs().clamp(min=1).float()
+    buckets = (diff.log() / 0.693).long().clamp(0, 7)
+    expect = tb.temporal_attention_bias(buckets).permute(0, 3, 1, 2)
+    assert torch.allclose(bias, expect)
+    # same timestamp -> bucket 0; huge gap -> clamped top bucket
+    assert buckets[0, 0, 0] == 0 and buckets[0, 0, 4] == 7
