@@ -132,3 +132,26 @@ def test_notellm_hardneg_and_labels():
     out = m(tok["input_ids"], tok["attention_mask"], tok["emb_token_idx"],
             hardneg=tok["hardneg"])
     assert torch.isfinite(out["loss"])
+
+
+def test_notellm_trainer_smoke(tmp_path):
+    """End-to-end NoteLLM contrastive trainer on a tiny backbone
+    (beyond-reference: the reference ships no NoteLLM trainer)."""
+    from genrec_amd.data.notellm_synthetic import SyntheticNotePairDataset
+    from genrec_amd.trainers import notellm_trainer
+
+    class Tiny(SyntheticNotePairDataset):
+        def __init__(self, **kw):
+            kw.update(num_pairs=24)
+            super().__init__(**kw)
+
+    notellm_trainer.train(
+        epochs=1, max_steps=2, batch_size=4, dataset=Tiny,
+        backbone_config=dict(vocab_size=512, hidden_size=32, num_layers=1,
+                             num_heads=4, num_kv_heads=2,
+                             intermediate_size=64),
+        gradient_checkpointing=False, do_eval=True, eval_max_batches=2,
+        save_dir_root=str(tmp_path), save_every_epoch=1, num_workers=0)
+    import os
+
+    assert os.path.isdir(os.path.join(str(tmp_path), "epoch_0"))
