@@ -138,7 +138,8 @@ def train(
         model.train()
         if hasattr(train_loader.sampler, "set_epoch"):
             train_loader.sampler.set_epoch(epoch)
-        ep = {"acc_c": 0.0, "acc_t": 0.0, "rec_c": 0.0, "rec_t": 0.0}
+        # device-resident epoch counters: no per-batch .item() syncs
+        ep_t = torch.zeros(4, device=device)
         for it, batch in enumerate(train_loader):
             micro = (it + 1) % gradient_accumulate_every == 0
             reducer.skip_sync = not micro
@@ -149,10 +150,10 @@ def train(
                 loss = sparse_loss_weight * out.loss_sparse \
                     + dense_loss_weight * out.loss_dense
             (loss / gradient_accumulate_every).backward()
-            ep["acc_c"] += out.acc_correct.item()
-            ep["acc_t"] += out.acc_total.item()
-            ep["rec_c"] += out.recall_correct.item()
-            ep["rec_t"] += out.recall_total.item()
+            with torch.no_grad():
+                ep_t += torch.stack([out.acc_correct, out.acc_total,
+                                     out.recall_correct,
+                                     out.recall_total]).to(ep_t)
             if micro:
                 reducer.finalize()
                 torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
@@ -171,7 +172,9 @@ def train(
                                 out.codebook_entropy.item()})
             if max_steps is not None and step >= max_steps:
                 break
-        ep = reduce_scalars(ep, device)
+        ep = reduce_scalars(
+            {k: v.item() for k, v in zip(
+                ("acc_c", "acc_t", "rec_c", "rec_t"), ep_t)}, device)
         if ctx.is_main and ep["acc_t"] > 0:
             logger.info("epoch %d codebook_acc %.4f item_recall %.4f",
                         epoch, ep["acc_c"] / max(ep["acc_t"], 1),
