@@ -33,7 +33,7 @@ def summarize(db_path: str, out_path: str, title: str,
         f.write("| kernel | calls | total ms | avg us | VGPR | LDS B |\n")
         f.write("|---|---|---|---|---|---|\n")
         for name, n, ms, avg, vgpr, lds in rows:
-            short = name.split("(")[0][:80]
+            short = name[:200].replace("|", "\\|")
             f.write(f"| `{short}` | {n} | {ms:.2f} | {avg:.1f} "
                     f"| {vgpr or ''} | {lds or ''} |\n")
     print(f"wrote {out_path}")
