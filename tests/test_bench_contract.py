@@ -28,6 +28,8 @@ def test_bench_json_contract():
     assert d["config"]["model"] == "tiger-amazon-beauty"
     assert d["config"]["global_batch"] == 8
     assert d["config"]["parallelism"] == "dp1"
+    assert d["config"]["hip_graph"] is False  # no GPU here
+    assert d["config"]["seq_len"] == 61
     assert isinstance(d["ms_per_step"], float)
 
 
