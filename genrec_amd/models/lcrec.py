@@ -23,11 +23,15 @@ from __future__ import annotations
 import os
 from typing import Callable, Dict, List, Optional, Tuple
 
+import logging
+
 import torch
 import torch.nn.functional as F
 from torch import nn
 
 from genrec_amd.config import ginlite
+
+logger = logging.getLogger("genrec_amd")
 
 
 def default_qwen_config(vocab_size: int = 512, hidden_size: int = 1536,
@@ -115,7 +119,7 @@ class LCRec(nn.Module):
         self.tokenizer = AutoTokenizer.from_pretrained(load_dir)
         self.model = AutoModelForCausalLM.from_pretrained(
             load_dir, torch_dtype=torch.bfloat16)
-        print(f"Loaded checkpoint from {load_dir}")
+        logger.info("Loaded checkpoint from %s", load_dir)
 
     @torch.no_grad()
     def generate_topk(

@@ -20,6 +20,8 @@ from __future__ import annotations
 from enum import Enum
 from typing import List, NamedTuple
 
+import logging
+
 import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
@@ -33,6 +35,8 @@ from genrec_amd.modules.losses import (
 )
 from genrec_amd.modules.mlp import MLP
 from genrec_amd.modules.norms import L2Norm, l2norm
+
+logger = logging.getLogger("genrec_amd")
 
 
 class QuantizeForwardMode(Enum):
@@ -247,7 +251,8 @@ class RqVae(nn.Module):
         state = torch.load(path, map_location=self.device, weights_only=False)
         self.load_state_dict(state["model"])
         tag = state.get("iter", state.get("epoch", "?"))
-        print(f"---Loaded RQVAE checkpoint (step/epoch {tag}) from {path}---")
+        logger.info("Loaded RQVAE checkpoint (step/epoch %s) from %s",
+                    tag, path)
 
     def encode(self, x: Tensor) -> Tensor:
         return self.encoder(x)
