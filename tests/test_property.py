@@ -92,3 +92,17 @@ def test_sem_id_flat_index_bijective(vocab, dim, seed):
     allp = torch.cartesian_prod(torch.arange(dim), torch.arange(vocab))
     rows = allp[:, 0] * vocab + allp[:, 1]
     assert rows.unique().numel() == dim * vocab
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 40), st.integers(1, 8))
+def test_user_id_embedding_modulo(num, uid):
+    """UserIdEmbedding hashing: id % num_embeddings (ref embedding.py:62-74)
+    for any id, including ids far beyond the table."""
+    from genrec_amd.modules import UserIdEmbedding
+
+    emb = UserIdEmbedding(num_embeddings=num, embeddings_dim=4)
+    big = uid * 1_000_003 + 7
+    out = emb(torch.tensor([[big]]))
+    ref = emb.emb(torch.tensor([[big % num]]))
+    assert torch.equal(out, ref)
