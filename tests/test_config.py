@@ -260,3 +260,21 @@ def test_circular_include_raises(tmp_path):
 
     with pytest.raises(ginlite.GinError, match="circular"):
         ginlite.parse_file(str(a))
+
+
+def test_config_run_comments_reference_real_trainers():
+    """Every `python -m genrec_amd.trainers.X` mentioned in config files
+    and docs refers to an importable module."""
+    import glob
+    import importlib
+    import re
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    mods = set()
+    for pattern in ("config/**/*.gin", "docs/*.md", "README.md"):
+        for f in glob.glob(os.path.join(root, pattern), recursive=True):
+            for m in re.finditer(r"genrec_amd\.trainers\.(\w+)", open(f).read()):
+                mods.add(m.group(1))
+    assert mods  # sanity: the docs do reference trainers
+    for name in sorted(mods):
+        importlib.import_module(f"genrec_amd.trainers.{name}")
