@@ -106,3 +106,30 @@ def test_user_id_embedding_modulo(num, uid):
     out = emb(torch.tensor([[big]]))
     ref = emb.emb(torch.tensor([[big % num]]))
     assert torch.equal(out, ref)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 6), st.integers(1, 5), st.integers(0, 2**31),
+       st.sampled_from(["left", "right"]))
+def test_tiger_collate_shape_invariants(batch_size, sem_dim, seed, side):
+    from genrec_amd.data.collate import tiger_pad_collate
+    from genrec_amd.data.schemas import SeqData
+
+    g = np.random.default_rng(seed)
+    batch = []
+    for b in range(batch_size):
+        n_items = int(g.integers(1, 7))
+        batch.append(SeqData(
+            user_id=int(g.integers(0, 1000)),
+            item_ids=[int(x) for x in g.integers(0, 16, n_items * sem_dim)],
+            target_ids=[int(x) for x in g.integers(0, 16, sem_dim)]))
+    out = tiger_pad_collate(batch, sem_id_dim=sem_dim, padding_side=side)
+    L = out["item_input_ids"].shape[1]
+    assert L == max(len(s.item_ids) for s in batch)
+    assert out["seq_mask"].shape == (batch_size, L)
+    assert out["token_type_ids"].max() < sem_dim
+    # mask counts exactly the real tokens
+    assert out["seq_mask"].sum().item() == sum(len(s.item_ids)
+                                               for s in batch)
+    # masked positions' token ids are pad (0)
+    assert (out["item_input_ids"][out["seq_mask"] == 0] == 0).all()
