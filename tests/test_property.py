@@ -3,6 +3,7 @@ parsing, metric math, bucketing."""
 
 import math
 
+import numpy as np
 import torch
 from hypothesis import given, settings, strategies as st
 
