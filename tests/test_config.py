@@ -278,3 +278,13 @@ def test_config_run_comments_reference_real_trainers():
     assert mods  # sanity: the docs do reference trainers
     for name in sorted(mods):
         importlib.import_module(f"genrec_amd.trainers.{name}")
+
+
+def test_mkdocs_nav_files_exist():
+    import yaml
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = yaml.safe_load(open(os.path.join(root, "mkdocs.yml")))
+    for entry in cfg["nav"]:
+        for _, path in entry.items():
+            assert os.path.exists(os.path.join(root, path)), path
