@@ -46,8 +46,17 @@ class TopKAccumulator:
         self.total += actual.size(0)
 
     def reduce(self, all_reduce: bool = False) -> Dict[str, float]:
+        dist_up = (all_reduce and torch.distributed.is_available()
+                   and torch.distributed.is_initialized())
         if self._counters is None:
-            return {f"{m}@{k}": 0.0 for k in self.ks for m in ("Recall", "NDCG")}
+            if not dist_up:
+                return {f"{m}@{k}": 0.0
+                        for k in self.ks for m in ("Recall", "NDCG")}
+            # this rank's eval shard was empty: still enter the collective
+            # with zero counters, or the other ranks hang
+            dev = (torch.device("cuda", torch.cuda.current_device())
+                   if torch.cuda.is_available() else torch.device("cpu"))
+            self._ensure(dev)
         c = self._counters.clone()
         total = torch.tensor([float(self.total)], dtype=torch.float64,
                              device=c.device)

@@ -132,9 +132,10 @@ class _FusedAttnFn(torch.autograd.Function):
 
 
 class _FlashAttnFn(torch.autograd.Function):
-    """Flash-tiled attention (Lk/Lq beyond one 64-tile). STAGED: compiled
-    + bound but dispatched only with GENREC_ATTN_FLASH=1 until
-    GPU-validated (tile math proven in tools/sim_flash_tiles.py)."""
+    """Flash-tiled attention (Lk/Lq beyond one 64-tile): running-softmax
+    forward, single-pass backward via the flash identity (tile math proven
+    in tools/sim_flash_tiles.py; GPU-validated round 2 — default for
+    Lk>64 softmax attention; GENREC_DISABLE_ATTN_FLASH=1 opts out)."""
 
     @staticmethod
     def forward(ctx, q, k, v, bias, key_pad_mask, additive_mask, query_mask,
@@ -201,7 +202,7 @@ def fused_attention(
         qm = query_mask.contiguous() if query_mask is not None else None
         return _FusedAttnFn.apply(q, k, v, b, kp, am, qm,
                                   scale, causal, act, dropout_p, training)
-    flash_ok = (os.environ.get("GENREC_ATTN_FLASH", "0") == "1"
+    flash_ok = (os.environ.get("GENREC_DISABLE_ATTN_FLASH", "0") != "1"
                 and act == _ACT_SOFTMAX and q.dtype == torch.bfloat16
                 and q.size(3) % 32 == 0 and q.size(3) <= 64
                 and (additive_mask is None or additive_mask.dim() == 2)

@@ -136,17 +136,26 @@ class GradReducer:
         for bi, ps in enumerate(self._buckets):
             self._pending[bi] = len(ps)
         self._works = []
+        self._next = 0
 
     def _on_grad_ready(self, p: Tensor) -> None:
         if self.world <= 1 or self.skip_sync:
             return
         bi = self._bucket_of[id(p)]
         self._pending[bi] -= 1
-        if self._pending[bi] == 0:
-            self._launch(bi)
+        self._maybe_launch()
 
-    def _launch(self, bi: int, params=None) -> None:
-        ps = params if params is not None else self._buckets[bi]
+    def _maybe_launch(self) -> None:
+        # Launch strictly in bucket-index order: RCCL requires the same
+        # collective sequence on every rank, and ready-order can diverge
+        # when parameter usage is rank-dependent.
+        while (self._next < len(self._buckets)
+               and self._pending[self._next] == 0):
+            self._launch(self._next)
+            self._next += 1
+
+    def _launch(self, bi: int) -> None:
+        ps = self._buckets[bi]
         grads = [p.grad for p in ps]
         flat = torch._utils._flatten_dense_tensors(grads)
         if self._comm_stream is not None:
@@ -165,19 +174,21 @@ class GradReducer:
 
         Buckets that never completed through the hooks (parameters the
         forward did not touch, e.g. TIGER's unused positional embeddings)
-        are reduced here over the SUBSET of parameters that do have
-        grads — otherwise a mixed used/unused bucket would silently stay
-        unsynchronized.
+        are reduced here over the FULL bucket with zero grads substituted
+        for untouched params — every rank contributes an identically
+        shaped flat buffer even when parameter usage is rank-dependent
+        (matching torch-DDP semantics: unused params end up with the
+        cross-rank average, zero if unused everywhere).
         """
         if self.world <= 1 or self.skip_sync:
             return
         for bi in range(len(self._buckets)):
             if self._pending[bi] > 0:
-                with_grads = [p for p in self._buckets[bi]
-                              if p.grad is not None]
-                if with_grads:
-                    self._launch(bi, with_grads)
+                for p in self._buckets[bi]:
+                    if p.grad is None:
+                        p.grad = torch.zeros_like(p)
                 self._pending[bi] = 0
+        self._maybe_launch()
         for bi, ps, work in self._works:
             work.wait()
         if self._comm_stream is not None:

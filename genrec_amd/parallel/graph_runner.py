@@ -124,6 +124,11 @@ class GraphedTrainStep:
         self._loss = None
         if use_graph and device.type == "cuda":
             try:
+                # Warmup + capture + verification replay each run the full
+                # optimizer update on the example batch; snapshot the
+                # optimizer state first and restore it after, so real
+                # training starts from step 0 exactly as eager mode would.
+                snap = self._opt_snapshot()
                 s = torch.cuda.Stream()
                 s.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(s):
@@ -135,6 +140,7 @@ class GraphedTrainStep:
                     self._loss = self._inner()
                 graph.replay()
                 torch.cuda.synchronize()
+                self._opt_restore(snap)
                 self._graph = graph
                 logger.info("GraphedTrainStep: hipGraph capture active")
             except Exception as e:
@@ -142,6 +148,32 @@ class GraphedTrainStep:
                     "GraphedTrainStep: capture failed (%s); running the "
                     "same step eagerly", e)
                 self._graph = None
+
+    def _opt_snapshot(self):
+        with torch.no_grad():
+            if self.fused:
+                return (self.flat_master.clone(), self.m.clone(),
+                        self.v.clone(), self.step_t.clone())
+            import copy
+
+            return ([m.detach().clone() for m in self.masters],
+                    copy.deepcopy(self.opt.state_dict()))
+
+    def _opt_restore(self, snap) -> None:
+        with torch.no_grad():
+            if self.fused:
+                fm, m, v, st = snap
+                self.flat_master.copy_(fm)
+                self.m.copy_(m)
+                self.v.copy_(v)
+                self.step_t.copy_(st)
+                self.flat_params.copy_(self.flat_master.to(torch.bfloat16))
+            else:
+                masters, opt_state = snap
+                for mt, s0 in zip(self.masters, masters):
+                    mt.copy_(s0)
+                self.opt.load_state_dict(opt_state)
+                torch._foreach_copy_(self.params, self.masters)
 
     def _inner(self):
         if self.device.type == "cuda":

@@ -71,14 +71,40 @@ def autocast_ctx(device: torch.device, mixed_precision: Optional[str]):
     return contextlib.nullcontext()
 
 
+class UnpaddedShardSampler(torch.utils.data.Sampler):
+    """Contiguous per-rank eval shard WITHOUT duplicate padding.
+
+    DistributedSampler pads the last shard by repeating samples; the
+    all-reduced Recall/NDCG counters would count those duplicates and bias
+    metrics versus the reference's single-process eval. Here rank r takes
+    indices [r*q + min(r, rem), ...) — shard sizes differ by at most 1 and
+    every sample appears exactly once across ranks.
+    """
+
+    def __init__(self, dataset, num_replicas: int, rank: int):
+        n = len(dataset)
+        q, rem = divmod(n, num_replicas)
+        start = rank * q + min(rank, rem)
+        self._indices = list(range(start, start + q + (1 if rank < rem else 0)))
+
+    def __iter__(self):
+        return iter(self._indices)
+
+    def __len__(self):
+        return len(self._indices)
+
+
 def make_loader(dataset: Dataset, batch_size: int, ctx: DistributedContext,
                 shuffle: bool, collate_fn=None, num_workers: int = 4,
                 drop_last: bool = False, seed: int = 0) -> DataLoader:
     sampler = None
     if ctx.world_size > 1:
-        sampler = DistributedSampler(
-            dataset, num_replicas=ctx.world_size, rank=ctx.rank,
-            shuffle=shuffle, seed=seed, drop_last=drop_last)
+        if shuffle:
+            sampler = DistributedSampler(
+                dataset, num_replicas=ctx.world_size, rank=ctx.rank,
+                shuffle=True, seed=seed, drop_last=drop_last)
+        else:  # eval: no duplicate padding (metrics are all-reduced)
+            sampler = UnpaddedShardSampler(dataset, ctx.world_size, ctx.rank)
         shuffle = False
     return DataLoader(
         dataset, batch_size=batch_size, shuffle=shuffle, sampler=sampler,
@@ -90,9 +116,13 @@ def make_loader(dataset: Dataset, batch_size: int, ctx: DistributedContext,
 def save_checkpoint(path: str, model: torch.nn.Module, optimizer, scheduler,
                     *, epoch: Optional[int] = None, step: Optional[int] = None,
                     model_config: Optional[dict] = None,
-                    is_main: bool = True) -> None:
+                    is_main: bool = True, runner=None) -> None:
     """Dict checkpoint, reference-compatible layout
-    ({epoch|iter, model, model_config, optimizer, scheduler})."""
+    ({epoch|iter, model, model_config, optimizer, scheduler}).
+
+    `runner` (a GraphedTrainStep) saves the fused-path optimizer state —
+    fp32 flat masters, AdamW moments, step counter, lr — under "runner";
+    without it a hipGraph-mode resume would silently restart AdamW."""
     if not is_main:
         return
     os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
@@ -107,6 +137,8 @@ def save_checkpoint(path: str, model: torch.nn.Module, optimizer, scheduler,
         state["optimizer"] = optimizer.state_dict()
     if scheduler is not None:
         state["scheduler"] = scheduler.state_dict()
+    if runner is not None:
+        state["runner"] = runner.state_dict()
     tmp = path + ".tmp"
     torch.save(state, tmp)
     os.replace(tmp, path)
@@ -114,13 +146,16 @@ def save_checkpoint(path: str, model: torch.nn.Module, optimizer, scheduler,
 
 
 def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
-                    scheduler=None, map_location="cpu") -> Dict[str, Any]:
+                    scheduler=None, map_location="cpu",
+                    runner=None) -> Dict[str, Any]:
     state = torch.load(path, map_location=map_location, weights_only=False)
     model.load_state_dict(state["model"])
     if optimizer is not None and "optimizer" in state:
         optimizer.load_state_dict(state["optimizer"])
     if scheduler is not None and "scheduler" in state:
         scheduler.load_state_dict(state["scheduler"])
+    if runner is not None and "runner" in state:
+        runner.load_state_dict(state["runner"])
     logger.info("resumed from %s (epoch=%s iter=%s)", path,
                 state.get("epoch"), state.get("iter"))
     return state

@@ -183,7 +183,7 @@ def train(
             common.save_checkpoint(
                 os.path.join(save_dir_root, f"checkpoint_epoch_{epoch}.pt"),
                 model, runner.opt if runner else opt, None,
-                epoch=epoch, is_main=True)
+                epoch=epoch, is_main=True, runner=runner)
         if max_steps is not None and step >= max_steps:
             break
     if do_eval:

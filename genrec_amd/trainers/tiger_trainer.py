@@ -135,10 +135,11 @@ def train(
     reducer = None if graph_mode else GradReducer(model)
 
     start_epoch, step = 0, 0
+    resume_state = None
     if resume_path and os.path.exists(resume_path):
-        state = common.load_checkpoint(resume_path, model, opt, sched,
-                                       map_location=device)
-        start_epoch = state.get("epoch", -1) + 1
+        resume_state = common.load_checkpoint(resume_path, model, opt, sched,
+                                              map_location=device)
+        start_epoch = resume_state.get("epoch", -1) + 1
 
     wb = common.init_wandb(wandb_project, {"model": "tiger"},
                            wandb_logging, ctx.is_main)
@@ -160,6 +161,9 @@ def train(
             model, example, loss_getter=lambda out: out.loss,
             lr=learning_rate, weight_decay=weight_decay,
             clip_norm=1.0, world=ctx.world_size)
+        if resume_state is not None and "runner" in resume_state:
+            # restore fp32 masters + AdamW moments/step for the fused path
+            runner.load_state_dict(resume_state["runner"])
 
         import math as _math
 
@@ -231,7 +235,8 @@ def train(
             common.save_checkpoint(
                 os.path.join(save_dir_root, f"checkpoint_epoch_{epoch}.pt"),
                 model, runner.opt if runner else opt,
-                None if runner else sched, epoch=epoch, is_main=True)
+                None if runner else sched, epoch=epoch, is_main=True,
+                runner=runner)
         if max_steps is not None and step >= max_steps:
             break
     prof.report()
@@ -240,7 +245,8 @@ def train(
         common.save_checkpoint(
             os.path.join(save_dir_root, "checkpoint_final.pt"),
             model, runner.opt if runner else opt,
-            None if runner else sched, epoch=epochs - 1, is_main=True)
+            None if runner else sched, epoch=epochs - 1, is_main=True,
+            runner=runner)
     wb.finish()
 
 

@@ -751,13 +751,11 @@ def test_mfma_attention_strided_views_match_contiguous():
             (a.float() - b.float()).abs().max()
 
 
-@pytest.mark.skipif(os.environ.get("GENREC_ATTN_FLASH") != "1",
-                    reason="staged flash path; enable GENREC_ATTN_FLASH=1")
 @pytest.mark.parametrize("case", ["plain128", "bias80", "causal128",
                                   "padmask96", "dropout128"])
 def test_flash_attention_vs_eager(case):
-    """Flash-tiled kernels (Lk>64) vs the eager fp32 reference. Skipped
-    until the staged path is enabled (BACKLOG.md item 3)."""
+    """Flash-tiled kernels (Lk>64) vs the eager fp32 reference. Default
+    dispatch path for Lk>64 since round 2."""
     from genrec_amd.ops.attention import fused_attention
 
     torch.manual_seed(0)

@@ -270,16 +270,16 @@ def test_eager_attention_matches_naive_formula(case):
     assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
 
 
-def test_flash_dispatch_inert_by_default(monkeypatch):
-    """Long-Lk attention falls back to eager on CPU regardless of the
-    staged-flash env flag (the flash kernels need the GPU extension)."""
+def test_flash_dispatch_inert_on_cpu(monkeypatch):
+    """Long-Lk attention falls back to eager on CPU (the flash kernels
+    need the GPU extension); the opt-out env changes nothing there."""
     from genrec_amd.ops.attention import fused_attention
 
     q = torch.randn(2, 2, 80, 32)
     k = torch.randn(2, 2, 128, 32)
     v = torch.randn(2, 2, 128, 32)
     out1 = fused_attention(q, k, v, scale=0.1, causal=True)
-    monkeypatch.setenv("GENREC_ATTN_FLASH", "1")
+    monkeypatch.setenv("GENREC_DISABLE_ATTN_FLASH", "1")
     out2 = fused_attention(q, k, v, scale=0.1, causal=True)
     assert torch.allclose(out1, out2)
     assert out1.shape == (2, 2, 80, 32)
