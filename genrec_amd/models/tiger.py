@@ -210,12 +210,43 @@ class Tiger(nn.Module):
             memory_key_padding_mask=memory_mask)
         return self.output_head(out)[:, -1, :]
 
+    def _decode_step_cached(self, memory, memory_mask, last_tok, step,
+                            kv_caches):
+        """Incremental decode: feed only the newest target row, reusing
+        per-layer self/cross K/V caches (redesign of the reference's full
+        re-forward per step, tiger.py:283-310)."""
+        b = memory.size(0)
+        if step == 0:
+            decoder_input = self.bos_embedding.expand(b, 1, -1)
+        else:
+            tgt_type = torch.full((b, 1), step - 1, dtype=torch.long,
+                                  device=memory.device)
+            decoder_input = self.sem_id_embedding(last_tok.view(b, 1),
+                                                  tgt_type)
+        decoder_input = self.in_proj(self.drop(self.norm(decoder_input)))
+        out = self.transformer.decoder(
+            decoder_input, memory=memory,
+            memory_key_padding_mask=memory_mask, kv_caches=kv_caches)
+        return self.output_head(out)[:, -1, :]
+
+    @staticmethod
+    def _permute_kv_caches(kv_caches, flat_parent):
+        """Reorder cached SELF K/V rows after beam re-ranking. Cross-attn
+        K/V rows are identical for all k beams of a user (memory was
+        expanded), so gathering within a user is a no-op — skipped."""
+        for layer_cache in kv_caches:
+            sc = layer_cache.get("self", {})
+            for key in ("k", "v"):
+                if key in sc:
+                    sc[key] = sc[key].index_select(0, flat_parent)
+
     @torch.no_grad()
     def generate(self, user_input_ids: Tensor, item_input_ids: Tensor,
                  token_type_ids: Tensor, seq_mask: Optional[Tensor] = None,
                  temperature: float = 0.2, n_top_k_candidates: int = 10,
                  valid_item_ids: Optional[Tensor] = None,
-                 use_trie: bool = True) -> TigerGenerationOutput:
+                 use_trie: bool = True,
+                 use_kv_cache: bool = True) -> TigerGenerationOutput:
         b, k = user_input_ids.size(0), n_top_k_candidates
         device = user_input_ids.device
         vlevel = self.num_item_embeddings
@@ -236,15 +267,24 @@ class Tiger(nn.Module):
             nodes = torch.ones(b, k, dtype=torch.long, device=device)  # root
 
         kk = min(k * 6, vlevel)  # reference R=6 (tiger.py:350-351)
+        kv_caches = ([{} for _ in self.transformer.decoder.layers]
+                     if use_kv_cache else None)
 
         for step in range(self.sem_id_dim):
-            if step == 0:
-                tgt_ids, tgt_type = None, None
+            if use_kv_cache:
+                last = (beam_seqs[:, :, step - 1].reshape(b * k)
+                        if step > 0 else None)
+                logits = self._decode_step_cached(
+                    memory, memory_mask, last, step, kv_caches)
             else:
-                tgt_ids = beam_seqs[:, :, :step].reshape(b * k, step)
-                tgt_type = torch.arange(step, device=device).unsqueeze(0) \
-                    .expand(b * k, -1)
-            logits = self._decode_step(memory, memory_mask, tgt_ids, tgt_type)
+                if step == 0:
+                    tgt_ids, tgt_type = None, None
+                else:
+                    tgt_ids = beam_seqs[:, :, :step].reshape(b * k, step)
+                    tgt_type = torch.arange(step, device=device) \
+                        .unsqueeze(0).expand(b * k, -1)
+                logits = self._decode_step(memory, memory_mask, tgt_ids,
+                                           tgt_type)
 
             offset = step * vlevel
             full_mask = torch.full_like(logits, False, dtype=torch.bool)
@@ -307,6 +347,10 @@ class Tiger(nn.Module):
                                           new_tok.clamp(min=0).reshape(-1)
                                           ).reshape(b, k)
                 nodes[padding] = 1  # reference resets padded beams to root
+            if use_kv_cache and step < self.sem_id_dim - 1:
+                flat_parent = (torch.arange(b, device=device).unsqueeze(1) * k
+                               + new_parent).reshape(-1)
+                self._permute_kv_caches(kv_caches, flat_parent)
 
         return TigerGenerationOutput(sem_ids=beam_seqs, log_probas=beam_logps)
 

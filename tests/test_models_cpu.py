@@ -266,3 +266,31 @@ def test_hstu_temporal_bucket_formula():
     assert torch.allclose(bias, expect)
     # same timestamp -> bucket 0; huge gap -> clamped top bucket
     assert buckets[0, 0, 0] == 0 and buckets[0, 0, 4] == 7
+
+
+def test_tiger_generate_kv_cache_equivalent():
+    """KV-cached incremental beam decode == the full re-forward path
+    (same RNG stream -> identical Gumbel draws; dropout off). Uses a
+    dense trie (all codes valid) so no beams are NEG_INF-padded."""
+    torch.manual_seed(7)
+    m = Tiger(embedding_dim=16, attn_dim=24, dropout=0.0, num_heads=4,
+              n_layers=2, num_item_embeddings=8, num_user_embeddings=10,
+              sem_id_dim=3)
+    m.eval()
+    B, NI, K = 3, 4, 4
+    L = NI * 3
+    item = torch.randint(0, 8, (B, L))
+    ttype = (torch.arange(L) % 3).unsqueeze(0).expand(B, -1)
+    mask = torch.ones(B, L, dtype=torch.long)
+    valid = torch.cartesian_prod(*[torch.arange(8)] * 3)  # dense trie
+    user = torch.zeros(B, 1, dtype=torch.long)
+
+    torch.manual_seed(123)
+    g1 = m.generate(user, item, ttype, mask, n_top_k_candidates=K,
+                    valid_item_ids=valid, use_kv_cache=False)
+    m.trie = None  # rebuild to keep paths independent
+    torch.manual_seed(123)
+    g2 = m.generate(user, item, ttype, mask, n_top_k_candidates=K,
+                    valid_item_ids=valid, use_kv_cache=True)
+    assert torch.equal(g1.sem_ids, g2.sem_ids)
+    assert torch.allclose(g1.log_probas, g2.log_probas, atol=1e-5)
