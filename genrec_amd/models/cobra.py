@@ -19,10 +19,11 @@ behavior:
   * beam_fusion: alpha-blend of softmaxed beam scores and max item
     similarity (cobra.py:679-760)
 
-MI355X notes: interleaving is pure index arithmetic (gather/scatter —
-K18); the decoder runs on torch TransformerDecoder (self-attention over
-interleaved length ~T*(C+1) <= 80, within ATen/rocBLAS); the InfoNCE sim
-GEMM and heads run on hipBLASLt.
+MI355X notes: interleaving is ONE indexed gather (K18); the decoder is
+native CobraDecoderLayers on ops.fused_attention (flash-tiled MFMA
+kernels at the interleaved length ~T*(C+1) <= 80, head_dim 64); the
+sparse-head CE and dense InfoNCE run on the fused CE kernel (K3); the
+InfoNCE sim GEMM and heads run on hipBLASLt.
 """
 
 from __future__ import annotations
@@ -33,6 +34,7 @@ import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
 
+from genrec_amd import ops
 from genrec_amd.ops.linear import SplitKLinear
 
 from genrec_amd.config import ginlite
@@ -95,25 +97,35 @@ class CobraEmbedding(nn.Module):
             valid, input_ids + ttype * self.id_vocab_size, input_ids)
         sparse_emb = self.id_embed(flat_ids)  # (B, L, D)
 
+        # K18 interleave as ONE indexed gather from [sparse_emb | vecs]
+        # (the reference's chunk/cat loop is ~2T kernel launches,
+        # cobra.py:323-377; a gather is bandwidth-optimal on HBM3E).
+        # Sparse token j (chunk i=j//C) shifts right by the number of
+        # dense slots inserted before it; dense slot i lands at
+        # i*(C+1)+C.
         n_ct = n_complete_items * self.C
-        chunks = []
-        if n_ct > 0:
-            for i, chunk in enumerate(sparse_emb[:, :n_ct].split(self.C, dim=1)):
-                chunks.append(chunk)
-                if i < t_vecs:
-                    chunks.append(input_vecs[:, i].unsqueeze(1))
-        if l - n_ct > 0:
-            chunks.append(sparse_emb[:, n_ct:])
-        h = torch.cat(chunks, dim=1)
+        n_ins = min(n_complete_items, t_vecs)
+        out_len = l + n_ins
+        orig = torch.arange(l, device=device)
+        shift = torch.where(orig < n_ct,
+                            torch.minimum(orig // self.C,
+                                          torch.tensor(n_ins, device=device)),
+                            torch.tensor(n_ins, device=device))
+        new_pos = orig + shift
+        src_idx = torch.empty(out_len, dtype=torch.long, device=device)
+        src_idx.scatter_(0, new_pos, orig)
+        is_dense = torch.zeros(out_len, dtype=torch.bool, device=device)
+        if n_ins > 0:
+            ins_pos = torch.arange(n_ins, device=device) * (self.C + 1) \
+                + self.C
+            src_idx.scatter_(0, ins_pos,
+                             torch.arange(n_ins, device=device) + l)
+            is_dense.scatter_(0, ins_pos, True)
+        src = torch.cat([sparse_emb, input_vecs[:, :n_ins]], dim=1)
+        h = src.gather(1, src_idx.view(1, -1, 1).expand(b, -1, src.size(-1)))
 
-        out_len = h.size(1)
         pos = torch.arange(out_len, device=device).unsqueeze(0).expand(b, -1)
-        types = []
-        for _ in range(n_complete_items):
-            types.extend([0] * self.C + [1])
-        types.extend([0] * (l - n_ct))
-        type_idx = torch.tensor(types[:out_len], device=device).unsqueeze(0) \
-            .expand(b, -1)
+        type_idx = is_dense.long().unsqueeze(0).expand(b, -1)
         m = mask.unsqueeze(-1).float()
         h = h * m
         h = h + self.pos_embed(pos) * m
@@ -121,30 +133,80 @@ class CobraEmbedding(nn.Module):
         return h
 
 
+class CobraDecoderLayer(nn.Module):
+    """Native causal self-attention layer, numerics-equivalent to torch's
+    post-LN ``nn.TransformerDecoderLayer`` driven with a zero-length
+    memory (as the reference does, cobra.py:150-224): the cross-attn
+    sublayer there contributes exactly zero, but its LayerNorm still
+    applies, so the effective layer is
+        x = norm1(x + SA(x)); x = norm2(x); x = norm3(x + FF(x)).
+    Attention runs on ops.fused_attention — the flash-tiled MFMA kernels
+    at the interleaved COBRA length L=(C+1)*T (~80), head_dim 64."""
+
+    def __init__(self, d_model: int, n_heads: int, ff_dim: int,
+                 dropout: float) -> None:
+        super().__init__()
+        assert d_model % n_heads == 0
+        self.h = n_heads
+        self.hd = d_model // n_heads
+        self.scale = 1.0 / (self.hd ** 0.5)
+        self.qkv = SplitKLinear(d_model, 3 * d_model, bias=True)
+        self.out = SplitKLinear(d_model, d_model, bias=True)
+        self.linear1 = SplitKLinear(d_model, ff_dim, bias=True)
+        self.linear2 = SplitKLinear(ff_dim, d_model, bias=True)
+        self.norm1 = nn.LayerNorm(d_model)
+        self.norm2 = nn.LayerNorm(d_model)
+        self.norm3 = nn.LayerNorm(d_model)
+        self.dropout_p = dropout
+
+    def forward(self, x: Tensor, key_pad_mask: Optional[Tensor],
+                query_mask: Optional[Tensor]) -> Tensor:
+        from genrec_amd import ops
+
+        b, l, d = x.shape
+        qkv = self.qkv(x).view(b, l, 3, self.h, self.hd) \
+            .permute(2, 0, 3, 1, 4)
+        att = ops.fused_attention(
+            qkv[0].contiguous(), qkv[1].contiguous(), qkv[2].contiguous(),
+            scale=self.scale, causal=True, key_pad_mask=key_pad_mask,
+            query_mask=query_mask, dropout_p=self.dropout_p,
+            training=self.training)
+        att = self.out(att.transpose(1, 2).reshape(b, l, d))
+        x = self.norm1(x + F.dropout(att, self.dropout_p, self.training))
+        x = self.norm2(x)  # vestigial cross-attn sublayer's norm
+        ff = self.linear2(F.dropout(F.relu(self.linear1(x)),
+                                    self.dropout_p, self.training))
+        return self.norm3(x + F.dropout(ff, self.dropout_p, self.training))
+
+
 class CobraDecoder(nn.Module):
-    """Causal transformer decoder (no cross-attn memory, ref cobra.py:150-224)."""
+    """Causal transformer decoder (no cross-attn memory, ref cobra.py:150-224).
+
+    Round 2: native layers on the fused/flash attention kernels replace
+    the round-1 stock nn.TransformerDecoder (numerics-equivalence covered
+    by tests/test_cobra_notellm.py::test_cobra_decoder_matches_torch)."""
 
     def __init__(self, hidden_dim: int = 768, n_layers: int = 6,
                  n_heads: int = 12, ff_dim: int = 2048,
                  dropout: float = 0.1) -> None:
         super().__init__()
-        layer = nn.TransformerDecoderLayer(
-            d_model=hidden_dim, nhead=n_heads, dim_feedforward=ff_dim,
-            dropout=dropout, batch_first=True)
-        self.decoder = nn.TransformerDecoder(layer, num_layers=n_layers)
+        self.layers = nn.ModuleList([
+            CobraDecoderLayer(hidden_dim, n_heads, ff_dim, dropout)
+            for _ in range(n_layers)])
 
     def forward(self, tgt: Tensor, memory: Optional[Tensor] = None,
                 tgt_key_padding_mask: Optional[Tensor] = None,
                 memory_key_padding_mask: Optional[Tensor] = None) -> Tensor:
-        l = tgt.size(1)
-        causal = torch.triu(
-            torch.ones(l, l, device=tgt.device, dtype=torch.bool), diagonal=1)
-        if memory is None:
-            memory = torch.zeros(tgt.size(0), 0, tgt.size(2),
-                                 dtype=tgt.dtype, device=tgt.device)
-        return self.decoder(tgt, memory, tgt_mask=causal,
-                            tgt_key_padding_mask=tgt_key_padding_mask,
-                            memory_key_padding_mask=memory_key_padding_mask)
+        # query_mask zeroes fully-padded query rows (torch emits NaN
+        # there; those positions are never consumed — losses use
+        # ignore_index and generation gathers valid last positions)
+        qm = None
+        if tgt_key_padding_mask is not None:
+            qm = (~tgt_key_padding_mask).to(tgt.dtype)
+        x = tgt
+        for layer in self.layers:
+            x = layer(x, tgt_key_padding_mask, qm)
+        return x
 
 
 @ginlite.configurable(name="Cobra")
@@ -256,11 +318,12 @@ class Cobra(nn.Module):
                 target_pos = torch.arange(1, t, device=h.device) * self.C + c
             logits = self.sparse_head[c](h[:, pos_c, :])
             target = input_ids[:, target_pos]
-            loss_c = F.cross_entropy(
-                logits.reshape(-1, logits.size(-1)), target.reshape(-1),
-                ignore_index=self.pad_id, reduction="sum")
-            n_valid = (target != self.pad_id).sum()
-            loss_sparse = loss_sparse + loss_c / n_valid.clamp(min=1)
+            # fused CE (K3 kernel): mean over valid == the reference's
+            # sum/n_valid (cobra.py:417-457)
+            loss_c = ops.softmax_ce(
+                logits.reshape(-1, logits.size(-1)).contiguous(),
+                target.reshape(-1), ignore_index=self.pad_id)
+            loss_sparse = loss_sparse + loss_c
             with torch.no_grad():
                 valid = target != self.pad_id
                 if all_valid is None:
@@ -295,7 +358,7 @@ class Cobra(nn.Module):
         sim = (vec_pred @ vec_gt.T) / self.temperature
         sim = sim.masked_fill(same, -1e4)
         labels = torch.arange(sim.size(0), device=sim.device)
-        loss_dense = F.cross_entropy(sim, labels, reduction="mean")
+        loss_dense = ops.softmax_ce(sim.contiguous(), labels)
 
         vec_cos_sim = F.cosine_similarity(vec_pred, vec_gt).mean()
         with torch.no_grad():

@@ -44,11 +44,12 @@ def relative_position_bucket(relative_positions: Tensor, num_buckets: int = 32,
         ret = torch.clamp_min(ret, 0)
     max_exact = num_buckets // 2
     is_small = ret < max_exact
+    # guard degenerate max_distance <= max_exact (log denominator -> 0)
+    log_denom = math.log(max(max_distance / max_exact, 1.0 + 1e-6))
     large = max_exact + (
-        (torch.log(ret.float() / max_exact + 1e-6)
-         / math.log(max_distance / max_exact))
+        (torch.log(ret.float() / max_exact + 1e-6) / log_denom)
         * (num_buckets - max_exact)
-    ).long().clamp(max=num_buckets - max_exact - 1)
+    ).long().clamp(min=0, max=num_buckets - max_exact - 1)
     ret = torch.where(is_small, ret, large)
     if bidirectional:
         ret = ret + sign * num_buckets

@@ -77,13 +77,14 @@ def use_hip(*tensors: torch.Tensor) -> bool:
 from genrec_amd.ops import eager  # noqa: E402
 from genrec_amd.ops.norms import rms_norm, l2norm_op, swish_layer_norm  # noqa: E402
 from genrec_amd.ops.attention import (  # noqa: E402
+    fused_attention,
     sasrec_attention,
     t5_attention,
     hstu_pointwise_attention,
     hstu_fused_attention,
 )
 from genrec_amd.ops.quantize import residual_quantize_step  # noqa: E402
-from genrec_amd.ops.losses import tied_softmax_ce, summed_ce  # noqa: E402
+from genrec_amd.ops.losses import softmax_ce, tied_softmax_ce, summed_ce  # noqa: E402
 from genrec_amd.ops.metrics import topk_hit_ranks  # noqa: E402
 from genrec_amd.ops.embedding import embedding  # noqa: E402
 from genrec_amd.ops.fused import dropout_add, relu_dropout  # noqa: E402
@@ -96,11 +97,13 @@ __all__ = [
     "rms_norm",
     "l2norm_op",
     "swish_layer_norm",
+    "fused_attention",
     "sasrec_attention",
     "t5_attention",
     "hstu_pointwise_attention",
     "hstu_fused_attention",
     "residual_quantize_step",
+    "softmax_ce",
     "tied_softmax_ce",
     "summed_ce",
     "topk_hit_ranks",
