@@ -157,8 +157,43 @@ def bench_lcrec(device, steps, warmup, full_size=False):
                 ms_per_step=el / steps * 1e3)
 
 
+def bench_cobra(device, steps, warmup):
+    """COBRA train step (config/cobra/amazon.gin: B=32, C=3, d_model=384,
+    6 heads, 4 layers, T=20 items -> interleaved L=80 on the flash
+    attention path), bf16 autocast."""
+    from genrec_amd.models.cobra import Cobra
+
+    torch.manual_seed(0)
+    B, T, Ltxt = 32, 20, 64
+    model = Cobra(encoder_n_layers=2, encoder_hidden_dim=384,
+                  encoder_num_heads=6, encoder_vocab_size=32128,
+                  id_vocab_size=256, n_codebooks=3, d_model=384,
+                  decoder_n_layers=4, decoder_num_heads=6,
+                  decoder_dropout=0.1).to(device)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-4, weight_decay=0.01)
+    ids = torch.randint(0, 256, (B, T * 3), device=device)
+    enc = torch.randint(1, 32128, (B, T, Ltxt), device=device)
+    model.train()
+    import contextlib
+
+    amp = (torch.autocast("cuda", dtype=torch.bfloat16)
+           if device.type == "cuda" else contextlib.nullcontext())
+
+    def step(i):
+        opt.zero_grad(set_to_none=False)
+        with amp:
+            out = model(ids, enc)
+        out.loss.backward()
+        opt.step()
+
+    el = _timeit(step, steps, warmup, device)
+    return dict(model="cobra-amazon-beauty", batch=B,
+                samples_per_s=B * steps / el,
+                ms_per_step=el / steps * 1e3)
+
+
 BENCHES = {"sasrec": bench_sasrec, "hstu": bench_hstu, "rqvae": bench_rqvae,
-           "lcrec": bench_lcrec}
+           "lcrec": bench_lcrec, "cobra": bench_cobra}
 
 
 def main():

@@ -1,0 +1,262 @@
+"""Round-2 GPU coverage: COBRA on the flash/native path, the Qwen-backbone
+models (LCRec SFT fwd/bwd + KV-cached constrained beam, NoteLLM contrastive
+step), and a 2-rank RCCL proof on a single GPU (process-group init, bucketed
+GradReducer, hipGraph-captured flat all-reduce).
+"""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+TINY_COBRA = dict(encoder_n_layers=1, encoder_hidden_dim=64,
+                  encoder_num_heads=4, encoder_vocab_size=100,
+                  id_vocab_size=16, n_codebooks=3, d_model=128,
+                  decoder_n_layers=2, decoder_num_heads=4,
+                  decoder_dropout=0.0)
+
+
+def _cobra_batch(C=3, B=3, T=20, L=6, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, 16, (B, T * C), generator=g)
+    enc = torch.randint(1, 100, (B, T, L), generator=g)
+    return ids, enc
+
+
+def test_cobra_shape_hits_flash_kernel():
+    """The COBRA decoder attention shape (L=(C+1)*T=80 > 64, head_dim 32)
+    must dispatch to the flash-tiled kernel, not eager ATen."""
+    from genrec_amd.ops.attention import fused_attention
+
+    q = torch.randn(2, 4, 80, 32, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    out = fused_attention(q, k, v, scale=0.1, causal=True)
+    assert "FlashAttnFn" in type(out.grad_fn).__name__, type(out.grad_fn)
+
+
+def test_cobra_gpu_step_matches_cpu():
+    """COBRA bf16 train step on the native decoder (flash attention +
+    fused CE) vs the same model in fp32 on CPU."""
+    from genrec_amd.models.cobra import Cobra
+
+    torch.manual_seed(0)
+    m = Cobra(**TINY_COBRA)
+    ids, enc = _cobra_batch()
+    m.eval()  # dropout off for comparability
+    out_cpu = m(ids, enc)
+
+    mg = Cobra(**TINY_COBRA)
+    mg.load_state_dict(m.state_dict())
+    mg = mg.to(DEV)
+    mg.eval()
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out_gpu = mg(ids.to(DEV), enc.to(DEV))
+    assert torch.isfinite(out_gpu.loss)
+    assert abs(out_gpu.loss.item() - out_cpu.loss.item()) < 0.15, \
+        (out_gpu.loss.item(), out_cpu.loss.item())
+    # backward produces finite grads through flash bwd + fused CE bwd
+    out_gpu.loss.backward()
+    for n, p in mg.named_parameters():
+        if p.grad is not None:
+            assert torch.isfinite(p.grad).all(), n
+
+
+def test_cobra_gpu_generate_and_fusion():
+    from genrec_amd.models.cobra import Cobra
+
+    torch.manual_seed(0)
+    m = Cobra(**TINY_COBRA).to(DEV)
+    m.eval()
+    ids, enc = _cobra_batch(B=2, T=8)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        gen = m.generate(ids.to(DEV), enc.to(DEV), n_candidates=4)
+    assert gen.sem_ids.shape == (2, 4, 3)
+    assert torch.isfinite(gen.scores).all()
+    n_items, D = 30, 128
+    item_vecs = torch.randn(n_items, D, device=DEV)
+    item_sids = torch.randint(0, 16, (n_items, 3), device=DEV)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        fus = m.beam_fusion(ids.to(DEV), enc.to(DEV), item_vecs, item_sids,
+                            n_candidates=5, n_beam=8)
+    assert fus.item_ids.shape == (2, 5)
+    assert (fus.item_ids < n_items).all()
+
+
+TINY_QWEN = dict(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, intermediate_size=128)
+
+
+def test_lcrec_gpu_sft_step_and_beam():
+    """Qwen-backbone SFT fwd/bwd + AdamW step + KV-cached constrained
+    beam on GPU."""
+    from genrec_amd.models.lcrec import LCRec, default_qwen_config
+
+    torch.manual_seed(0)
+    m = LCRec(config=default_qwen_config(**TINY_QWEN))
+    m.add_codebook_tokens(3, 8)
+    m = m.to(DEV)
+    s = m.tokenize_sft_format("history: <C0_1><C1_2><C2_3>", "<C0_4>")
+    ids = s["input_ids"].to(DEV)
+    am = s["attention_mask"].to(DEV)
+    opt = torch.optim.AdamW(m.parameters(), lr=1e-4)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = m(ids, am, labels=ids)
+    assert torch.isfinite(out.loss)
+    out.loss.backward()
+    opt.step()
+
+    m.eval()
+    cb = m.codebook_token_ids(3, 8).to(DEV)
+    prompt = torch.randint(0, 256, (2, 6), device=DEV)
+    with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16):
+        res = m.generate_topk(prompt, max_new_tokens=3, beam_width=4,
+                              allowed_token_ids=[cb[0], cb[1], cb[2]])
+    assert len(res) == 2 and len(res[0]) == 4
+    allowed = [set(c.tolist()) for c in cb]
+    for b in range(2):
+        for seq, score in res[b]:
+            new = seq[6:].tolist()
+            for lvl in range(3):
+                assert new[lvl] in allowed[lvl]
+
+
+def test_notellm_gpu_contrastive_step():
+    from genrec_amd.models.lcrec import default_qwen_config
+    from genrec_amd.models.notellm import Query2Embedding
+
+    torch.manual_seed(0)
+    m = Query2Embedding(config=default_qwen_config(**TINY_QWEN),
+                        gradient_checkpointing=False).to(DEV)
+    queries = []
+    for i in range(4):
+        queries.append(f"note {i} text [EMB]")
+        queries.append(f"related note {i} [EMB]")
+    tok = m.tokenize(queries)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = m(tok["input_ids"].to(DEV), tok["attention_mask"].to(DEV),
+                tok["emb_token_idx"].to(DEV))
+    assert torch.isfinite(out["loss"])
+    out["loss"].backward()
+    assert m.tau.grad is not None and torch.isfinite(m.tau.grad)
+    emb = out["sentence_embedding"].detach().float()
+    assert torch.allclose(emb.norm(dim=1).cpu(), torch.ones(8), atol=1e-2)
+
+
+# --------------------------------------------------------------- RCCL proof
+
+def _rccl_worker(rank, world, port, results):
+    """2 ranks co-located on ONE GPU: RCCL init, bucketed GradReducer,
+    GraphedTrainStep with a graph-captured flat all-reduce."""
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = "0"  # both ranks share cuda:0
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch
+    import torch.distributed as dist
+
+    from genrec_amd.parallel import GradReducer, init_distributed
+    from genrec_amd.parallel.ddp import broadcast_parameters
+
+    ctx = init_distributed()
+    assert ctx.backend == "nccl"
+    dev = ctx.device
+
+    # 1) plain all-reduce sanity
+    t = torch.full((4,), float(rank + 1), device=dev)
+    dist.all_reduce(t)
+    assert torch.allclose(t, torch.full((4,), 3.0, device=dev))
+
+    # 2) bucketed GradReducer averages grads across ranks
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 32),
+                                torch.nn.Linear(32, 4)).to(dev)
+    broadcast_parameters(model)
+    reducer = GradReducer(model, bucket_cap_mb=0.001)  # force >1 bucket
+    x = torch.randn(8, 16, device=dev,
+                    generator=torch.Generator(dev).manual_seed(100 + rank))
+    model(x).sum().backward()
+    reducer.finalize()
+    g = model[0].weight.grad.clone()
+    # compare to the explicit average of both ranks' grads, computed
+    # locally on every rank (hooks disarmed so no collectives fire)
+    reducer.skip_sync = True
+    xs = [torch.randn(8, 16, device=dev,
+                      generator=torch.Generator(dev).manual_seed(100 + r))
+          for r in range(world)]
+    ref = None
+    for xr in xs:
+        model.zero_grad(set_to_none=True)
+        model(xr).sum().backward()
+        gr = model[0].weight.grad.clone()
+        ref = gr if ref is None else ref + gr
+    ref /= world
+    assert torch.allclose(g, ref, atol=1e-5), (g - ref).abs().max().item()
+
+    # 3) hipGraph capture of a step containing the RCCL all-reduce
+    from genrec_amd.models.tiger import Tiger
+    from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+    tig = Tiger(embedding_dim=16, attn_dim=32, dropout=0.0, num_heads=4,
+                n_layers=2, num_item_embeddings=16, num_user_embeddings=8,
+                sem_id_dim=3).to(dev)
+    broadcast_parameters(tig)
+    B, NI = 4, 4
+    L = NI * 3
+    batch = {
+        "user_input_ids": torch.zeros(B, 1, dtype=torch.long, device=dev),
+        "item_input_ids": torch.randint(0, 16, (B, L), device=dev),
+        "token_type_ids": (torch.arange(L, device=dev) % 3)
+        .unsqueeze(0).expand(B, -1).contiguous(),
+        "target_input_ids": torch.randint(0, 16, (B, 3), device=dev),
+        "target_token_type_ids": torch.arange(3, device=dev)
+        .unsqueeze(0).expand(B, -1).contiguous(),
+        "seq_mask": torch.ones(B, L, dtype=torch.long, device=dev),
+    }
+    runner = GraphedTrainStep(tig, batch, lambda o: o.loss, lr=1e-3,
+                              weight_decay=0.0, clip_norm=1.0,
+                              world=world, use_graph=True)
+    for _ in range(3):
+        loss = runner.step(batch)
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+    # parameters must remain bitwise identical across ranks after graphed
+    # steps (same data + averaged grads)
+    flat = runner.flat_params.clone()
+    flats = [torch.empty_like(flat) for _ in range(world)]
+    dist.all_gather(flats, flat)
+    assert torch.equal(flats[0], flats[1])
+    results[rank] = "ok:captured=%s" % runner.captured
+    dist.destroy_process_group()
+
+
+def test_rccl_two_ranks_one_gpu():
+    """RCCL process-group + bucketed reduce + graphed all-reduce executed
+    on hardware. Two ranks share the single GPU."""
+    import torch.multiprocessing as mp
+
+    if torch.cuda.device_count() < 1:
+        pytest.skip("needs a GPU")
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        procs = [ctx.Process(target=_rccl_worker, args=(r, 2, 29611, results))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+                pytest.fail("RCCL worker hung")
+        assert results.get(0, "").startswith("ok"), dict(results)
+        assert results.get(1, "").startswith("ok"), dict(results)
+        # the graphed path must actually have captured on at least rank 0
+        assert "captured=True" in results[0], dict(results)
