@@ -290,11 +290,23 @@ attn_fwd_mfma_kernel(
   }
 }
 
-// ---- backward dS kernel: dP = dO V^T (MFMA), then Jacobian in-register.
-// Writes dS (bf16, unscaled) and A_d (bf16, the post-mask probabilities)
-// to global for the host-side batched GEMMs.
+// ---- backward dS kernel: dP = dO V^T (MFMA), then Jacobian in-register,
+// then dQ/dK/dV MFMA in the same launch.
+//
+// 8-wave occupancy layout (round 2): 512 threads; wave pair p = wid>>1
+// owns q/j-strip [16p, 16p+16) and half = wid&1 computes dP column
+// fragments {2*half, 2*half+1}. The softmax-Jacobian row dot becomes
+// cross-wave: each half reduces its 2-fragment partial over the 16-lane
+// group and writes it to a per-half LDS slot (dotbuf[2][64], 512 B — no
+// atomics: every slot has exactly one writer); one barrier later both
+// halves read the summed dot. dS/dS^T/A_d^T stores stay disjoint per half
+// (different column fragments -> different j rows of the transposed
+// tiles), and dQ/dK/dV split their OUTPUT d-fragments across the halves
+// with full k-loops. LDS grows by only 512 B, so still 2 blocks/CU but
+// 16 resident waves — double the latency hiding for the staging-bound
+// phases (round-1 PMC: 8 waves, ~49k LDS conflict cycles/dispatch).
 template <bool IS_SILU>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 attn_bwd_ds_kernel(
     const __hip_bfloat16* __restrict__ dout,  // [B,H,Lq,D]
     const __hip_bfloat16* __restrict__ q,     // [B,H,Lq,D]
@@ -328,10 +340,14 @@ attn_bwd_ds_kernel(
   char* dsn = dot + TILE * 128;     // [64(i)][128B(j)] dS (A for dQ)
   char* dst = dsn + TILE * 128;     // [64(j)][128B(i)] dS^T (A for dK)
   char* adt = dst + TILE * 128;     // [64(j)][128B(i)] A_d^T (A for dV)
+  float* dotbuf = reinterpret_cast<float*>(adt + TILE * 128);  // [2][64]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;
+  const int wid = tid >> 6;          // 0..7
+  const int pair = wid >> 1;         // strip owner
+  const int half = wid & 1;          // fragment split
+  const int fbase = half * 2;
 
   for (int idx = tid; idx < TILE * (TILE / 8); idx += blockDim.x) {
     int row = idx / (TILE / 8);
@@ -383,30 +399,31 @@ attn_bwd_ds_kernel(
   }
   __syncthreads();
 
-  const int strip = wid * 16;
+  const int strip = pair * 16;
   const float inv_keep = dropout_p > 0.f ? 1.0f / (1.0f - dropout_p) : 1.0f;
 
-  // dP = dO V^T : A = dO strip rows (k=d), B = V rows (n=j, k=d)
-  float4v acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                    {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  // dP = dO V^T : A = dO strip rows (k=d), B = V rows (n=j, k=d).
+  // This half computes column fragments fbase and fbase+1 only.
+  float4v acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
-  for (int f = 0; f < 4; ++f) {
+  for (int fi = 0; fi < 2; ++fi) {
     for (int kk = 0; kk < D; kk += 32) {
       short8v a = frag_load(dos, strip, kk, lane);
-      short8v bfr = frag_load(vs, f * 16, kk, lane);
-      acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc[f], 0, 0, 0);
+      short8v bfr = frag_load(vs, (fbase + fi) * 16, kk, lane);
+      acc[fi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc[fi],
+                                                        0, 0, 0);
     }
   }
 
   const int col_base = lane & 15;
   const int row_grp = (lane >> 4) << 2;
-  float pv[4][4], da[4][4], ad[4][4];
+  float pv[2][4], da[2][4], ad[2][4];
 #pragma unroll
-  for (int f = 0; f < 4; ++f) {
+  for (int fi = 0; fi < 2; ++fi) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int i = strip + row_grp + r;
-      int j = f * 16 + col_base;
+      int j = (fbase + fi) * 16 + col_base;
       bool ok = (i < Lq) && (j < Lk);
       float p = ok ? p_saved[IDX4M(b, h, i, j, H, Lq, Lk)] : 0.f;
       float m = 1.f;
@@ -416,33 +433,47 @@ attn_bwd_ds_kernel(
           m *= drop_mask[IDX4M(b, h, i, j, H, Lq, Lk)] ? inv_keep : 0.f;
         }
       }
-      pv[f][r] = p;          // P (softmax) or S (silu)
-      da[f][r] = ok ? acc[f][r] * (IS_SILU ? 1.f : m) : 0.f;
-      ad[f][r] = IS_SILU ? (ok ? p * sigmoidf_dev(p) : 0.f) : p * m;
+      pv[fi][r] = p;          // P (softmax) or S (silu)
+      da[fi][r] = ok ? acc[fi][r] * (IS_SILU ? 1.f : m) : 0.f;
+      ad[fi][r] = IS_SILU ? (ok ? p * sigmoidf_dev(p) : 0.f) : p * m;
     }
   }
 
-  float ds[4][4];
+  float ds[2][4];
   if (IS_SILU) {
 #pragma unroll
-    for (int f = 0; f < 4; ++f)
+    for (int fi = 0; fi < 2; ++fi)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float s = pv[f][r];
+        float s = pv[fi][r];
         float sg = sigmoidf_dev(s);
-        ds[f][r] = da[f][r] * sg * (1.f + s * (1.f - sg));
+        ds[fi][r] = da[fi][r] * sg * (1.f + s * (1.f - sg));
       }
   } else {
+    // Jacobian row dot across both halves: reduce this half's 32-column
+    // partial in-wave, publish to dotbuf[half], barrier, read the sum.
+    float part[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float dot = da[0][r] * pv[0][r] + da[1][r] * pv[1][r] +
-                  da[2][r] * pv[2][r] + da[3][r] * pv[3][r];
+      float d0 = da[0][r] * pv[0][r] + da[1][r] * pv[1][r];
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
-        dot += __shfl_xor(dot, off, 64);
+        d0 += __shfl_xor(d0, off, 64);
+      part[r] = d0;
+    }
+    if ((lane & 15) == 0) {
 #pragma unroll
-      for (int f = 0; f < 4; ++f)
-        ds[f][r] = pv[f][r] * (da[f][r] - dot);
+      for (int r = 0; r < 4; ++r)
+        dotbuf[half * TILE + strip + row_grp + r] = part[r];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float dot_r = dotbuf[strip + row_grp + r] +
+                    dotbuf[TILE + strip + row_grp + r];
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        ds[fi][r] = pv[fi][r] * (da[fi][r] - dot_r);
     }
   }
 
@@ -451,18 +482,18 @@ attn_bwd_ds_kernel(
   // columns of row j, so they pack into ONE aligned 8-byte ds_write
   // (4x fewer LDS stores than per-element bf16 scatter -> fewer bank
   // conflict cycles; rocprofv3 SQ_LDS_BANK_CONFLICT was dominated by
-  // these stores).
+  // these stores). Halves write disjoint j rows / column fragments.
   typedef __attribute__((ext_vector_type(4))) short short4v;
   const int i0 = strip + row_grp;  // multiple of 4 -> byte i0*2 is 8B-aligned
 #pragma unroll
-  for (int f = 0; f < 4; ++f) {
+  for (int fi = 0; fi < 2; ++fi) {
     short4v dpack, apack;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int i = i0 + r;
-      int j = f * 16 + col_base;
-      float dval = (i < Lq && j < Lk) ? ds[f][r] : 0.f;
-      float aval = (i < Lq && j < Lk) ? ad[f][r] : 0.f;
+      int j = (fbase + fi) * 16 + col_base;
+      float dval = (i < Lq && j < Lk) ? ds[fi][r] : 0.f;
+      float aval = (i < Lq && j < Lk) ? ad[fi][r] : 0.f;
       __hip_bfloat16 dh = __float2bfloat16(dval * scale);
       __hip_bfloat16 ah = __float2bfloat16(aval);
       dpack[r] = *reinterpret_cast<short*>(&dh);
@@ -472,74 +503,72 @@ attn_bwd_ds_kernel(
         ds_saved[IDX4M(b, h, i, j, H, Lq, Lk)] = dval;
       }
     }
-    int j = f * 16 + col_base;
+    int j = (fbase + fi) * 16 + col_base;
     *reinterpret_cast<short4v*>(dst + swz(j, i0 * 2)) = dpack;
     *reinterpret_cast<short4v*>(adt + swz(j, i0 * 2)) = apack;
   }
-  // dQ uses only this wave's dS rows — no barrier needed yet
-  __builtin_amdgcn_wave_barrier();
+  // dQ's A-operand (dsn strip rows) now mixes both halves' column
+  // fragments, and dK/dV read all pairs' dS^T/A_d^T columns: one
+  // workgroup barrier covers every consumer below.
+  __syncthreads();
 
   const int nfrag_d = (D + 15) / 16;
-  {  // dQ[strip rows] = (scale*dS) @ K  : A = dsn strip, B = kt d-rows
-    float4v accq[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  {  // dQ[strip rows] = (scale*dS) @ K : output d-fragments split by half
+    float4v accq[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      if (f >= nfrag_d) break;
+    for (int fi = 0; fi < 2; ++fi) {
+      if (fbase + fi >= nfrag_d) break;
       for (int kk = 0; kk < TILE; kk += 32) {
         short8v a = frag_load(dsn, strip, kk, lane);
-        short8v bfr = frag_load(kt, f * 16, kk, lane);
-        accq[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, accq[f],
-                                                          0, 0, 0);
+        short8v bfr = frag_load(kt, (fbase + fi) * 16, kk, lane);
+        accq[fi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, accq[fi],
+                                                           0, 0, 0);
       }
     }
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      if (f >= nfrag_d) break;
+    for (int fi = 0; fi < 2; ++fi) {
+      if (fbase + fi >= nfrag_d) break;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int i = strip + row_grp + r;
-        int d = f * 16 + col_base;
+        int d = (fbase + fi) * 16 + col_base;
         if (i < Lq && d < D) {
           dq_out[(int64_t)b * dq_sb + h * dq_sh + i * dq_sl + d] =
-              __float2bfloat16(accq[f][r]);
+              __float2bfloat16(accq[fi][r]);
         }
       }
     }
   }
-  __syncthreads();  // dK/dV read other waves' dS^T / A_d^T columns
 
   {  // dK[j strip] = (scale*dS)^T @ Q ; dV[j strip] = A_d^T @ dO
-    float4v acck[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-    float4v accv[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                       {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    float4v acck[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    float4v accv[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      if (f >= nfrag_d) break;
+    for (int fi = 0; fi < 2; ++fi) {
+      if (fbase + fi >= nfrag_d) break;
       for (int kk = 0; kk < TILE; kk += 32) {
         short8v a1 = frag_load(dst, strip, kk, lane);
-        short8v b1 = frag_load(qt, f * 16, kk, lane);
-        acck[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acck[f],
-                                                          0, 0, 0);
+        short8v b1 = frag_load(qt, (fbase + fi) * 16, kk, lane);
+        acck[fi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acck[fi],
+                                                           0, 0, 0);
         short8v a2 = frag_load(adt, strip, kk, lane);
-        short8v b2 = frag_load(dot, f * 16, kk, lane);
-        accv[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, accv[f],
-                                                          0, 0, 0);
+        short8v b2 = frag_load(dot, (fbase + fi) * 16, kk, lane);
+        accv[fi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, accv[fi],
+                                                           0, 0, 0);
       }
     }
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      if (f >= nfrag_d) break;
+    for (int fi = 0; fi < 2; ++fi) {
+      if (fbase + fi >= nfrag_d) break;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int j = strip + row_grp + r;
-        int d = f * 16 + col_base;
+        int d = (fbase + fi) * 16 + col_base;
         if (j < Lk && d < D) {
           dk_out[(int64_t)b * dk_sb + h * dk_sh + j * dk_sl + d] =
-              __float2bfloat16(acck[f][r]);
+              __float2bfloat16(acck[fi][r]);
           dv_out[(int64_t)b * dv_sb + h * dv_sh + j * dv_sl + d] =
-              __float2bfloat16(accv[f][r]);
+              __float2bfloat16(accv[fi][r]);
         }
       }
     }
@@ -649,9 +678,9 @@ std::vector<torch::Tensor> attn_bwd_mfma(
   torch::Tensor qm_f;
   if (query_mask.has_value())
     qm_f = query_mask->to(torch::kFloat32).contiguous();
-  dim3 block(256);
+  dim3 block(512);  // 8 waves: pairs own strips, halves split fragments
   dim3 grid(B * H);
-  size_t smem = 8 * TILE * 128;
+  size_t smem = 8 * TILE * 128 + 2 * TILE * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_BWD_DS(SILU)                                                    \
