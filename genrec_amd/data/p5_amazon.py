@@ -108,6 +108,7 @@ class AmazonReviews:
             texts.append(f"{m.get('title') or ''} {m.get('brand') or ''} "
                          f"{cats} {m.get('price') or ''}".strip()
                          or f"item {i}")
+        self.item_texts = texts
         if encoder_model_name and encoder_model_name != "light" \
                 and os.path.exists(encoder_model_name):
             from sentence_transformers import SentenceTransformer
@@ -119,6 +120,66 @@ class AmazonReviews:
             from genrec_amd.data.amazon import _hashed_text_embeddings
 
             self.item_embeddings = _hashed_text_embeddings(texts, embed_dim)
+
+    # ------------------------------------------------ processed artifact
+
+    def processed_path(self) -> str:
+        return os.path.join(self.root, "processed",
+                            f"data_{self.split}.pt")
+
+    def build_processed(self, max_seq_len: int = 20) -> dict:
+        """Dependency-free equivalent of the reference's HeteroData
+        artifact (p5_amazon.py:322-368): per-split histories with the
+        leave-two-out layout ([-1]-padded eval windows), item embedding
+        matrix, item text, and the 95/5 is_train mask (generator seed 42).
+        Layout: plain dict of tensors, loadable anywhere torch is."""
+        hist = {"train": {"itemId": [], "itemId_fut": []},
+                "val": {"itemId": [], "itemId_fut": []},
+                "test": {"itemId": [], "itemId_fut": []}}
+        user_ids = []
+        for u, seq in enumerate(self.sequences):
+            items = [i - 1 for i in seq]  # remap to 0-based (ref :281)
+            if len(items) < 3:
+                continue
+            user_ids.append(u)
+            hist["train"]["itemId"].append(torch.tensor(items[:-2]))
+            hist["train"]["itemId_fut"].append(items[-2])
+            ev = items[-(max_seq_len + 2):-2]
+            hist["val"]["itemId"].append(torch.tensor(
+                ev + [-1] * (max_seq_len - len(ev))))
+            hist["val"]["itemId_fut"].append(items[-2])
+            te = items[-(max_seq_len + 1):-1]
+            hist["test"]["itemId"].append(torch.tensor(
+                te + [-1] * (max_seq_len - len(te))))
+            hist["test"]["itemId_fut"].append(items[-1])
+        for sp in ("val", "test"):
+            hist[sp]["itemId"] = torch.stack(hist[sp]["itemId"]) \
+                if hist[sp]["itemId"] else torch.zeros(0, max_seq_len,
+                                                       dtype=torch.long)
+        for sp in hist:
+            hist[sp]["itemId_fut"] = torch.tensor(hist[sp]["itemId_fut"])
+            hist[sp]["userId"] = torch.tensor(user_ids)
+        gen = torch.Generator()
+        gen.manual_seed(42)
+        n = self.item_embeddings.shape[0]
+        return {
+            "history": hist,
+            "item_x": torch.from_numpy(np.ascontiguousarray(
+                self.item_embeddings)),
+            "item_text": list(getattr(self, "item_texts", [])),
+            "item_is_train": torch.rand(n, generator=gen) > 0.05,
+            "max_seq_len": max_seq_len,
+        }
+
+    def save_processed(self, max_seq_len: int = 20) -> str:
+        path = self.processed_path()
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        torch.save(self.build_processed(max_seq_len), path)
+        return path
+
+    @staticmethod
+    def load_processed(path: str) -> dict:
+        return torch.load(path, map_location="cpu", weights_only=False)
 
     def rolling_windows(self, window_size: int = 20, stride: int = 1,
                         train_split: float = 0.8):

@@ -76,3 +76,38 @@ def test_p5_item_and_seq_datasets(p5_root, tmp_path):
         rqvae_embed_dim=8, rqvae_hidden_dims=[16], rqvae_codebook_size=8,
         rqvae_n_layers=3)
     assert len(ts) == 10  # one per user
+
+
+def test_processed_artifact_roundtrip(tmp_path):
+    """build_processed mirrors the reference HeteroData artifact layout
+    (p5_amazon.py:322-368): leave-two-out histories with -1-padded eval
+    windows, item_x embeddings, 95/5 is_train mask (seed 42)."""
+    import json
+
+    from genrec_amd.data.p5_amazon import AmazonReviews
+
+    base = tmp_path / "beauty"
+    base.mkdir()
+    seqs = [[1, 2, 3, 4, 5], [2, 3, 4], [5, 1, 2, 3]]
+    (base / "sequential_data.txt").write_text("\n".join(
+        f"{u} " + " ".join(map(str, s)) for u, s in enumerate(seqs)))
+    (base / "datamaps.json").write_text(json.dumps(
+        {"id2item": {str(i): f"A{i}" for i in range(1, 6)}}))
+    ar = AmazonReviews(root=str(tmp_path), split="beauty", embed_dim=32)
+    path = ar.save_processed(max_seq_len=4)
+    art = AmazonReviews.load_processed(path)
+
+    h = art["history"]
+    # first user 1..5 -> 0-based 0..4: train=[0,1,2], fut=3;
+    # test window = last max_seq_len+1 minus the target = [0,1,2,3], fut=4
+    assert h["train"]["itemId"][0].tolist() == [0, 1, 2]
+    assert h["train"]["itemId_fut"][0].item() == 3
+    assert h["test"]["itemId"][0].tolist() == [0, 1, 2, 3]
+    assert h["test"]["itemId_fut"][0].item() == 4
+    assert h["val"]["itemId"][0].tolist() == [0, 1, 2, -1]
+    # second user len 3: train=[1], fut=2
+    assert h["train"]["itemId"][1].tolist() == [1]
+    assert art["item_x"].shape == (5, 32)
+    assert len(art["item_text"]) == 5
+    assert art["item_is_train"].dtype == torch.bool
+    assert art["item_is_train"].shape == (5,)
