@@ -61,22 +61,26 @@ def main():
     with torch.no_grad():
         for B in (1, 32, args.batch):
             batches = [make_batch(B, 7 + i) for i in range(4)]
-            for i in range(args.warmup):
-                model.generate(**batches[i % 4], n_top_k_candidates=args.topk,
-                               valid_item_ids=valid)
-            if device.type == "cuda":
-                torch.cuda.synchronize()
-            t0 = time.perf_counter()
-            for i in range(args.steps):
-                model.generate(**batches[i % 4], n_top_k_candidates=args.topk,
-                               valid_item_ids=valid)
-            if device.type == "cuda":
-                torch.cuda.synchronize()
-            el = time.perf_counter() - t0
-            results[f"batch_{B}"] = {
-                "users_per_s": B * args.steps / el,
-                "ms_per_batch": el / args.steps * 1e3,
-            }
+            for kv in (True, False):
+                for i in range(args.warmup):
+                    model.generate(**batches[i % 4],
+                                   n_top_k_candidates=args.topk,
+                                   valid_item_ids=valid, use_kv_cache=kv)
+                if device.type == "cuda":
+                    torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for i in range(args.steps):
+                    model.generate(**batches[i % 4],
+                                   n_top_k_candidates=args.topk,
+                                   valid_item_ids=valid, use_kv_cache=kv)
+                if device.type == "cuda":
+                    torch.cuda.synchronize()
+                el = time.perf_counter() - t0
+                key = f"batch_{B}" + ("" if kv else "_nocache")
+                results[key] = {
+                    "users_per_s": B * args.steps / el,
+                    "ms_per_batch": el / args.steps * 1e3,
+                }
     print(json.dumps({
         "metric": "tiger_generate", "topk": args.topk,
         "n_items": n_items, "device": str(device),
