@@ -194,7 +194,11 @@ def fused_attention(
     act = _ACT_SOFTMAX if score_act == "softmax" else _ACT_SILU
     fits = (k.size(2) <= 64 and q.size(2) <= 64 and q.size(3) <= 64
             and (additive_mask is None or additive_mask.dim() == 2)
-            and os.environ.get("GENREC_DISABLE_ATTN", "0") != "1")
+            and os.environ.get("GENREC_DISABLE_ATTN", "0") != "1"
+            # experiment knob: route even Lk<=64 through the flash
+            # (recompute-P) kernels to trade HBM p_saved traffic for
+            # recompute — measured per-config on GPU
+            and os.environ.get("GENREC_FORCE_FLASH", "0") != "1")
     if fits and _kernel_available("attn_fwd", q, k, v):
         b = bias.contiguous() if bias is not None else None
         kp = key_pad_mask.contiguous() if key_pad_mask is not None else None
