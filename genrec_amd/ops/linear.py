@@ -39,12 +39,38 @@ def _pick_chunks(k: int) -> int:
     return 1
 
 
+def _use_skinny(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    """Hand MFMA GEMM for the tall-skinny fwd family (BACKLOG r1 item 4):
+    bf16, K % 64 == 0, many rows. Opt-in (GENREC_SKINNY_GEMM=1) until the
+    measured win generalizes across shapes."""
+    import os
+
+    if os.environ.get("GENREC_SKINNY_GEMM", "0") != "1":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16):
+        return False
+    k = weight.shape[1]
+    rows = x.numel() // x.shape[-1]
+    from genrec_amd import ops
+
+    return k % 64 == 0 and rows >= 4096 and ops.has_ext()
+
+
 class _SplitKLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, weight: torch.Tensor,
                 bias: Optional[torch.Tensor]):
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
+        if _use_skinny(x, weight):
+            from genrec_amd import ops
+
+            x2 = x.reshape(-1, x.shape[-1]).contiguous()
+            out = ops.ext().skinny_gemm(
+                x2, weight.contiguous(),
+                bias if bias is not None else None)
+            return out.view(*x.shape[:-1], weight.shape[0])
         out = x.matmul(weight.t())
         if bias is not None:
             out = out + bias

@@ -28,6 +28,7 @@ sources = [
     "csrc/kernels/fused_elementwise.hip",
     "csrc/kernels/adamw.hip",
     "csrc/kernels/attention_flash.hip",
+    "csrc/kernels/skinny_gemm.hip",
 ]
 
 cxx_flags = ["-O3", "-std=c++17"]

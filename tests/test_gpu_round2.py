@@ -260,3 +260,25 @@ def test_rccl_two_ranks_one_gpu():
         assert results.get(1, "").startswith("ok"), dict(results)
         # the graphed path must actually have captured on at least rank 0
         assert "captured=True" in results[0], dict(results)
+
+
+@pytest.mark.parametrize("shape", [(15616, 384, 384), (15616, 768, 384),
+                                   (15616, 1024, 384), (15616, 384, 1024),
+                                   (6400, 256, 64), (1000, 400, 128)])
+def test_skinny_gemm_matches_matmul(shape):
+    """Hand tall-skinny MFMA GEMM vs hipBLASLt (same bf16 inputs)."""
+    from genrec_amd import ops
+
+    M, N, K = shape
+    torch.manual_seed(0)
+    x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(N, K, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(N, device=DEV, dtype=torch.bfloat16)
+    y = ops.ext().skinny_gemm(x, w, b)
+    ref = (x.float() @ w.float().t() + b.float())
+    err = (y.float() - ref).abs()
+    denom = ref.abs() + 1.0
+    assert (err / denom).max() < 0.02, (err / denom).max()
+    y2 = ops.ext().skinny_gemm(x, w, None)
+    ref2 = x.float() @ w.float().t()
+    assert ((y2.float() - ref2).abs() / (ref2.abs() + 1.0)).max() < 0.02
