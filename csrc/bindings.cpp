@@ -79,6 +79,8 @@ void fused_adamw(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
                  double beta2, double eps, double weight_decay);
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias);
+torch::Tensor skinny_gemm_tn(torch::Tensor x, torch::Tensor w,
+                             c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> attn_fwd_flash(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> key_pad,
@@ -118,6 +120,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused flat AdamW step (device lr/scale/step scalars)");
   m.def("skinny_gemm", &genrec::skinny_gemm,
         "tall-skinny linear fwd GEMM (bf16, fp32 accum)");
+  m.def("skinny_gemm_tn", &genrec::skinny_gemm_tn,
+        "tall-skinny dX GEMM: x[M,K] @ w[K,N] (transposed-B staging)");
   m.def("attn_fwd_flash", &genrec::attn_fwd_flash,
         "flash-tiled attention fwd (staged; GENREC_ATTN_FLASH=1)");
   m.def("attn_bwd_flash", &genrec::attn_bwd_flash,

@@ -18,8 +18,8 @@ template <typename T>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               const int64_t* __restrict__ targets,
                               float* __restrict__ lse,
-                              float* __restrict__ loss_sum,
-                              int* __restrict__ n_valid,
+                              float* __restrict__ row_loss,
+                              float* __restrict__ row_valid,
                               int64_t n_rows, int64_t V, int64_t ignore_index) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -35,11 +35,13 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
     float l = m + __logf(s);
     int64_t t = targets[row];
     if (lane == 0) {
+      // per-row contributions; reduced by ONE fixed-order ATen sum on the
+      // host side — bitwise deterministic (fp32 atomicAdd into a scalar
+      // was run-order dependent; caught by tools/race_check.py)
       lse[row] = l;
-      if (t != ignore_index) {
-        atomicAdd(loss_sum, l - to_f32(lr[t]));
-        atomicAdd(n_valid, 1);
-      }
+      bool ok = t != ignore_index;
+      row_loss[row] = ok ? (l - to_f32(lr[t])) : 0.f;
+      row_valid[row] = ok ? 1.f : 0.f;
     }
   }
 }
@@ -79,8 +81,7 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
   TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 && logits.is_contiguous());
   const int64_t n = logits.size(0), V = logits.size(1);
   auto lse = torch::empty({n}, logits.options().dtype(torch::kFloat32));
-  auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
-  auto n_valid = torch::zeros({1}, logits.options().dtype(torch::kInt32));
+  auto rowbuf = torch::empty({2, n}, logits.options().dtype(torch::kFloat32));
   dim3 block(256);
   int64_t blocks = std::min<int64_t>((n + 3) / 4, 4096);
   dim3 grid((unsigned)blocks);
@@ -88,19 +89,20 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
   if (logits.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((ce_fwd_kernel<float>), grid, block, 0, stream,
                        logits.data_ptr<float>(), targets.data_ptr<int64_t>(),
-                       lse.data_ptr<float>(), loss.data_ptr<float>(),
-                       n_valid.data_ptr<int>(), n, V, ignore_index);
+                       lse.data_ptr<float>(), rowbuf.data_ptr<float>(),
+                       rowbuf.data_ptr<float>() + n, n, V, ignore_index);
   } else if (logits.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL((ce_fwd_kernel<__hip_bfloat16>), grid, block, 0, stream,
                        reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
                        targets.data_ptr<int64_t>(), lse.data_ptr<float>(),
-                       loss.data_ptr<float>(), n_valid.data_ptr<int>(), n, V,
-                       ignore_index);
+                       rowbuf.data_ptr<float>(), rowbuf.data_ptr<float>() + n,
+                       n, V, ignore_index);
   } else {
     TORCH_CHECK(false, "softmax_ce: unsupported dtype");
   }
-  auto n_valid_f = n_valid.to(torch::kFloat32);
-  auto loss_mean = loss / n_valid_f.clamp_min(1.0).squeeze();
+  auto sums = rowbuf.sum(1);  // fixed-order tree reduce: deterministic
+  auto n_valid_f = sums[1].reshape({1}).contiguous();
+  auto loss_mean = sums[0] / sums[1].clamp_min(1.0);
   return {loss_mean, lse, n_valid_f};
 }
 

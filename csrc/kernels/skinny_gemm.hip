@@ -46,9 +46,32 @@ __device__ __forceinline__ short8v ld_frag(const char* base, int row0,
   return *reinterpret_cast<const short8v*>(base + swz128(row, byte));
 }
 
+// In-register 8x8 bf16 block transpose across the 8-lane group (same
+// butterfly as attention_mfma.hip's V^T staging).
+__device__ __forceinline__ void xpose8x8g(short (&vals)[8], int g) {
+#pragma unroll
+  for (int m = 1; m < 8; m <<= 1) {
+    short nv[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int t = __shfl_xor((int)vals[e ^ m], 8 * m, 64);
+      nv[e] = ((e & m) != (g & m)) ? (short)t : vals[e];
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) vals[e] = nv[e];
+  }
+}
+
+// TRANS_B=false: y[M,N] = x[M,K] @ w[N,K]^T  (linear forward; w row-major
+//                in K — B image rows are w rows, staged directly)
+// TRANS_B=true:  y[M,N] = x[M,K] @ w[K,N]    (linear dX backward; w
+//                row-major in N — B image rows are w COLUMNS, staged via
+//                the 8x8 in-register transpose)
+// K is always the contraction dim of x's last axis.
+template <bool TRANS_B>
 __global__ void __launch_bounds__(256)
 skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,  // [M,K]
-                   const __hip_bfloat16* __restrict__ w,  // [N,K]
+                   const __hip_bfloat16* __restrict__ w,
                    const __hip_bfloat16* __restrict__ bias,  // null | [N]
                    __hip_bfloat16* __restrict__ y,        // [M,N]
                    int M, int N, int K) {
@@ -83,17 +106,46 @@ skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,  // [M,K]
       }
       *reinterpret_cast<short8v*>(xs + swz128(row, c8 * 2)) = val;
     }
-    // stage W slice: 384 rows x 64 k
-    for (int idx = tid; idx < BN * (BK / 8); idx += blockDim.x) {
-      int row = idx / (BK / 8);
-      int c8 = (idx % (BK / 8)) * 8;
-      short8v val = {};
-      int ni = n0 + row;
-      if (ni < N) {
-        val = *reinterpret_cast<const short8v*>(
-            &w[(int64_t)ni * K + k0 + c8]);
+    if (!TRANS_B) {
+      // stage W slice: 384 n-rows x 64 k, rows read natural-coalesced
+      for (int idx = tid; idx < BN * (BK / 8); idx += blockDim.x) {
+        int row = idx / (BK / 8);
+        int c8 = (idx % (BK / 8)) * 8;
+        short8v val = {};
+        int ni = n0 + row;
+        if (ni < N) {
+          val = *reinterpret_cast<const short8v*>(
+              &w[(int64_t)ni * K + k0 + c8]);
+        }
+        *reinterpret_cast<short8v*>(ws + swz128(row, c8 * 2)) = val;
       }
-      *reinterpret_cast<short8v*>(ws + swz128(row, c8 * 2)) = val;
+    } else {
+      // stage W^T slice: source slab w[k0..k0+64, n0..n0+384] read
+      // row-natural (coalesced 16 B chunks), 8x8-transposed in-register,
+      // landing as [384 n-rows][64 k] in LDS. Row mapping keeps the
+      // 8-lane shuffle group on 8 consecutive source k-rows.
+      const int g = (lane >> 3) & 7;
+      for (int idx = tid; idx < (BK / 8) * BN; idx += blockDim.x) {
+        int j = idx >> 6;                         // n chunk (8 cols)
+        int sub = idx & 63;
+        int row = ((sub & 7) << 3) | (sub >> 3);  // k row, group-aligned
+        short8v val = {};
+        int ki = k0 + row;
+        int nj = n0 + 8 * j;
+        if (ki < K && nj < N) {
+          val = *reinterpret_cast<const short8v*>(
+              &w[(int64_t)ki * N + nj]);
+        }
+        short tv[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) tv[e] = val[e];
+        xpose8x8g(tv, g);
+        short8v pack;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) pack[e] = tv[e];
+        *reinterpret_cast<short8v*>(
+            ws + swz128(8 * j + g, ((row & ~7) * 2))) = pack;
+      }
     }
     __syncthreads();
 
@@ -138,15 +190,22 @@ skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,  // [M,K]
 
 }  // namespace
 
-torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
-                          c10::optional<torch::Tensor> bias) {
+static torch::Tensor launch_skinny(torch::Tensor x, torch::Tensor w,
+                                   c10::optional<torch::Tensor> bias,
+                                   bool trans_b) {
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
               w.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.is_contiguous() &&
               w.is_contiguous());
-  const int M = x.size(0), K = x.size(1), N = w.size(0);
-  TORCH_CHECK(w.size(1) == K && K % BK == 0,
-              "skinny_gemm: K must be a multiple of 64");
+  const int M = x.size(0), K = x.size(1);
+  const int N = trans_b ? w.size(1) : w.size(0);
+  if (trans_b) {
+    TORCH_CHECK(w.size(0) == K && N % 8 == 0,
+                "skinny_gemm_tn: w must be [K,N], N % 8 == 0");
+  } else {
+    TORCH_CHECK(w.size(1) == K && K % BK == 0,
+                "skinny_gemm: K must be a multiple of 64");
+  }
   auto y = torch::empty({M, N}, x.options());
   torch::Tensor bias_c;
   const __hip_bfloat16* bias_p = nullptr;
@@ -158,13 +217,28 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
   size_t smem = (BM + BN) * 128;
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, smem, stream,
-      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
-      bias_p,
-      reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
-      M, N, K);
+#define LAUNCH_SKINNY(TB)                                                     \
+  hipLaunchKernelGGL((skinny_gemm_kernel<TB>), grid, block, smem, stream,     \
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),                  \
+      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                  \
+      bias_p,                                                                 \
+      reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),                        \
+      M, N, K)
+  if (trans_b) LAUNCH_SKINNY(true);
+  else LAUNCH_SKINNY(false);
+#undef LAUNCH_SKINNY
   return y;
+}
+
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias) {
+  return launch_skinny(x, w, bias, false);
+}
+
+// dX backward: y[M,N] = x[M,K] @ w[K,N] (w used un-transposed / B^T)
+torch::Tensor skinny_gemm_tn(torch::Tensor x, torch::Tensor w,
+                             c10::optional<torch::Tensor> bias) {
+  return launch_skinny(x, w, bias, true);
 }
 
 }  // namespace genrec

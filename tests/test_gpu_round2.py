@@ -328,3 +328,18 @@ def test_skinny_gemm_matches_matmul(shape):
     y2 = ops.ext().skinny_gemm(x, w, None)
     ref2 = x.float() @ w.float().t()
     assert ((y2.float() - ref2).abs() / (ref2.abs() + 1.0)).max() < 0.02
+
+
+@pytest.mark.parametrize("shape", [(15616, 384, 384), (15616, 384, 1024),
+                                   (1000, 384, 128)])
+def test_skinny_gemm_tn_matches_matmul(shape):
+    """Transposed-B variant (dX backward): y = x @ w with w [K,N]."""
+    from genrec_amd import ops
+
+    M, K, N = shape
+    torch.manual_seed(0)
+    x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(K, N, device=DEV, dtype=torch.bfloat16)
+    y = ops.ext().skinny_gemm_tn(x, w, None)
+    ref = x.float() @ w.float()
+    assert ((y.float() - ref).abs() / (ref.abs() + 1.0)).max() < 0.02
