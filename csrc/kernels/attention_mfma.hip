@@ -318,7 +318,9 @@ attn_bwd_ds_kernel(
     __hip_bfloat16* __restrict__ dq_out,      // [B,H,Lq,D]
     __hip_bfloat16* __restrict__ dk_out,      // [B,H,Lk,D]
     __hip_bfloat16* __restrict__ dv_out,      // [B,H,Lk,D]
-    float* __restrict__ ds_saved,             // null | [B,H,Lq,Lk] bias grad
+    __hip_bfloat16* __restrict__ ds_saved,    // null | [B,H,Lq,Lk] bias grad
+                                              // (bf16: halves the HBM
+                                              // round-trip; summed fp32)
     int B, int H, int Lq, int Lk, int D,
     int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t q_sb, int64_t q_sh, int64_t q_sl,
@@ -500,7 +502,7 @@ attn_bwd_ds_kernel(
       apack[r] = *reinterpret_cast<short*>(&ah);
       *reinterpret_cast<__hip_bfloat16*>(dsn + swz(i, j * 2)) = dh;
       if (ds_saved && i < Lq && j < Lk) {
-        ds_saved[IDX4M(b, h, i, j, H, Lq, Lk)] = dval;
+        ds_saved[IDX4M(b, h, i, j, H, Lq, Lk)] = __float2bfloat16(dval);
       }
     }
     int j = (fbase + fi) * 16 + col_base;
@@ -669,11 +671,11 @@ std::vector<torch::Tensor> attn_bwd_mfma(
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   torch::Tensor ds_saved;
-  float* ds_ptr = nullptr;
+  __hip_bfloat16* ds_ptr = nullptr;
   if (bias_grad) {
     ds_saved = torch::empty({B, H, Lq, Lk},
-                            q.options().dtype(torch::kFloat32));
-    ds_ptr = ds_saved.data_ptr<float>();
+                            q.options().dtype(torch::kBFloat16));
+    ds_ptr = reinterpret_cast<__hip_bfloat16*>(ds_saved.data_ptr());
   }
   torch::Tensor qm_f;
   if (query_mask.has_value())
@@ -711,7 +713,10 @@ std::vector<torch::Tensor> attn_bwd_mfma(
 
   torch::Tensor dbias;
   if (bias_grad) {
-    dbias = (bias_dim == 3) ? ds_saved.sum(0) : ds_saved;
+    // bf16 per-element grads, fp32-accumulated batch reduce
+    dbias = (bias_dim == 3)
+        ? ds_saved.sum(0, /*keepdim=*/false, torch::kFloat32)
+        : ds_saved.to(torch::kFloat32);
   } else {
     dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
   }
