@@ -577,6 +577,19 @@ attn_bwd_ds_kernel(
   }
 }
 
+// Batch-reduce of the bf16 bias-grad buffer: out[h,i,j] = sum_b ds[b,h,i,j]
+// in fp32 — replaces an ATen sum(0) that cost ~11.7 us/layer (BACKLOG r1
+// item 2b). Fixed-order loop over B: deterministic.
+__global__ void batch_sum_bf16_kernel(const __hip_bfloat16* __restrict__ in,
+                                      float* __restrict__ out,
+                                      int B, int64_t inner) {
+  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= inner) return;
+  float acc = 0.f;
+  for (int b = 0; b < B; ++b) acc += to_f32(in[(int64_t)b * inner + j]);
+  out[j] = acc;
+}
+
 // ------------------------------------------------------------------ hosts
 
 std::vector<torch::Tensor> attn_fwd_mfma(
@@ -713,10 +726,18 @@ std::vector<torch::Tensor> attn_bwd_mfma(
 
   torch::Tensor dbias;
   if (bias_grad) {
-    // bf16 per-element grads, fp32-accumulated batch reduce
-    dbias = (bias_dim == 3)
-        ? ds_saved.sum(0, /*keepdim=*/false, torch::kFloat32)
-        : ds_saved.to(torch::kFloat32);
+    if (bias_dim == 3) {  // bf16 per-element grads, fp32 batch reduce
+      const int64_t inner = (int64_t)H * Lq * Lk;
+      dbias = torch::empty({H, Lq, Lk},
+                           q.options().dtype(torch::kFloat32));
+      dim3 rblock(256);
+      dim3 rgrid((unsigned)((inner + 255) / 256));
+      hipLaunchKernelGGL(batch_sum_bf16_kernel, rgrid, rblock, 0, stream,
+          reinterpret_cast<const __hip_bfloat16*>(ds_saved.data_ptr()),
+          dbias.data_ptr<float>(), B, inner);
+    } else {
+      dbias = ds_saved.to(torch::kFloat32);
+    }
   } else {
     dbias = torch::empty({0}, q.options().dtype(torch::kFloat32));
   }

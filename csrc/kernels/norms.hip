@@ -236,6 +236,20 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
   return {y, inv_rms};
 }
 
+// Column-sum of the per-block dw partials + cast in ONE small kernel
+// (the ATen sum(0) + .to() pair cost ~12+5 us per call, ~22 calls/step
+// on TIGER — round-2 trace). Fixed-order loop: deterministic.
+template <typename WT>
+__global__ void rms_dw_reduce_kernel(const float* __restrict__ part,
+                                     WT* __restrict__ out,
+                                     int n_part, int d) {
+  int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= d) return;
+  float acc = 0.f;
+  for (int r = 0; r < n_part; ++r) acc += part[(int64_t)r * d + j];
+  out[j] = from_f32<WT>(acc);
+}
+
 std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                         torch::Tensor w, torch::Tensor inv_rms,
                                         bool t5_style) {
@@ -280,7 +294,19 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   }
 #undef LAUNCH_RMSB_W
 #undef LAUNCH_RMSB
-  auto dw_out = dw.sum(0).to(w.scalar_type());
+  auto dw_out = torch::empty({(int64_t)d}, w.options());
+  dim3 rblock(256);
+  dim3 rgrid((d + 255) / 256);
+  if (w_bf16) {
+    hipLaunchKernelGGL((rms_dw_reduce_kernel<__hip_bfloat16>), rgrid, rblock,
+                       0, stream, dw.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(dw_out.data_ptr()),
+                       n_blocks, d);
+  } else {
+    hipLaunchKernelGGL((rms_dw_reduce_kernel<float>), rgrid, rblock, 0,
+                       stream, dw.data_ptr<float>(),
+                       dw_out.data_ptr<float>(), n_blocks, d);
+  }
   return {dx, dw_out};
 }
 

@@ -347,20 +347,48 @@ class Cobra(nn.Module):
         vec_gt = vecs[:, 1:, :].detach()
         q = b * n_pos
         valid_d = seq_mask[:, (self.C + 1)::(self.C + 1)].reshape(-1)
-        vec_pred = F.normalize(
-            vec_pred.reshape(q, -1)[valid_d], p=2, dim=-1, eps=1e-12)
-        vec_gt = F.normalize(
-            vec_gt.reshape(q, -1)[valid_d], p=2, dim=-1, eps=1e-12)
-        seq_ids = torch.arange(b, device=h.device).unsqueeze(1) \
-            .expand(-1, n_pos).reshape(-1)[valid_d]
-        same = seq_ids.unsqueeze(0) == seq_ids.unsqueeze(1)
-        same.fill_diagonal_(False)
-        sim = (vec_pred @ vec_gt.T) / self.temperature
-        sim = sim.masked_fill(same, -1e4)
-        labels = torch.arange(sim.size(0), device=sim.device)
-        loss_dense = ops.softmax_ce(sim.contiguous(), labels)
+        if getattr(self, "static_infonce", False):
+            # hipGraph-capturable variant: no boolean-mask gather (its
+            # output shape is data-dependent and would be baked into the
+            # capture). All q rows stay; invalid keys are masked to -1e4
+            # (exp(-1e4/T) ~ 0 == excluded) and invalid rows contribute
+            # zero to the masked mean. Numerically equivalent to the
+            # filtered reference formulation.
+            vp = F.normalize(vec_pred.reshape(q, -1), p=2, dim=-1,
+                             eps=1e-12)
+            vg = F.normalize(vec_gt.reshape(q, -1), p=2, dim=-1, eps=1e-12)
+            seq_ids = torch.arange(b, device=h.device).unsqueeze(1) \
+                .expand(-1, n_pos).reshape(-1)
+            same = seq_ids.unsqueeze(0) == seq_ids.unsqueeze(1)
+            same.fill_diagonal_(False)
+            sim = (vp @ vg.T) / self.temperature
+            sim = sim.masked_fill(same, -1e4)
+            sim = sim.masked_fill(~valid_d.unsqueeze(0), -1e4)
+            # keep each row's own positive so no row is all -1e4
+            diag = torch.eye(q, dtype=torch.bool, device=sim.device)
+            sim = torch.where(diag, (vp * vg).sum(-1, keepdim=True)
+                              .expand(-1, q) / self.temperature, sim)
+            labels = torch.arange(q, device=sim.device)
+            per_row = F.cross_entropy(sim, labels, reduction="none")
+            vmask = valid_d.float()
+            nv = vmask.sum().clamp(min=1.0)
+            loss_dense = (per_row * vmask).sum() / nv
+            vec_cos_sim = (F.cosine_similarity(vp, vg) * vmask).sum() / nv
+        else:
+            vec_pred = F.normalize(
+                vec_pred.reshape(q, -1)[valid_d], p=2, dim=-1, eps=1e-12)
+            vec_gt = F.normalize(
+                vec_gt.reshape(q, -1)[valid_d], p=2, dim=-1, eps=1e-12)
+            seq_ids = torch.arange(b, device=h.device).unsqueeze(1) \
+                .expand(-1, n_pos).reshape(-1)[valid_d]
+            same = seq_ids.unsqueeze(0) == seq_ids.unsqueeze(1)
+            same.fill_diagonal_(False)
+            sim = (vec_pred @ vec_gt.T) / self.temperature
+            sim = sim.masked_fill(same, -1e4)
+            labels = torch.arange(sim.size(0), device=sim.device)
+            loss_dense = ops.softmax_ce(sim.contiguous(), labels)
 
-        vec_cos_sim = F.cosine_similarity(vec_pred, vec_gt).mean()
+            vec_cos_sim = F.cosine_similarity(vec_pred, vec_gt).mean()
         with torch.no_grad():
             usage = torch.stack([
                 F.one_hot(input_ids[:, c::self.C], self.pad_id + 1)

@@ -201,3 +201,25 @@ def test_cobra_decoder_matches_torch():
     valid = ~pad
     assert torch.allclose(out_nat[valid], out_ref[valid], atol=1e-5), \
         (out_nat[valid] - out_ref[valid]).abs().max()
+
+
+def test_cobra_static_infonce_equivalent(cobra):
+    """Graph-capturable fixed-shape InfoNCE (mask-to--1e4) == the
+    filtered reference formulation, including padded rows."""
+    torch.manual_seed(3)
+    m = cobra
+    ids, enc = _batch(m, B=3, T=5)
+    ids[1, 9:] = m.pad_id  # pad the tail of one sequence
+    m.eval()
+    out_dyn = m(ids, enc)
+    m.static_infonce = True
+    try:
+        out_sta = m(ids, enc)
+    finally:
+        m.static_infonce = False
+    assert torch.allclose(out_sta.loss_dense, out_dyn.loss_dense,
+                          atol=1e-5), (out_sta.loss_dense,
+                                       out_dyn.loss_dense)
+    assert torch.allclose(out_sta.vec_cos_sim, out_dyn.vec_cos_sim,
+                          atol=1e-5)
+    assert torch.allclose(out_sta.loss, out_dyn.loss, atol=1e-5)
