@@ -29,6 +29,7 @@ class SyntheticCobraDataset(Dataset):
         self.C = n_codebooks
         self.id_vocab_size = id_vocab_size
         self.num_items = num_items
+        self.max_items_per_seq = max_items_per_seq
         rng = np.random.default_rng(seed + 71)
         self.item_sem_ids = rng.integers(
             0, id_vocab_size, size=(num_items + 1, n_codebooks))
@@ -76,9 +77,12 @@ class SyntheticCobraDataset(Dataset):
 
 
 def cobra_collate_fn(batch: List[Dict], pad_id: int, n_codebooks: int,
-                     train: bool = True) -> Dict[str, torch.Tensor]:
+                     train: bool = True,
+                     fixed_items: int = 0) -> Dict[str, torch.Tensor]:
     """Train: target appended to history (model's shifted loss covers it).
-    Eval: history only, target kept aside (ref cobra_trainer.py:25-88)."""
+    Eval: history only, target kept aside (ref cobra_trainer.py:25-88).
+    fixed_items > 0 pads every batch to that item count (static shapes
+    for hipGraph capture)."""
     B = len(batch)
     if train:
         items = [b["history_sem_ids"] + [b["target_sem_ids"]] for b in batch]
@@ -87,6 +91,8 @@ def cobra_collate_fn(batch: List[Dict], pad_id: int, n_codebooks: int,
         items = [b["history_sem_ids"] for b in batch]
         texts = [list(b["history_text"]) for b in batch]
     max_t = max(len(x) for x in items)
+    if fixed_items:
+        max_t = max(max_t, fixed_items)
     text_len = len(batch[0]["target_text"])
     input_ids = torch.full((B, max_t * n_codebooks), pad_id, dtype=torch.long)
     enc = torch.zeros(B, max_t, text_len, dtype=torch.long)

@@ -90,6 +90,7 @@ def train(
     resume_path: Optional[str] = None,
     num_workers: int = 0,
     eval_max_batches: Optional[int] = None,
+    use_hip_graph: bool = False,
 ):
     common.enable_tuned_gemms()
     ctx = init_distributed()
@@ -113,7 +114,13 @@ def train(
     opt = AdamW(model.parameters(), lr=learning_rate,
                 weight_decay=weight_decay)
     pad_id = model.pad_id
-    tcoll = lambda b: cobra_collate_fn(b, pad_id, n_codebooks, train=True)
+    graph_mode = (use_hip_graph and device.type == "cuda"
+                  and gradient_accumulate_every == 1)
+    fixed_items = (getattr(train_ds, "max_items_per_seq", 20) + 1
+                   if graph_mode else 0)
+    model.static_infonce = graph_mode  # fixed-shape InfoNCE for capture
+    tcoll = lambda b: cobra_collate_fn(b, pad_id, n_codebooks, train=True,
+                                       fixed_items=fixed_items)
     ecoll = lambda b: cobra_collate_fn(b, pad_id, n_codebooks, train=False)
     train_loader = common.make_loader(train_ds, batch_size, ctx, True, tcoll,
                                       num_workers=num_workers, seed=seed,
@@ -122,17 +129,48 @@ def train(
                                       ecoll, num_workers=num_workers)
     sched = get_cosine_schedule_with_warmup(
         opt, num_warmup_steps, max(1, len(train_loader)) * epochs)
-    reducer = GradReducer(model)
+    reducer = None if graph_mode else GradReducer(model)
 
     start_epoch, step = 0, 0
+    resume_state = None
     if resume_path and os.path.exists(resume_path):
-        state = common.load_checkpoint(resume_path, model, opt, sched,
-                                       map_location=device)
-        start_epoch = state.get("epoch", -1) + 1
+        resume_state = common.load_checkpoint(resume_path, model, opt, sched,
+                                              map_location=device)
+        start_epoch = resume_state.get("epoch", -1) + 1
 
     wb = common.init_wandb(wandb_project, {"model": "cobra"},
                            wandb_logging, ctx.is_main)
     amp_ctx = common.autocast_ctx(device, mixed_precision_type if amp else None)
+
+    runner = None
+    if graph_mode:
+        # hipGraph-captured full step (fwd+bwd+allreduce+clip+AdamW) at
+        # fixed shapes; eager fallback on capture failure. The fixed-shape
+        # InfoNCE variant (models/cobra.py) keeps capture shape-safe.
+        from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+        example = common.to_device(next(iter(train_loader)), device)
+        runner = GraphedTrainStep(
+            model, {"input_ids": example["input_ids"],
+                    "encoder_input_ids": example["encoder_input_ids"]},
+            loss_getter=lambda out: (sparse_loss_weight * out.loss_sparse
+                                     + dense_loss_weight * out.loss_dense),
+            lr=learning_rate, weight_decay=weight_decay, clip_norm=1.0,
+            world=ctx.world_size)
+        if resume_state is not None and "runner" in resume_state:
+            runner.load_state_dict(resume_state["runner"])
+
+        import math as _math
+
+        total_sched = max(1, len(train_loader)) * epochs
+
+        def _cosine_lr(st: int) -> float:
+            if st < num_warmup_steps:
+                return learning_rate * st / max(1, num_warmup_steps)
+            prog = (st - num_warmup_steps) / max(
+                1, total_sched - num_warmup_steps)
+            return learning_rate * max(
+                0.0, 0.5 * (1.0 + _math.cos(_math.pi * prog)))
 
     for epoch in range(start_epoch, epochs):
         model.train()
@@ -141,6 +179,20 @@ def train(
         # device-resident epoch counters: no per-batch .item() syncs
         ep_t = torch.zeros(4, device=device)
         for it, batch in enumerate(train_loader):
+            if runner is not None:
+                runner.set_lr(_cosine_lr(step))
+                loss_t = runner.step(
+                    {"input_ids": batch["input_ids"].to(device),
+                     "encoder_input_ids": batch["encoder_input_ids"]
+                     .to(device)})
+                step += 1
+                if ctx.is_main and step % wandb_log_interval == 0:
+                    logger.info("epoch %d step %d loss %.4f", epoch, step,
+                                loss_t.item())
+                    wb.log({"train/loss": loss_t.item()})
+                if max_steps is not None and step >= max_steps:
+                    break
+                continue
             micro = (it + 1) % gradient_accumulate_every == 0
             reducer.skip_sync = not micro
             with amp_ctx:
@@ -193,13 +245,15 @@ def train(
         if ctx.is_main and (epoch + 1) % save_every_epoch == 0:
             common.save_checkpoint(
                 os.path.join(save_dir_root, f"checkpoint_epoch_{epoch}.pt"),
-                model, opt, sched, epoch=epoch, is_main=True)
+                model, opt, sched, epoch=epoch, is_main=True,
+                runner=runner)
         if max_steps is not None and step >= max_steps:
             break
     if ctx.is_main:
         common.save_checkpoint(
             os.path.join(save_dir_root, "checkpoint_final.pt"),
-            model, opt, sched, epoch=epochs - 1, is_main=True)
+            model, opt, sched, epoch=epochs - 1, is_main=True,
+            runner=runner)
     wb.finish()
     return model
 
