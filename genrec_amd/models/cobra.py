@@ -126,7 +126,7 @@ class CobraEmbedding(nn.Module):
 
         pos = torch.arange(out_len, device=device).unsqueeze(0).expand(b, -1)
         type_idx = is_dense.long().unsqueeze(0).expand(b, -1)
-        m = mask.unsqueeze(-1).float()
+        m = mask.unsqueeze(-1).to(h.dtype)  # dtype-preserving (bf16 path)
         h = h * m
         h = h + self.pos_embed(pos) * m
         h = h + self.type_embed(type_idx) * m

@@ -49,7 +49,9 @@ class LightT5Encoder(nn.Module):
         x = self.embedding(flat) + self.pos_embedding(pos)
         pad = flat == 0
         hidden = self.layer_norm(self.encoder(x, src_key_padding_mask=pad))
-        m = (~pad).unsqueeze(-1).float()
+        m = (~pad).unsqueeze(-1).to(hidden.dtype)  # keep bf16 under the
+        # pure-bf16 graph runner (a .float() mask silently promoted the
+        # whole pooled path to fp32)
         pooled = (hidden * m).sum(dim=1) / m.sum(dim=1).clamp(min=1e-9)
         out = F.normalize(self.proj(pooled), p=2, dim=-1)
         return out.view(b, t, -1) if t > 1 else out

@@ -121,3 +121,34 @@ def test_fused_checkpoint_loads_into_cpu_runner():
     flat = torch.cat([m.reshape(-1) for m in r.masters])
     assert torch.allclose(flat, fake["flat_master"])
     assert r.opt.param_groups[0]["lr"] == 5e-4
+
+
+def test_graphed_step_cobra_cpu():
+    """GraphedTrainStep drives the COBRA model (static-InfoNCE mode) —
+    the bf16-conversion path must keep every intermediate bf16 on CPU
+    too (fp32 mask promotions broke the pure-bf16 runner in round 2)."""
+    import torch
+
+    from genrec_amd.data.cobra_synthetic import (SyntheticCobraDataset,
+                                                 cobra_collate_fn)
+    from genrec_amd.models.cobra import Cobra
+    from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+    torch.manual_seed(0)
+    ds = SyntheticCobraDataset(num_users=30, num_items=50, split="train",
+                               n_codebooks=3, id_vocab_size=16)
+    m = Cobra(encoder_n_layers=1, encoder_hidden_dim=32,
+              encoder_num_heads=4, encoder_vocab_size=1000,
+              id_vocab_size=16, n_codebooks=3, d_model=32,
+              decoder_n_layers=2, decoder_num_heads=4, decoder_dropout=0.0)
+    m.static_infonce = True
+    batch = cobra_collate_fn([ds[i] for i in range(4)], m.pad_id, 3,
+                             train=True,
+                             fixed_items=ds.max_items_per_seq + 1)
+    ex = {"input_ids": batch["input_ids"],
+          "encoder_input_ids": batch["encoder_input_ids"]}
+    runner = GraphedTrainStep(
+        m, ex, loss_getter=lambda out: out.loss_sparse + out.loss_dense,
+        lr=1e-3, weight_decay=0.0, clip_norm=1.0, world=1, use_graph=False)
+    losses = [float(runner.step(ex).detach()) for _ in range(3)]
+    assert losses[-1] < losses[0]  # optimizes
