@@ -136,7 +136,8 @@ def test_graphed_step_cobra_cpu():
 
     torch.manual_seed(0)
     ds = SyntheticCobraDataset(num_users=30, num_items=50, split="train",
-                               n_codebooks=3, id_vocab_size=16)
+                               n_codebooks=3, id_vocab_size=16,
+                               text_vocab_size=1000)
     m = Cobra(encoder_n_layers=1, encoder_hidden_dim=32,
               encoder_num_heads=4, encoder_vocab_size=1000,
               id_vocab_size=16, n_codebooks=3, d_model=32,
