@@ -579,14 +579,30 @@ attn_bwd_ds_kernel(
 
 // Batch-reduce of the bf16 bias-grad buffer: out[h,i,j] = sum_b ds[b,h,i,j]
 // in fp32 — replaces an ATen sum(0) that cost ~11.7 us/layer (BACKLOG r1
-// item 2b). Fixed-order loop over B: deterministic.
-__global__ void batch_sum_bf16_kernel(const __hip_bfloat16* __restrict__ in,
-                                      float* __restrict__ out,
-                                      int B, int64_t inner) {
+// item 2b). Two stages keep enough resident waves; fixed-order loops:
+// deterministic.
+constexpr int BS_CH = 16;
+
+__global__ void batch_sum1_kernel(const __hip_bfloat16* __restrict__ in,
+                                  float* __restrict__ tmp,
+                                  int B, int64_t inner) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= (int64_t)BS_CH * inner) return;
+  int64_t j = tid % inner;
+  int chunk = (int)(tid / inner);
+  float acc = 0.f;
+  for (int b = chunk; b < B; b += BS_CH)
+    acc += to_f32(in[(int64_t)b * inner + j]);
+  tmp[(int64_t)chunk * inner + j] = acc;
+}
+
+__global__ void batch_sum2_kernel(const float* __restrict__ tmp,
+                                  float* __restrict__ out, int64_t inner) {
   int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (j >= inner) return;
   float acc = 0.f;
-  for (int b = 0; b < B; ++b) acc += to_f32(in[(int64_t)b * inner + j]);
+#pragma unroll
+  for (int r = 0; r < BS_CH; ++r) acc += tmp[(int64_t)r * inner + j];
   out[j] = acc;
 }
 
@@ -730,11 +746,16 @@ std::vector<torch::Tensor> attn_bwd_mfma(
       const int64_t inner = (int64_t)H * Lq * Lk;
       dbias = torch::empty({H, Lq, Lk},
                            q.options().dtype(torch::kFloat32));
+      auto tmp = torch::empty({16, inner},
+                              q.options().dtype(torch::kFloat32));
       dim3 rblock(256);
-      dim3 rgrid((unsigned)((inner + 255) / 256));
-      hipLaunchKernelGGL(batch_sum_bf16_kernel, rgrid, rblock, 0, stream,
+      dim3 rgrid1((unsigned)((16 * inner + 255) / 256));
+      hipLaunchKernelGGL(batch_sum1_kernel, rgrid1, rblock, 0, stream,
           reinterpret_cast<const __hip_bfloat16*>(ds_saved.data_ptr()),
-          dbias.data_ptr<float>(), B, inner);
+          tmp.data_ptr<float>(), B, inner);
+      dim3 rgrid2((unsigned)((inner + 255) / 256));
+      hipLaunchKernelGGL(batch_sum2_kernel, rgrid2, rblock, 0, stream,
+          tmp.data_ptr<float>(), dbias.data_ptr<float>(), inner);
     } else {
       dbias = ds_saved.to(torch::kFloat32);
     }

@@ -236,17 +236,34 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
   return {y, inv_rms};
 }
 
-// Column-sum of the per-block dw partials + cast in ONE small kernel
-// (the ATen sum(0) + .to() pair cost ~12+5 us per call, ~22 calls/step
-// on TIGER — round-2 trace). Fixed-order loop: deterministic.
+// Column-sum of the per-block dw partials + cast, replacing the ATen
+// sum(0) + .to() pair (~12+5 us per call, ~22 calls/step on TIGER).
+// Two stages so the reduce has enough resident waves (a single-stage
+// d-thread loop left only 6 waves on the whole chip and was LOSING to
+// ATen): stage 1 folds n_part rows 32-fold with CH*d threads, stage 2
+// finishes 32 rows with d threads. Fixed-order loops: deterministic.
+constexpr int RED_CH = 32;
+
+__global__ void rms_dw_reduce1_kernel(const float* __restrict__ part,
+                                      float* __restrict__ tmp,
+                                      int n_part, int d) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= (int64_t)RED_CH * d) return;
+  int j = (int)(tid % d);
+  int chunk = (int)(tid / d);
+  float acc = 0.f;
+  for (int r = chunk; r < n_part; r += RED_CH) acc += part[(int64_t)r * d + j];
+  tmp[(int64_t)chunk * d + j] = acc;
+}
+
 template <typename WT>
-__global__ void rms_dw_reduce_kernel(const float* __restrict__ part,
-                                     WT* __restrict__ out,
-                                     int n_part, int d) {
+__global__ void rms_dw_reduce2_kernel(const float* __restrict__ tmp,
+                                      WT* __restrict__ out, int d) {
   int j = blockIdx.x * blockDim.x + threadIdx.x;
   if (j >= d) return;
   float acc = 0.f;
-  for (int r = 0; r < n_part; ++r) acc += part[(int64_t)r * d + j];
+#pragma unroll
+  for (int r = 0; r < RED_CH; ++r) acc += tmp[(int64_t)r * d + j];
   out[j] = from_f32<WT>(acc);
 }
 
@@ -295,17 +312,23 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
 #undef LAUNCH_RMSB_W
 #undef LAUNCH_RMSB
   auto dw_out = torch::empty({(int64_t)d}, w.options());
+  auto dw_tmp = torch::empty({RED_CH, (int64_t)d},
+                             x.options().dtype(torch::kFloat32));
   dim3 rblock(256);
-  dim3 rgrid((d + 255) / 256);
+  dim3 rgrid1((unsigned)(((int64_t)RED_CH * d + 255) / 256));
+  hipLaunchKernelGGL(rms_dw_reduce1_kernel, rgrid1, rblock, 0, stream,
+                     dw.data_ptr<float>(), dw_tmp.data_ptr<float>(),
+                     n_blocks, d);
+  dim3 rgrid2((d + 255) / 256);
   if (w_bf16) {
-    hipLaunchKernelGGL((rms_dw_reduce_kernel<__hip_bfloat16>), rgrid, rblock,
-                       0, stream, dw.data_ptr<float>(),
+    hipLaunchKernelGGL((rms_dw_reduce2_kernel<__hip_bfloat16>), rgrid2,
+                       rblock, 0, stream, dw_tmp.data_ptr<float>(),
                        reinterpret_cast<__hip_bfloat16*>(dw_out.data_ptr()),
-                       n_blocks, d);
+                       d);
   } else {
-    hipLaunchKernelGGL((rms_dw_reduce_kernel<float>), rgrid, rblock, 0,
-                       stream, dw.data_ptr<float>(),
-                       dw_out.data_ptr<float>(), n_blocks, d);
+    hipLaunchKernelGGL((rms_dw_reduce2_kernel<float>), rgrid2, rblock, 0,
+                       stream, dw_tmp.data_ptr<float>(),
+                       dw_out.data_ptr<float>(), d);
   }
   return {dx, dw_out};
 }
