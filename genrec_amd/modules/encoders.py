@@ -31,7 +31,10 @@ class LightT5Encoder(nn.Module):
         layer = nn.TransformerEncoderLayer(
             d_model=hidden_dim, nhead=num_heads, dim_feedforward=ff_dim,
             dropout=dropout, batch_first=True)
-        self.encoder = nn.TransformerEncoder(layer, num_layers=n_layers)
+        # nested-tensor fast path does host-side mask inspection
+        # (.item() syncs) — illegal inside hipGraph capture
+        self.encoder = nn.TransformerEncoder(layer, num_layers=n_layers,
+                                             enable_nested_tensor=False)
         self.proj = nn.Linear(hidden_dim, output_dim)
         self.layer_norm = nn.LayerNorm(hidden_dim)
 
