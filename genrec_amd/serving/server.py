@@ -30,7 +30,8 @@ from genrec_amd.models.tiger import Tiger
 class RecommendationService:
     def __init__(self, model: Tiger, item_sem_ids: torch.Tensor,
                  device: Optional[torch.device] = None,
-                 max_items_per_seq: int = 20, top_k: int = 10) -> None:
+                 max_items_per_seq: int = 20, top_k: int = 10,
+                 use_graph: bool = True) -> None:
         self.device = device or (
             torch.device("cuda:0") if torch.cuda.is_available()
             else torch.device("cpu"))
@@ -51,6 +52,14 @@ class RecommendationService:
                 torch.ones(1, self.sem_dim, dtype=torch.long,
                            device=self.device),
                 n_top_k_candidates=1, valid_item_ids=self.item_sem_ids)
+        # hipGraph-captured decode (one graph per batch-shape) removes the
+        # per-dispatch host cost of the launch-bound beam
+        self._graphed = None
+        if use_graph and self.device.type == "cuda":
+            from genrec_amd.serving.graphed_generate import GraphedGenerate
+
+            self._graphed = GraphedGenerate(
+                self.model, self.item_sem_ids, n_top_k_candidates=top_k)
 
     @torch.no_grad()
     def recommend_batch(self, user_ids: List[int],
@@ -76,10 +85,17 @@ class RecommendationService:
             item_ids[i, :flat.numel()] = flat
             mask[i, :flat.numel()] = 1
         users = torch.tensor(user_ids, dtype=torch.long).unsqueeze(1)
-        gen = self.model.generate(
-            users.to(self.device), item_ids.to(self.device),
-            ttype.to(self.device), mask.to(self.device),
-            n_top_k_candidates=k, valid_item_ids=self.item_sem_ids)
+        if self._graphed is not None and k == self.top_k:
+            gen = self._graphed(
+                user_input_ids=users.to(self.device),
+                item_input_ids=item_ids.to(self.device),
+                token_type_ids=ttype.to(self.device),
+                seq_mask=mask.to(self.device))
+        else:
+            gen = self.model.generate(
+                users.to(self.device), item_ids.to(self.device),
+                ttype.to(self.device), mask.to(self.device),
+                n_top_k_candidates=k, valid_item_ids=self.item_sem_ids)
         out: List[List[Dict]] = []
         for i in range(b):
             row = []
