@@ -36,31 +36,45 @@ def _timeit(step, steps, warmup, device):
     return time.perf_counter() - t0
 
 
+def _graph_step(model, batch, loss_getter, device, betas=(0.9, 0.999)):
+    """hipGraph-captured full step on CUDA (the trainers' production
+    path); eager AdamW fallback elsewhere."""
+    if device.type == "cuda":
+        from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+        runner = GraphedTrainStep(model, batch, loss_getter, lr=1e-3,
+                                  betas=betas, weight_decay=0.0,
+                                  clip_norm=None, world=1, use_graph=True)
+        return lambda i: runner.step(batch)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3, betas=betas)
+
+    def step(i):
+        opt.zero_grad(set_to_none=False)
+        loss_getter(model(**batch)).backward()
+        opt.step()
+
+    return step
+
+
 def bench_sasrec(device, steps, warmup):
-    """SASRec: B=128, L=50, D=64, H=2, 2 blocks, V=12101 (sasrec/amazon.gin)."""
+    """SASRec: B=128, L=50, D=64, H=2, 2 blocks, V=12101 (sasrec/amazon.gin),
+    graph-captured step on GPU."""
     from genrec_amd.models.sasrec import SASRec
 
     torch.manual_seed(0)
     B, L, V = 128, 50, 12101
     model = SASRec(num_items=V - 1, max_seq_len=L, embed_dim=64, num_heads=2,
                    num_blocks=2, ffn_dim=256, dropout=0.2).to(device)
-    if device.type == "cuda":
-        model = model.to(torch.bfloat16)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3, betas=(0.9, 0.98))
     ids = torch.randint(1, V, (B, L), device=device)
     ids[::4, :20] = 0
     model.train()
-
-    def step(i):
-        opt.zero_grad(set_to_none=False)
-        _, loss = model(ids, ids)
-        loss.backward()
-        opt.step()
+    step = _graph_step(model, {"input_ids": ids, "targets": ids},
+                       lambda out: out[1], device, betas=(0.9, 0.98))
 
     el = _timeit(step, steps, warmup, device)
     return dict(model="sasrec-amazon-beauty", batch=B,
                 samples_per_s=B * steps / el,
-                ms_per_step=el / steps * 1e3)
+                ms_per_step=el / steps * 1e3, hip_graph=device.type == "cuda")
 
 
 def bench_hstu(device, steps, warmup):
@@ -71,24 +85,18 @@ def bench_hstu(device, steps, warmup):
     B, L, V = 128, 50, 12101
     model = HSTU(num_items=V - 1, max_seq_len=L, embed_dim=64, num_heads=2,
                  num_blocks=2, dropout=0.2, use_temporal_bias=True).to(device)
-    if device.type == "cuda":
-        model = model.to(torch.bfloat16)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3, betas=(0.9, 0.98))
     ids = torch.randint(1, V, (B, L), device=device)
     ts = (torch.arange(L, device=device) * 86400 + 10 ** 9).unsqueeze(0) \
         .expand(B, -1).contiguous()
     model.train()
-
-    def step(i):
-        opt.zero_grad(set_to_none=False)
-        _, loss = model(ids, ts, ids)
-        loss.backward()
-        opt.step()
+    step = _graph_step(
+        model, {"input_ids": ids, "timestamps": ts, "targets": ids},
+        lambda out: out[1], device, betas=(0.9, 0.98))
 
     el = _timeit(step, steps, warmup, device)
     return dict(model="hstu-amazon-beauty-bf16", batch=B,
                 samples_per_s=B * steps / el,
-                ms_per_step=el / steps * 1e3)
+                ms_per_step=el / steps * 1e3, hip_graph=device.type == "cuda")
 
 
 def bench_rqvae(device, steps, warmup):
@@ -170,26 +178,18 @@ def bench_cobra(device, steps, warmup):
                   id_vocab_size=256, n_codebooks=3, d_model=384,
                   decoder_n_layers=4, decoder_num_heads=6,
                   decoder_dropout=0.1).to(device)
-    opt = torch.optim.AdamW(model.parameters(), lr=1e-4, weight_decay=0.01)
     ids = torch.randint(0, 256, (B, T * 3), device=device)
     enc = torch.randint(1, 32128, (B, T, Ltxt), device=device)
     model.train()
-    import contextlib
-
-    amp = (torch.autocast("cuda", dtype=torch.bfloat16)
-           if device.type == "cuda" else contextlib.nullcontext())
-
-    def step(i):
-        opt.zero_grad(set_to_none=False)
-        with amp:
-            out = model(ids, enc)
-        out.loss.backward()
-        opt.step()
+    model.static_infonce = device.type == "cuda"  # capture-safe variant
+    step = _graph_step(
+        model, {"input_ids": ids, "encoder_input_ids": enc},
+        lambda out: out.loss_sparse + out.loss_dense, device)
 
     el = _timeit(step, steps, warmup, device)
     return dict(model="cobra-amazon-beauty", batch=B,
                 samples_per_s=B * steps / el,
-                ms_per_step=el / steps * 1e3)
+                ms_per_step=el / steps * 1e3, hip_graph=device.type == "cuda")
 
 
 BENCHES = {"sasrec": bench_sasrec, "hstu": bench_hstu, "rqvae": bench_rqvae,
