@@ -107,10 +107,11 @@ class CobraEmbedding(nn.Module):
         n_ins = min(n_complete_items, t_vecs)
         out_len = l + n_ins
         orig = torch.arange(l, device=device)
+        # clamp/full_like, NOT torch.tensor(scalar, device=...): an H2D
+        # scalar upload is illegal inside hipGraph capture
         shift = torch.where(orig < n_ct,
-                            torch.minimum(orig // self.C,
-                                          torch.tensor(n_ins, device=device)),
-                            torch.tensor(n_ins, device=device))
+                            (orig // self.C).clamp(max=n_ins),
+                            torch.full_like(orig, n_ins))
         new_pos = orig + shift
         src_idx = torch.empty(out_len, dtype=torch.long, device=device)
         src_idx.scatter_(0, new_pos, orig)
