@@ -146,17 +146,16 @@ def bench_lcrec(device, steps, warmup, full_size=False):
     model = model.to(device)
     if device.type == "cuda":
         model = model.to(torch.bfloat16)
-    opt = torch.optim.AdamW(model.parameters(), lr=3e-5)
     V = model.model.config.vocab_size
     ids = torch.randint(0, V, (B, L), device=device)
     attn = torch.ones_like(ids)
     model.train()
-
-    def step(i):
-        opt.zero_grad(set_to_none=False)
-        out = model(ids, attn, labels=ids)
-        out.loss.backward()
-        opt.step()
+    # GraphedTrainStep: fused flat AdamW + steal-grads even when hipGraph
+    # capture falls back to eager (HF checkpointing is not capture-safe)
+    step = _graph_step(model,
+                       {"input_ids": ids, "attention_mask": attn,
+                        "labels": ids},
+                       lambda out: out.loss, device)
 
     el = _timeit(step, steps, warmup, device)
     return dict(model="lcrec-qwen2-1.5b" if full_size else "lcrec-tiny",
