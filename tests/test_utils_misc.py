@@ -83,3 +83,26 @@ def test_enable_tuned_gemms_env(monkeypatch):
     monkeypatch.delenv("PYTORCH_TUNABLEOP_ENABLED", raising=False)
     common.enable_tuned_gemms()
     assert "PYTORCH_TUNABLEOP_ENABLED" not in os.environ
+
+
+def test_unpadded_shard_sampler_partition():
+    """Eval sharding: every sample exactly once across ranks, no
+    duplicate padding (advisor r1: DistributedSampler padding biased
+    all-reduced metrics)."""
+    from genrec_amd.trainers.common import UnpaddedShardSampler
+
+    class DS:
+        def __len__(self):
+            return 10
+
+    ds = DS()
+    seen = []
+    sizes = []
+    for r in range(3):
+        s = UnpaddedShardSampler(ds, num_replicas=3, rank=r)
+        idx = list(iter(s))
+        assert len(s) == len(idx)
+        sizes.append(len(idx))
+        seen.extend(idx)
+    assert sorted(seen) == list(range(10))
+    assert max(sizes) - min(sizes) <= 1
