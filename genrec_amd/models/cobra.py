@@ -90,12 +90,14 @@ class CobraEmbedding(nn.Module):
         if n_complete_items is None:
             n_complete_items = l // self.C
 
-        # codebook-offset gather
+        # codebook-offset gather (graph-safe genrec embedding op: ATen's
+        # rocprim embedding backward faults under hipGraph replay)
         ttype = (torch.arange(l, device=device) % self.C).unsqueeze(0)
         valid = input_ids != self.pad_id
         flat_ids = torch.where(
             valid, input_ids + ttype * self.id_vocab_size, input_ids)
-        sparse_emb = self.id_embed(flat_ids)  # (B, L, D)
+        sparse_emb = ops.embedding(self.id_embed.weight, flat_ids,
+                                   self.id_embed.padding_idx)
 
         # K18 interleave as ONE indexed gather from [sparse_emb | vecs]
         # (the reference's chunk/cat loop is ~2T kernel launches,
@@ -129,8 +131,8 @@ class CobraEmbedding(nn.Module):
         type_idx = is_dense.long().unsqueeze(0).expand(b, -1)
         m = mask.unsqueeze(-1).to(h.dtype)  # dtype-preserving (bf16 path)
         h = h * m
-        h = h + self.pos_embed(pos) * m
-        h = h + self.type_embed(type_idx) * m
+        h = h + ops.embedding(self.pos_embed.weight, pos) * m
+        h = h + ops.embedding(self.type_embed.weight, type_idx) * m
         return h
 
 

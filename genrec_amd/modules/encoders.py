@@ -49,7 +49,11 @@ class LightT5Encoder(nn.Module):
             t = 1
             flat = batch_tokens
         pos = torch.arange(l, device=flat.device).unsqueeze(0)
-        x = self.embedding(flat) + self.pos_embedding(pos)
+        # graph-safe embedding gathers (ATen rocprim bwd faults in replay)
+        from genrec_amd import ops
+
+        x = ops.embedding(self.embedding.weight, flat) \
+            + ops.embedding(self.pos_embedding.weight, pos)
         pad = flat == 0
         hidden = self.layer_norm(self.encoder(x, src_key_padding_mask=pad))
         m = (~pad).unsqueeze(-1).to(hidden.dtype)  # keep bf16 under the
