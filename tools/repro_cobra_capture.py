@@ -52,15 +52,21 @@ def main():
     ids = torch.randint(0, 256, (B, T * 3), device=dev)
     enc = torch.randint(1, 32128, (B, T, Ltxt), device=dev)
 
+    with torch.no_grad():
+        vecs = m.encoder(enc)
+        seq_mask = m.interleave_seq_mask(ids != m.pad_id, m.C)
+        emb = m.cobra_emb(ids, vecs, seq_mask)
     try_capture("encoder", lambda: m.encoder(enc))
-    vecs = m.encoder(enc)
-    seq_mask = m.interleave_seq_mask(ids != m.pad_id, m.C)
     try_capture("interleave+embed",
                 lambda: m.cobra_emb(ids, vecs, seq_mask))
-    emb = m.cobra_emb(ids, vecs, seq_mask)
     try_capture("decoder",
                 lambda: m.decoder(emb, tgt_key_padding_mask=~seq_mask))
     try_capture("full fwd", lambda: m(ids, enc).loss)
+    # fwd+bwd must not hold stale autograd state from the phases above
+    # (AccumulateGrad nodes pinned to the default stream break capture)
+    del vecs, emb
+    for p in m.parameters():
+        p.grad = None
 
     def fwd_bwd():
         out = m(ids, enc)
