@@ -30,6 +30,12 @@ logger = logging.getLogger("genrec_amd")
 class GraphedGenerate:
     """Capture-once / replay-per-request wrapper around Tiger.generate."""
 
+    # replay of a captured decode faults in an ATen gather at B=512 on
+    # ROCm 7 (HSA exception on the 2nd replay; eager B=512 and all
+    # graphed B<=256 runs are clean — tools/repro_gen512.py). Cap the
+    # captured path until root-caused; larger batches run eager.
+    MAX_GRAPH_BATCH = 256
+
     def __init__(self, model: Tiger, valid_item_ids: torch.Tensor,
                  n_top_k_candidates: int = 10, temperature: float = 0.2,
                  warmup_iters: int = 2) -> None:
@@ -73,6 +79,10 @@ class GraphedGenerate:
     def __call__(self, **inputs) -> TigerGenerationOutput:
         key = (inputs["item_input_ids"].size(0),
                inputs["item_input_ids"].size(1))
+        if key[0] > self.MAX_GRAPH_BATCH:
+            return self.model.generate(
+                **inputs, n_top_k_candidates=self.k,
+                valid_item_ids=self.valid, temperature=self.temperature)
         state = self._graphs.get(key)
         if state is None:
             state = self._build(inputs)
