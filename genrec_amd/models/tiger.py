@@ -299,7 +299,13 @@ class Tiger(nn.Module):
             # Gumbel-top-K == multinomial without replacement (tiger.py:386)
             gumbel = -torch.log(-torch.log(
                 torch.rand_like(log_probs) + 1e-20) + 1e-20)
-            cand_vocab = torch.topk(log_probs + gumbel, kk, dim=-1).indices
+            # sort-based top-KK instead of torch.topk: ATen's multi-block
+            # topk (chosen at b*k >= ~512 rows) keeps a workspace that is
+            # not hipGraph-replay-safe (2nd replay faults, ROCm 7 —
+            # tools/repro_gen512.py); radix sort replays cleanly and ties
+            # are measure-zero under Gumbel noise
+            cand_vocab = (log_probs + gumbel).sort(
+                dim=-1, descending=True).indices[:, :kk].contiguous()
             cand_logp = torch.gather(log_probs, 1, cand_vocab)
             cand_tok = cand_vocab - offset
 
@@ -328,7 +334,9 @@ class Tiger(nn.Module):
                 torch.full_like(score_desc, NEG_INF_SCORE))
             # index back into the (k*kk) flat candidate list
             flat_idx = score_order.gather(1, ko2)
-            top_scores, top_pos = dedup_score.topk(k, dim=1)
+            srt = dedup_score.sort(dim=1, descending=True)
+            top_scores = srt.values[:, :k]
+            top_pos = srt.indices[:, :k]
             chosen = flat_idx.gather(1, top_pos)  # [b, k]
 
             new_tok = total_tok.gather(1, chosen)

@@ -30,11 +30,10 @@ logger = logging.getLogger("genrec_amd")
 class GraphedGenerate:
     """Capture-once / replay-per-request wrapper around Tiger.generate."""
 
-    # replay of a captured decode faults in an ATen gather at B=512 on
-    # ROCm 7 (HSA exception on the 2nd replay; eager B=512 and all
-    # graphed B<=256 runs are clean — tools/repro_gen512.py). Cap the
-    # captured path until root-caused; larger batches run eager.
-    MAX_GRAPH_BATCH = 256
+    # ATen's multi-block topk was not hipGraph-replay-safe (2nd replay
+    # faulted at B=512, ROCm 7); generate now uses sort-based selection
+    # (models/tiger.py). Cap kept as a generous guard for huge batches.
+    MAX_GRAPH_BATCH = 1024
 
     def __init__(self, model: Tiger, valid_item_ids: torch.Tensor,
                  n_top_k_candidates: int = 10, temperature: float = 0.2,
