@@ -93,62 +93,76 @@ skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,  // [M,K]
 #pragma unroll
     for (int mf = 0; mf < 4; ++mf) acc[nf][mf] = float4v{0.f, 0.f, 0.f, 0.f};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage A tile: 64 rows x 64 k (8 chunks of 16 B per row)
-    for (int idx = tid; idx < BM * (BK / 8); idx += blockDim.x) {
-      int row = idx / (BK / 8);
-      int c8 = (idx % (BK / 8)) * 8;
-      short8v val = {};
-      int mi = m0 + row;
+  // T14 register-pipelined staging (guide §5): ONE register set holds
+  // tile t+1's global loads while the MFMAs consume tile t from LDS;
+  // the write pass runs after the read barrier. Synchronous staging
+  // measured 2.6x slower on the guide's GEMM — and was why skinny v1
+  // lost to hipBLASLt.
+  // Per-thread chunks: A 512/256 = 2, W 3072/256 = 12 (16 B each).
+  const int a_row = tid >> 3;              // A chunk coords (x2, +32 rows)
+  const int a_c8 = (tid & 7) * 8;
+  short8v a_reg[2], w_reg[12];
+  const int g = (lane >> 3) & 7;           // TN transpose group
+  const int t_row = ((lane & 7) << 3) | (lane >> 3);  // TN source k-row
+
+  auto load_regs = [&](int k0) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int mi = m0 + a_row + i * 32;
+      a_reg[i] = short8v{};
       if (mi < M) {
-        val = *reinterpret_cast<const short8v*>(
-            &x[(int64_t)mi * K + k0 + c8]);
+        a_reg[i] = *reinterpret_cast<const short8v*>(
+            &x[(int64_t)mi * K + k0 + a_c8]);
       }
-      *reinterpret_cast<short8v*>(xs + swz128(row, c8 * 2)) = val;
     }
-    if (!TRANS_B) {
-      // stage W slice: 384 n-rows x 64 k, rows read natural-coalesced
-      for (int idx = tid; idx < BN * (BK / 8); idx += blockDim.x) {
-        int row = idx / (BK / 8);
-        int c8 = (idx % (BK / 8)) * 8;
-        short8v val = {};
-        int ni = n0 + row;
+#pragma unroll
+    for (int j = 0; j < 12; ++j) {
+      w_reg[j] = short8v{};
+      if (!TRANS_B) {
+        int ni = n0 + a_row + j * 32;      // same row/chunk walk as A
         if (ni < N) {
-          val = *reinterpret_cast<const short8v*>(
-              &w[(int64_t)ni * K + k0 + c8]);
+          w_reg[j] = *reinterpret_cast<const short8v*>(
+              &w[(int64_t)ni * K + k0 + a_c8]);
         }
-        *reinterpret_cast<short8v*>(ws + swz128(row, c8 * 2)) = val;
-      }
-    } else {
-      // stage W^T slice: source slab w[k0..k0+64, n0..n0+384] read
-      // row-natural (coalesced 16 B chunks), 8x8-transposed in-register,
-      // landing as [384 n-rows][64 k] in LDS. Row mapping keeps the
-      // 8-lane shuffle group on 8 consecutive source k-rows.
-      const int g = (lane >> 3) & 7;
-      for (int idx = tid; idx < (BK / 8) * BN; idx += blockDim.x) {
-        int j = idx >> 6;                         // n chunk (8 cols)
-        int sub = idx & 63;
-        int row = ((sub & 7) << 3) | (sub >> 3);  // k row, group-aligned
-        short8v val = {};
-        int ki = k0 + row;
-        int nj = n0 + 8 * j;
+      } else {
+        int jn = wid + j * 4;              // n chunk (8 cols)
+        int ki = k0 + t_row;
+        int nj = n0 + 8 * jn;
         if (ki < K && nj < N) {
-          val = *reinterpret_cast<const short8v*>(
+          w_reg[j] = *reinterpret_cast<const short8v*>(
               &w[(int64_t)ki * N + nj]);
         }
+      }
+    }
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      *reinterpret_cast<short8v*>(
+          xs + swz128(a_row + i * 32, a_c8 * 2)) = a_reg[i];
+    }
+#pragma unroll
+    for (int j = 0; j < 12; ++j) {
+      if (!TRANS_B) {
+        *reinterpret_cast<short8v*>(
+            ws + swz128(a_row + j * 32, a_c8 * 2)) = w_reg[j];
+      } else {
         short tv[8];
 #pragma unroll
-        for (int e = 0; e < 8; ++e) tv[e] = val[e];
+        for (int e = 0; e < 8; ++e) tv[e] = w_reg[j][e];
         xpose8x8g(tv, g);
         short8v pack;
 #pragma unroll
         for (int e = 0; e < 8; ++e) pack[e] = tv[e];
+        int jn = wid + j * 4;
         *reinterpret_cast<short8v*>(
-            ws + swz128(8 * j + g, ((row & ~7) * 2))) = pack;
+            ws + swz128(8 * jn + g, ((t_row & ~7) * 2))) = pack;
       }
     }
-    __syncthreads();
+  };
 
+  auto compute = [&]() {
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       short8v a[4];
@@ -164,7 +178,20 @@ skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,  // [M,K]
         }
       }
     }
-    __syncthreads();
+  };
+
+  const int n_tiles = (K + BK - 1) / BK;
+  load_regs(0);
+  write_lds();
+  __syncthreads();
+  for (int t = 0; t < n_tiles; ++t) {
+    if (t + 1 < n_tiles) load_regs((t + 1) * BK);  // overlaps compute(t)
+    compute();
+    __syncthreads();                 // all reads of tile t done
+    if (t + 1 < n_tiles) {
+      write_lds();                   // waits the in-flight loads here
+      __syncthreads();               // writes visible to every wave
+    }
   }
 
   // epilogue: C layout row=(lane>>4)*4+r (within m-frag), col=lane&15
