@@ -144,14 +144,24 @@ def main() -> None:
     # Extra UNTIMED warmup until GPU clocks settle: a fresh box ramps
     # sclk over the first ~2 s of load; short driver settings (warmup 5)
     # otherwise time the ramp. The timed region below is still exactly
-    # args.steps steps.
+    # args.steps steps. The extra count is derived from ONE timed step
+    # and MAX-agreed across ranks — every rank must run the same number
+    # of steps (each contains a collective for N>1) or the job deadlocks.
     if use_gpu:
         torch.cuda.synchronize()
         t_w = time.perf_counter()
-        i = args.warmup
-        while time.perf_counter() - t_w < 2.5:
-            step(i)
-            i += 1
+        step(args.warmup)
+        torch.cuda.synchronize()
+        dt = max(time.perf_counter() - t_w, 1e-4)
+        n_extra = min(int(2.5 / dt), 2000)
+        if world > 1:
+            import torch.distributed as dist
+
+            t = torch.tensor([n_extra], dtype=torch.int64, device=device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            n_extra = int(t.item())
+        for i in range(n_extra):
+            step(args.warmup + 1 + i)
         torch.cuda.synchronize()
 
     ctx.barrier()
