@@ -69,12 +69,24 @@ class RecommendationService:
         k = top_k or self.top_k
         b = len(histories)
         dim = self.sem_dim
-        max_len = max(max((len(h) for h in histories), default=1), 1)
-        max_len = min(max_len, self.max_items)
+        use_graph = self._graphed is not None and k == self.top_k
+        if use_graph:
+            # fixed shapes so ragged traffic reuses a handful of graphs:
+            # histories padded to max_items, batch rounded up to the next
+            # power of two (dummy rows sliced off the result)
+            max_len = self.max_items
+            b_pad = 1
+            while b_pad < b:
+                b_pad *= 2
+        else:
+            max_len = max(max((len(h) for h in histories), default=1), 1)
+            max_len = min(max_len, self.max_items)
+            b_pad = b
         L = max_len * dim
-        item_ids = torch.zeros(b, L, dtype=torch.long)
-        mask = torch.zeros(b, L, dtype=torch.long)
-        ttype = (torch.arange(L) % dim).unsqueeze(0).expand(b, -1).clone()
+        item_ids = torch.zeros(b_pad, L, dtype=torch.long)
+        mask = torch.zeros(b_pad, L, dtype=torch.long)
+        ttype = (torch.arange(L) % dim).unsqueeze(0).expand(b_pad, -1) \
+            .clone()
         n_items = self.item_sem_ids.size(0)
         sem_cpu = self.item_sem_ids.cpu()
         for i, h in enumerate(histories):
@@ -84,8 +96,10 @@ class RecommendationService:
                 if h else torch.zeros(0, dtype=torch.long)
             item_ids[i, :flat.numel()] = flat
             mask[i, :flat.numel()] = 1
-        users = torch.tensor(user_ids, dtype=torch.long).unsqueeze(1)
-        if self._graphed is not None and k == self.top_k:
+        users = torch.tensor(
+            list(user_ids) + [0] * (b_pad - b), dtype=torch.long) \
+            .unsqueeze(1)
+        if use_graph:
             gen = self._graphed(
                 user_input_ids=users.to(self.device),
                 item_input_ids=item_ids.to(self.device),
