@@ -278,7 +278,10 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   bool w_bf16 = w.scalar_type() == torch::kBFloat16;
   auto dx = torch::empty_like(x);
   dim3 block(256);
-  int n_blocks = std::min(grid_for_rows(n_rows, 4), 1024);
+  // 512 blocks keeps 2048 waves for the main pass (still ~8x over-
+  // subscribed at TIGER shapes) while halving the dw partial matrix the
+  // two-stage reduce has to chew through
+  int n_blocks = std::min(grid_for_rows(n_rows, 4), 512);
   dim3 grid(n_blocks);
   auto dw = torch::empty({n_blocks, (int64_t)d},
                          x.options().dtype(torch::kFloat32));

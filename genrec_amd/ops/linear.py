@@ -57,6 +57,28 @@ def _use_skinny(x: torch.Tensor, weight: torch.Tensor) -> bool:
     return k % 64 == 0 and rows >= 4096 and ops.has_ext()
 
 
+def _use_skinny_dx(dy: torch.Tensor, weight: torch.Tensor) -> bool:
+    """Wide-to-narrow dX backward (dy [M, C] @ W [C, N<=384]) through the
+    hand TN kernel: one 384-strip reads dy from HBM exactly once, where
+    hipBLASLt's macro-tiles measured ~4x off roofline at C>=1024
+    (MT128x128x32, 46.5 us on the TIGER qkv dX). Opt-out with
+    GENREC_DISABLE_SKINNY_DX=1."""
+    import os
+
+    if os.environ.get("GENREC_DISABLE_SKINNY_DX", "0") == "1":
+        return False
+    if not (dy.is_cuda and dy.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16):
+        return False
+    c, n = weight.shape
+    rows = dy.numel() // dy.shape[-1]
+    if not (rows >= 4096 and c >= 768 and n <= 384 and n % 8 == 0):
+        return False
+    from genrec_amd import ops
+
+    return ops.has_ext()
+
+
 class _SplitKLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, weight: torch.Tensor,
@@ -81,8 +103,17 @@ class _SplitKLinearFn(torch.autograd.Function):
         x, weight = ctx.saved_tensors
         # under autocast the saved weight/x may be fp32 while dy is bf16;
         # match dtypes explicitly (no-op casts on the pure-bf16 path)
-        dx = (dy.matmul(weight.to(dy.dtype))
-              if ctx.needs_input_grad[0] else None)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            if _use_skinny_dx(dy, weight):
+                from genrec_amd import ops
+
+                dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+                dx = ops.ext().skinny_gemm_tn(
+                    dy2, weight.contiguous(), None) \
+                    .view(*dy.shape[:-1], weight.shape[1])
+            else:
+                dx = dy.matmul(weight.to(dy.dtype))
         dw = None
         if ctx.needs_input_grad[1]:
             x2 = x.reshape(-1, x.shape[-1]).to(dy.dtype)

@@ -34,7 +34,19 @@ def main():
         t_blas = timeit(lambda: x @ w.t())
         t_hand = timeit(lambda: ops.ext().skinny_gemm(x, w, None))
         sol_us = (M * K + N * K + M * N) * 2 / 8e12 * 1e6
-        print(f"M={M:6d} N={N:5d} K={K:5d}: blaslt {t_blas:7.2f}us  "
+        print(f"NN M={M:6d} N={N:5d} K={K:5d}: blaslt {t_blas:7.2f}us  "
+              f"hand {t_hand:7.2f}us  ({t_blas / t_hand:4.2f}x, "
+              f"SOL~{sol_us:.2f}us)")
+
+    # TN (dX backward) family: dx[M,N] = dy[M,C] @ W[C,N]
+    for (M, C, N) in [(15616, 1152, 384), (15616, 1024, 384),
+                      (15616, 768, 384), (15616, 384, 384)]:
+        dy = torch.randn(M, C, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(C, N, device="cuda", dtype=torch.bfloat16)
+        t_blas = timeit(lambda: dy @ w)
+        t_hand = timeit(lambda: ops.ext().skinny_gemm_tn(dy, w, None))
+        sol_us = (M * C + C * N + M * N) * 2 / 8e12 * 1e6
+        print(f"TN M={M:6d} C={C:5d} N={N:5d}: blaslt {t_blas:7.2f}us  "
               f"hand {t_hand:7.2f}us  ({t_blas / t_hand:4.2f}x, "
               f"SOL~{sol_us:.2f}us)")
 
