@@ -59,13 +59,14 @@ def _use_skinny(x: torch.Tensor, weight: torch.Tensor) -> bool:
 
 def _use_skinny_dx(dy: torch.Tensor, weight: torch.Tensor) -> bool:
     """Wide-to-narrow dX backward (dy [M, C] @ W [C, N<=384]) through the
-    hand TN kernel: one 384-strip reads dy from HBM exactly once, where
-    hipBLASLt's macro-tiles measured ~4x off roofline at C>=1024
-    (MT128x128x32, 46.5 us on the TIGER qkv dX). Opt-out with
-    GENREC_DISABLE_SKINNY_DX=1."""
+    hand TN kernel. MEASURED SLOWER than hipBLASLt (111 vs 21 us at
+    C=1152: the in-register 8x8 transpose in the staging write pass
+    serializes the load pipeline — each shuffle forces a full vmcnt
+    wait), so opt-IN only (GENREC_SKINNY_DX=1); kept for kernel-layout
+    experiments."""
     import os
 
-    if os.environ.get("GENREC_DISABLE_SKINNY_DX", "0") == "1":
+    if os.environ.get("GENREC_SKINNY_DX", "0") != "1":
         return False
     if not (dy.is_cuda and dy.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16):
