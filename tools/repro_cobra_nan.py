@@ -1,0 +1,101 @@
+"""Localize the NaN in COBRA's pure-bf16 GraphedTrainStep path.
+
+Run on GPU: python -u tools/repro_cobra_nan.py
+Runs the fused step EAGERLY (use_graph=False), sweeps loss/grads/masters
+for the first non-finite value, then re-runs the offending step with
+forward hooks to name the first non-finite activation.
+"""
+
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    from genrec_amd.data.cobra_synthetic import (SyntheticCobraDataset,
+                                                 cobra_collate_fn)
+    from genrec_amd.models.cobra import Cobra
+    from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+    torch.manual_seed(42)
+    dev = "cuda:0"
+    ds = SyntheticCobraDataset(num_users=500, num_items=2000, split="train")
+    m = Cobra(encoder_n_layers=2, encoder_hidden_dim=384,
+              encoder_num_heads=6, encoder_vocab_size=32128,
+              id_vocab_size=256, n_codebooks=3, d_model=384,
+              decoder_n_layers=4, decoder_num_heads=6,
+              decoder_dropout=0.1).to(dev)
+    m.static_infonce = True
+    fixed = ds.max_items_per_seq + 1
+    bs = 32
+
+    def mk(i0):
+        b = cobra_collate_fn([ds[(i0 + j) % len(ds)] for j in range(bs)],
+                             m.pad_id, 3, train=True, fixed_items=fixed)
+        return {"input_ids": b["input_ids"].to(dev),
+                "encoder_input_ids": b["encoder_input_ids"].to(dev)}
+
+    runner = GraphedTrainStep(
+        m, mk(0), loss_getter=lambda o: o.loss_sparse + o.loss_dense,
+        lr=3e-4, weight_decay=0.01, clip_norm=1.0, world=1,
+        use_graph=False)
+
+    def bad(t):
+        return not torch.isfinite(t.float()).all()
+
+    off = []
+    o = 0
+    for p in runner.params:
+        off.append((o, o + p.numel()))
+        o += p.numel()
+    names = [n for n, p in m.named_parameters() if p.requires_grad]
+
+    for step in range(60):
+        batch = mk(step * bs)
+        loss = runner.step(batch)
+        lf = float(loss.detach().float())
+        gbad = bad(runner.flat_grads)
+        mbad = bad(runner.flat_master)
+        print(f"step {step:3d} loss {lf:9.4f} grads_bad={gbad} "
+              f"master_bad={mbad}", flush=True)
+        if lf != lf or gbad or mbad:
+            fg = runner.flat_grads.float()
+            fm = runner.flat_master
+            for (a, b), n in zip(off, names):
+                g = fg[a:b]
+                mm = fm[a:b]
+                if not torch.isfinite(g).all() or not torch.isfinite(mm).all():
+                    print(f"  BAD {n}: grad finite={torch.isfinite(g).all()}"
+                          f" absmax={g.abs().max():.3e} "
+                          f"master finite={torch.isfinite(mm).all()}",
+                          flush=True)
+            # name the first non-finite activation
+            hooks = []
+
+            def mkhook(name):
+                def h(mod, inp, out):
+                    ts = out if isinstance(out, (tuple, list)) else [out]
+                    for t in ts:
+                        if torch.is_tensor(t) and t.is_floating_point() \
+                                and not torch.isfinite(t.float()).all():
+                            print(f"  ACT-NAN {name} "
+                                  f"({type(mod).__name__})", flush=True)
+                            break
+                return h
+
+            for n, mod in m.named_modules():
+                hooks.append(mod.register_forward_hook(mkhook(n)))
+            out = m(**batch)
+            print(f"  refwd loss_sparse={float(out.loss_sparse):.4f} "
+                  f"loss_dense={float(out.loss_dense):.4f}", flush=True)
+            for h in hooks:
+                h.remove()
+            break
+    print("done", flush=True)
+
+
+if __name__ == "__main__":
+    main()
