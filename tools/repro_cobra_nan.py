@@ -38,10 +38,12 @@ def main():
         return {"input_ids": b["input_ids"].to(dev),
                 "encoder_input_ids": b["encoder_input_ids"].to(dev)}
 
+    use_graph = os.environ.get("NAN_GRAPH", "0") == "1"
     runner = GraphedTrainStep(
         m, mk(0), loss_getter=lambda o: o.loss_sparse + o.loss_dense,
         lr=3e-4, weight_decay=0.01, clip_norm=1.0, world=1,
-        use_graph=False)
+        use_graph=use_graph)
+    print("captured:", runner.captured, flush=True)
 
     def bad(t):
         return not torch.isfinite(t.float()).all()
