@@ -38,9 +38,22 @@ def main():
         return {"input_ids": b["input_ids"].to(dev),
                 "encoder_input_ids": b["encoder_input_ids"].to(dev)}
 
+    if os.environ.get("NAN_SDPA_MATH", "0") == "1":
+        torch.backends.cuda.enable_flash_sdp(False)
+        torch.backends.cuda.enable_mem_efficient_sdp(False)
+        print("sdpa: math only", flush=True)
+    loss_mode = os.environ.get("NAN_LOSS", "both")  # both|sparse|dense
+
+    def get_loss(o):
+        if loss_mode == "sparse":
+            return o.loss_sparse
+        if loss_mode == "dense":
+            return o.loss_dense
+        return o.loss_sparse + o.loss_dense
+
     use_graph = os.environ.get("NAN_GRAPH", "0") == "1"
     runner = GraphedTrainStep(
-        m, mk(0), loss_getter=lambda o: o.loss_sparse + o.loss_dense,
+        m, mk(0), loss_getter=get_loss,
         lr=3e-4, weight_decay=0.01, clip_norm=1.0, world=1,
         use_graph=use_graph)
     print("captured:", runner.captured, flush=True)
@@ -55,7 +68,7 @@ def main():
         o += p.numel()
     names = [n for n, p in m.named_parameters() if p.requires_grad]
 
-    for step in range(60):
+    for step in range(30):
         batch = mk(step * bs)
         loss = runner.step(batch)
         lf = float(loss.detach().float())
