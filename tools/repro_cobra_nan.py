@@ -57,6 +57,9 @@ def main():
         lr=3e-4, weight_decay=0.01, clip_norm=1.0, world=1,
         use_graph=use_graph)
     print("captured:", runner.captured, flush=True)
+    if os.environ.get("NAN_LR0", "0") == "1":
+        runner.set_lr(0.0)
+        print("lr pinned to 0 (params frozen)", flush=True)
 
     def bad(t):
         return not torch.isfinite(t.float()).all()
