@@ -79,6 +79,19 @@ def main():
         mbad = bad(runner.flat_master)
         print(f"step {step:3d} loss {lf:9.4f} grads_bad={gbad} "
               f"master_bad={mbad}", flush=True)
+        if mbad or gbad:
+            for nm, t in (("master", runner.flat_master),
+                          ("params", runner.flat_params.float()),
+                          ("m", runner.m), ("v", runner.v),
+                          ("grads", runner.flat_grads.float())):
+                nf = ~torch.isfinite(t)
+                cnt = int(nf.sum())
+                if cnt:
+                    idx = nf.nonzero().flatten()
+                    print(f"  {nm}: {cnt} nonfinite, first={int(idx[0])} "
+                          f"last={int(idx[-1])} of {t.numel()}", flush=True)
+                else:
+                    print(f"  {nm}: clean", flush=True)
         if lf != lf or gbad or mbad:
             fg = runner.flat_grads.float()
             fm = runner.flat_master
