@@ -28,6 +28,16 @@ def main():
               id_vocab_size=256, n_codebooks=3, d_model=384,
               decoder_n_layers=4, decoder_num_heads=6,
               decoder_dropout=0.1).to(dev)
+    if os.environ.get("NAN_NODROP", "0") == "1":
+        import torch.nn as nn
+        for mod in m.modules():
+            if isinstance(mod, nn.Dropout):
+                mod.p = 0.0
+            if isinstance(mod, nn.MultiheadAttention):
+                mod.dropout = 0.0
+        for layer in m.decoder.layers:
+            layer.dropout_p = 0.0
+        print("all dropout off", flush=True)
     m.static_infonce = True
     fixed = ds.max_items_per_seq + 1
     bs = 32
