@@ -175,11 +175,16 @@ class CobraDecoderLayer(nn.Module):
             query_mask=query_mask, dropout_p=self.dropout_p,
             training=self.training)
         att = self.out(att.transpose(1, 2).reshape(b, l, d))
-        x = self.norm1(x + F.dropout(att, self.dropout_p, self.training))
+        # genrec fused dropout kernels (device seed counter): ATen
+        # native_dropout produces corrupt values on hipGraph REPLAY on
+        # ROCm 7 (single-element NaN grads — tools/repro_cobra_nan.py)
+        x = self.norm1(ops.dropout_add(att, x, self.dropout_p,
+                                       self.training))
         x = self.norm2(x)  # vestigial cross-attn sublayer's norm
-        ff = self.linear2(F.dropout(F.relu(self.linear1(x)),
-                                    self.dropout_p, self.training))
-        return self.norm3(x + F.dropout(ff, self.dropout_p, self.training))
+        ff = self.linear2(ops.relu_dropout(self.linear1(x), self.dropout_p,
+                                           self.training))
+        return self.norm3(ops.dropout_add(ff, x, self.dropout_p,
+                                          self.training))
 
 
 class CobraDecoder(nn.Module):

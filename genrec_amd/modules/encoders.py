@@ -20,6 +20,72 @@ import torch.nn.functional as F
 from torch import Tensor, nn
 
 
+class CaptureSafeEncoderLayer(nn.Module):
+    """Post-LN transformer encoder layer, numerics-equivalent to torch's
+    nn.TransformerEncoderLayer (relu activation), built on the genrec
+    fused attention + dropout kernels: ATen native_dropout produces
+    corrupt values on hipGraph REPLAY on ROCm 7 (single-element NaN
+    gradients, localized with tools/repro_cobra_nan.py), and the
+    nested-tensor fast path does host-side mask syncs."""
+
+    def __init__(self, d_model: int, nhead: int, ff_dim: int,
+                 dropout: float) -> None:
+        super().__init__()
+        assert d_model % nhead == 0
+        self.h = nhead
+        self.hd = d_model // nhead
+        self.scale = 1.0 / (self.hd ** 0.5)
+        self.qkv = nn.Linear(d_model, 3 * d_model)
+        self.out = nn.Linear(d_model, d_model)
+        self.linear1 = nn.Linear(d_model, ff_dim)
+        self.linear2 = nn.Linear(ff_dim, d_model)
+        self.norm1 = nn.LayerNorm(d_model)
+        self.norm2 = nn.LayerNorm(d_model)
+        self.dropout_p = dropout
+
+    def forward(self, x: Tensor,
+                src_key_padding_mask: Optional[Tensor] = None) -> Tensor:
+        from genrec_amd import ops
+
+        b, l, d = x.shape
+        qkv = self.qkv(x).view(b, l, 3, self.h, self.hd) \
+            .permute(2, 0, 3, 1, 4)
+        att = ops.fused_attention(
+            qkv[0].contiguous(), qkv[1].contiguous(), qkv[2].contiguous(),
+            scale=self.scale, key_pad_mask=src_key_padding_mask,
+            dropout_p=self.dropout_p, training=self.training)
+        att = self.out(att.transpose(1, 2).reshape(b, l, d))
+        x = self.norm1(ops.dropout_add(att, x, self.dropout_p,
+                                       self.training))
+        ff = self.linear2(ops.relu_dropout(self.linear1(x), self.dropout_p,
+                                           self.training))
+        return self.norm2(ops.dropout_add(ff, x, self.dropout_p,
+                                          self.training))
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        # accept nn.TransformerEncoderLayer checkpoints
+        ren = {"self_attn.in_proj_weight": "qkv.weight",
+               "self_attn.in_proj_bias": "qkv.bias",
+               "self_attn.out_proj.weight": "out.weight",
+               "self_attn.out_proj.bias": "out.bias"}
+        for old, new in ren.items():
+            if prefix + old in state_dict:
+                state_dict[prefix + new] = state_dict.pop(prefix + old)
+        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
+
+
+class _EncoderStack(nn.Module):
+    def __init__(self, layers) -> None:
+        super().__init__()
+        self.layers = nn.ModuleList(layers)
+
+    def forward(self, x: Tensor,
+                src_key_padding_mask: Optional[Tensor] = None) -> Tensor:
+        for layer in self.layers:
+            x = layer(x, src_key_padding_mask)
+        return x
+
+
 class LightT5Encoder(nn.Module):
     def __init__(self, n_layers: int = 1, hidden_dim: int = 768,
                  output_dim: int = 768, num_heads: int = 8,
@@ -28,13 +94,9 @@ class LightT5Encoder(nn.Module):
         super().__init__()
         self.embedding = nn.Embedding(vocab_size, hidden_dim)
         self.pos_embedding = nn.Embedding(max_seq_len, hidden_dim)
-        layer = nn.TransformerEncoderLayer(
-            d_model=hidden_dim, nhead=num_heads, dim_feedforward=ff_dim,
-            dropout=dropout, batch_first=True)
-        # nested-tensor fast path does host-side mask inspection
-        # (.item() syncs) — illegal inside hipGraph capture
-        self.encoder = nn.TransformerEncoder(layer, num_layers=n_layers,
-                                             enable_nested_tensor=False)
+        self.encoder = _EncoderStack([
+            CaptureSafeEncoderLayer(hidden_dim, num_heads, ff_dim, dropout)
+            for _ in range(n_layers)])
         self.proj = nn.Linear(hidden_dim, output_dim)
         self.layer_norm = nn.LayerNorm(hidden_dim)
 

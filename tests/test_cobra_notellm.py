@@ -223,3 +223,28 @@ def test_cobra_static_infonce_equivalent(cobra):
     assert torch.allclose(out_sta.vec_cos_sim, out_dyn.vec_cos_sim,
                           atol=1e-5)
     assert torch.allclose(out_sta.loss, out_dyn.loss, atol=1e-5)
+
+
+def test_capture_safe_encoder_layer_matches_torch():
+    """CaptureSafeEncoderLayer == nn.TransformerEncoderLayer (post-LN,
+    relu) with weights copied via the back-compat state-dict mapping."""
+    import torch.nn as tnn
+    from genrec_amd.modules.encoders import CaptureSafeEncoderLayer
+
+    torch.manual_seed(0)
+    d, heads, ff, B, L = 48, 4, 96, 3, 9
+    ref = tnn.TransformerEncoderLayer(d_model=d, nhead=heads,
+                                      dim_feedforward=ff, dropout=0.0,
+                                      batch_first=True)
+    ours = CaptureSafeEncoderLayer(d, heads, ff, dropout=0.0)
+    ours.load_state_dict(ref.state_dict())  # via _load_from_state_dict
+    ref.eval()
+    ours.eval()
+    x = torch.randn(B, L, d)
+    pad = torch.zeros(B, L, dtype=torch.bool)
+    pad[:, -2:] = True
+    o_ref = ref(x, src_key_padding_mask=pad)
+    o_our = ours(x, src_key_padding_mask=pad)
+    valid = ~pad
+    assert torch.allclose(o_our[valid], o_ref[valid], atol=1e-5), \
+        (o_our[valid] - o_ref[valid]).abs().max()
