@@ -70,6 +70,9 @@ std::vector<torch::Tensor> dropout_add_fwd(
 std::vector<torch::Tensor> relu_dropout_fwd(
     torch::Tensor x, double p, int64_t seed,
     c10::optional<torch::Tensor> seed_dev);
+std::vector<torch::Tensor> plain_dropout_fwd(
+    torch::Tensor x, double p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev);
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
@@ -114,6 +117,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_bwd", &genrec::embedding_bwd, "embedding scatter-add bwd");
   m.def("dropout_add_fwd", &genrec::dropout_add_fwd, "residual+dropout fwd");
   m.def("relu_dropout_fwd", &genrec::relu_dropout_fwd, "dropout(relu) fwd");
+  m.def("plain_dropout_fwd", &genrec::plain_dropout_fwd,
+        "graph-replay-safe plain dropout fwd");
   m.def("dropout_fuse_bwd", &genrec::dropout_fuse_bwd, "fused dropout bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
   m.def("fused_adamw", &genrec::fused_adamw,

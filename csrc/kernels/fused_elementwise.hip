@@ -99,6 +99,14 @@ std::vector<torch::Tensor> relu_dropout_fwd(
   return dropout_fuse_fwd(x, c10::nullopt, p, seed, seed_dev, true);
 }
 
+// plain dropout(x) — ATen native_dropout corrupts under hipGraph replay
+// on ROCm 7 (BACKLOG round-2 hazard ledger)
+std::vector<torch::Tensor> plain_dropout_fwd(
+    torch::Tensor x, double p, int64_t seed,
+    c10::optional<torch::Tensor> seed_dev) {
+  return dropout_fuse_fwd(x, c10::nullopt, p, seed, seed_dev, false);
+}
+
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu) {
   const int64_t n = dy.numel();

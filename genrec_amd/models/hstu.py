@@ -136,8 +136,12 @@ class HSTULayer(nn.Module):
                                                 padding_mask)
         attn = attn.transpose(1, 2).reshape(b, l, d)
         attn = self.attn_norm(attn) * u
-        x = residual + self.dropout(attn)
-        return x + self.ffn(self.ffn_norm(x))
+        # replay-safe genrec dropout (hipGraph hazard ledger, BACKLOG)
+        x = ops.dropout_add(attn, residual, self.dropout.p, self.training)
+        h = self.ffn[0](self.ffn_norm(x))           # lin1
+        h = ops.plain_dropout(F.silu(h), self.ffn[2].p, self.training)
+        h = self.ffn[3](h)                          # lin2
+        return ops.dropout_add(h, x, self.ffn[4].p, self.training)
 
 
 @ginlite.configurable(name="HSTU")
@@ -186,8 +190,10 @@ class HSTU(nn.Module):
                 targets: Optional[Tensor] = None
                 ) -> Tuple[Optional[Tensor], Optional[Tensor]]:
         padding_mask = input_ids == 0
-        x = self.emb_dropout(ops.embedding(self.item_embedding.weight,
-                                           input_ids, padding_idx=0))
+        x = ops.plain_dropout(
+            ops.embedding(self.item_embedding.weight, input_ids,
+                          padding_idx=0),
+            self.emb_dropout.p, self.training)
         for layer in self.layers:
             x = layer(x, padding_mask, timestamps)
         x = self.final_norm(x)

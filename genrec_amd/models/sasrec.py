@@ -40,8 +40,11 @@ class PointWiseFeedForward(nn.Module):
         self.dropout = nn.Dropout(dropout)
 
     def forward(self, x: Tensor, residual: Tensor) -> Tensor:
-        out = self.fc2(self.dropout(F.relu(self.fc1(x))))
-        return self.dropout(out) + residual
+        # genrec fused dropout kernels: replay-safe under hipGraph
+        out = self.fc2(ops.relu_dropout(self.fc1(x), self.dropout.p,
+                                        self.training))
+        return ops.dropout_add(out, residual, self.dropout.p,
+                               self.training)
 
 
 class MultiHeadAttention(nn.Module):
@@ -135,7 +138,7 @@ class SASRec(nn.Module):
                           padding_idx=0) * (self.embed_dim ** 0.5)
         pos = torch.arange(l, device=input_ids.device).unsqueeze(0).expand(b, l)
         x = x + ops.embedding(self.position_embedding.weight, pos)
-        x = self.emb_dropout(x) * mask
+        x = ops.plain_dropout(x, self.emb_dropout.p, self.training) * mask
         for block in self.blocks:
             x = block(x, mask) * mask
         x = self.final_norm(x)

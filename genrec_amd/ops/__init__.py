@@ -87,7 +87,7 @@ from genrec_amd.ops.quantize import residual_quantize_step  # noqa: E402
 from genrec_amd.ops.losses import softmax_ce, tied_softmax_ce, summed_ce  # noqa: E402
 from genrec_amd.ops.metrics import topk_hit_ranks  # noqa: E402
 from genrec_amd.ops.embedding import embedding  # noqa: E402
-from genrec_amd.ops.fused import dropout_add, relu_dropout  # noqa: E402
+from genrec_amd.ops.fused import dropout_add, plain_dropout, relu_dropout  # noqa: E402
 
 __all__ = [
     "ext",
@@ -109,5 +109,6 @@ __all__ = [
     "topk_hit_ranks",
     "embedding",
     "dropout_add",
+    "plain_dropout",
     "relu_dropout",
 ]

@@ -82,7 +82,12 @@ def fused_attention(
     if query_mask is not None:
         attn = attn * query_mask[:, None, :, None]
     if dropout_p > 0.0 and training:
-        attn = F.dropout(attn, p=dropout_p, training=True)
+        if attn.is_cuda:  # replay-safe on GPU (ATen dropout corrupts
+            from genrec_amd.ops.fused import plain_dropout  # in replay)
+
+            attn = plain_dropout(attn, dropout_p, True)
+        else:
+            attn = F.dropout(attn, p=dropout_p, training=True)
     return torch.matmul(attn, v)
 
 

@@ -61,6 +61,38 @@ class _ReluDropoutFn(torch.autograd.Function):
         return dx, None
 
 
+class _PlainDropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: Tensor, p: float):
+        from genrec_amd import ops
+
+        seed, seed_dev = _seed_args(x.device, True)
+        out, mask = ops.ext().plain_dropout_fwd(x.contiguous(), p, seed,
+                                                seed_dev)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return out
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        from genrec_amd import ops
+
+        (mask,) = ctx.saved_tensors
+        dx = ops.ext().dropout_fuse_bwd(dy, mask, ctx.p, False)
+        return dx, None
+
+
+def plain_dropout(x: Tensor, p: float, training: bool) -> Tensor:
+    """dropout(x, p), hipGraph-replay-safe on GPU (ATen native_dropout
+    corrupts on replay, ROCm 7 — see BACKLOG hazard ledger)."""
+    from genrec_amd import ops
+
+    if training and p > 0.0 and ops.use_hip(x) \
+            and hasattr(ops.ext(), "plain_dropout_fwd"):
+        return _PlainDropoutFn.apply(x, p)
+    return F.dropout(x, p=p, training=training)
+
+
 def dropout_add(x: Tensor, residual: Tensor, p: float,
                 training: bool) -> Tensor:
     """residual + dropout(x, p) — fused on GPU."""
