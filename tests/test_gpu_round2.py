@@ -388,3 +388,28 @@ def test_graphed_generate_serving():
     # graph captured (not eager fallback)
     key = next(iter(gg._graphs))
     assert gg._graphs[key]["graph"] is not None, "capture failed"
+
+
+def test_captured_dropout_replay_stays_finite():
+    """Regression for the ROCm 7 replay hazards (BACKLOG ledger): a
+    dropout-active model captured in a hipGraph must keep finite losses
+    and gradients across many replays. ATen native_dropout corrupted on
+    the 2nd+ replay (single-element NaN grads -> clip-norm NaN -> all
+    masters NaN) before the genrec dropout kernels took over."""
+    from genrec_amd.models.sasrec import SASRec
+    from genrec_amd.parallel.graph_runner import GraphedTrainStep
+
+    torch.manual_seed(0)
+    m = SASRec(num_items=500, max_seq_len=30, embed_dim=64, num_heads=2,
+               num_blocks=2, ffn_dim=128, dropout=0.3).to(DEV)
+    ids = torch.randint(1, 501, (16, 30), device=DEV)
+    runner = GraphedTrainStep(m, {"input_ids": ids, "targets": ids},
+                              lambda out: out[1], lr=1e-3,
+                              weight_decay=0.0, clip_norm=1.0, world=1,
+                              use_graph=True)
+    assert runner.captured, "capture must succeed"
+    for i in range(10):
+        loss = runner.step({"input_ids": ids, "targets": ids})
+        assert torch.isfinite(loss.detach()), f"loss NaN at replay {i}"
+    assert torch.isfinite(runner.flat_grads.float()).all()
+    assert torch.isfinite(runner.flat_master).all()
