@@ -23,11 +23,14 @@ def main():
     torch.manual_seed(42)
     dev = "cuda:0"
     ds = SyntheticCobraDataset(num_users=500, num_items=2000, split="train")
-    m = Cobra(encoder_n_layers=2, encoder_hidden_dim=384,
-              encoder_num_heads=6, encoder_vocab_size=32128,
+    ehd = int(os.environ.get("NAN_EHD", "384"))
+    ehds = int(os.environ.get("NAN_EHEADS", "6"))
+    m = Cobra(encoder_n_layers=2, encoder_hidden_dim=ehd,
+              encoder_num_heads=ehds, encoder_vocab_size=32128,
               id_vocab_size=256, n_codebooks=3, d_model=384,
               decoder_n_layers=4, decoder_num_heads=6,
               decoder_dropout=0.1).to(dev)
+    print(f"encoder {ehd}/{ehds} heads", flush=True)
     if os.environ.get("NAN_NODROP", "0") == "1":
         import torch.nn as nn
         for mod in m.modules():
@@ -81,7 +84,8 @@ def main():
         o += p.numel()
     names = [n for n, p in m.named_parameters() if p.requires_grad]
 
-    for step in range(30):
+    n_steps = int(os.environ.get("NAN_STEPS", "30"))
+    for step in range(n_steps):
         batch = mk(step * bs)
         loss = runner.step(batch)
         lf = float(loss.detach().float())
