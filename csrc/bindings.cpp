@@ -73,6 +73,7 @@ std::vector<torch::Tensor> relu_dropout_fwd(
 std::vector<torch::Tensor> plain_dropout_fwd(
     torch::Tensor x, double p, int64_t seed,
     c10::optional<torch::Tensor> seed_dev);
+torch::Tensor colsum(torch::Tensor x);
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
@@ -119,6 +120,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_dropout_fwd", &genrec::relu_dropout_fwd, "dropout(relu) fwd");
   m.def("plain_dropout_fwd", &genrec::plain_dropout_fwd,
         "graph-replay-safe plain dropout fwd");
+  m.def("colsum", &genrec::colsum,
+        "deterministic replay-safe column sum (bias grads)");
   m.def("dropout_fuse_bwd", &genrec::dropout_fuse_bwd, "fused dropout bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
   m.def("fused_adamw", &genrec::fused_adamw,

@@ -107,6 +107,70 @@ std::vector<torch::Tensor> plain_dropout_fwd(
   return dropout_fuse_fwd(x, c10::nullopt, p, seed, seed_dev, false);
 }
 
+// Deterministic column sum out[j] = sum_r x[r, j] (fp32 accum), for
+// linear BIAS gradients: ATen's outer-dim reduce at these shapes uses a
+// semaphore-based two-pass whose state is not hipGraph-replay-safe on
+// ROCm 7 (sporadic corrupt elements in bias grads — BACKLOG hazard
+// ledger). Two fixed-order stages keep the chip busy and the result
+// bit-stable.
+constexpr int CS_CH = 32;
+
+template <typename T>
+__global__ void colsum1_kernel(const T* __restrict__ x,
+                               float* __restrict__ tmp,
+                               int64_t rows, int64_t cols) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= (int64_t)CS_CH * cols) return;
+  int64_t j = tid % cols;
+  int chunk = (int)(tid / cols);
+  float acc = 0.f;
+  for (int64_t r = chunk; r < rows; r += CS_CH)
+    acc += to_f32(x[r * cols + j]);
+  tmp[(int64_t)chunk * cols + j] = acc;
+}
+
+template <typename T>
+__global__ void colsum2_kernel(const float* __restrict__ tmp,
+                               T* __restrict__ out, int64_t cols) {
+  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= cols) return;
+  float acc = 0.f;
+#pragma unroll
+  for (int r = 0; r < CS_CH; ++r) acc += tmp[(int64_t)r * cols + j];
+  out[j] = from_f32<T>(acc);
+}
+
+torch::Tensor colsum(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  const int64_t rows = x.size(0), cols = x.size(1);
+  auto tmp = torch::empty({CS_CH, cols},
+                          x.options().dtype(torch::kFloat32));
+  auto out = torch::empty({cols}, x.options());
+  dim3 block(256);
+  dim3 grid1((unsigned)((CS_CH * cols + 255) / 256));
+  dim3 grid2((unsigned)((cols + 255) / 256));
+  auto stream = at::cuda::getCurrentHIPStream();
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((colsum1_kernel<float>), grid1, block, 0, stream,
+                       x.data_ptr<float>(), tmp.data_ptr<float>(), rows,
+                       cols);
+    hipLaunchKernelGGL((colsum2_kernel<float>), grid2, block, 0, stream,
+                       tmp.data_ptr<float>(), out.data_ptr<float>(), cols);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((colsum1_kernel<__hip_bfloat16>), grid1, block, 0,
+                       stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       tmp.data_ptr<float>(), rows, cols);
+    hipLaunchKernelGGL((colsum2_kernel<__hip_bfloat16>), grid2, block, 0,
+                       stream, tmp.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                       cols);
+  } else {
+    TORCH_CHECK(false, "colsum: unsupported dtype");
+  }
+  return out;
+}
+
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu) {
   const int64_t n = dy.numel();

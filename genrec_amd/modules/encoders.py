@@ -19,6 +19,8 @@ import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
 
+from genrec_amd.ops.linear import SplitKLinear
+
 
 class CaptureSafeEncoderLayer(nn.Module):
     """Post-LN transformer encoder layer, numerics-equivalent to torch's
@@ -35,10 +37,12 @@ class CaptureSafeEncoderLayer(nn.Module):
         self.h = nhead
         self.hd = d_model // nhead
         self.scale = 1.0 / (self.hd ** 0.5)
-        self.qkv = nn.Linear(d_model, 3 * d_model)
-        self.out = nn.Linear(d_model, d_model)
-        self.linear1 = nn.Linear(d_model, ff_dim)
-        self.linear2 = nn.Linear(ff_dim, d_model)
+        # SplitKLinear: bias grads go through the replay-safe colsum
+        # kernel (drop-in nn.Linear, same state-dict keys)
+        self.qkv = SplitKLinear(d_model, 3 * d_model)
+        self.out = SplitKLinear(d_model, d_model)
+        self.linear1 = SplitKLinear(d_model, ff_dim)
+        self.linear2 = SplitKLinear(ff_dim, d_model)
         self.norm1 = nn.LayerNorm(d_model)
         self.norm2 = nn.LayerNorm(d_model)
         self.dropout_p = dropout
@@ -97,7 +101,7 @@ class LightT5Encoder(nn.Module):
         self.encoder = _EncoderStack([
             CaptureSafeEncoderLayer(hidden_dim, num_heads, ff_dim, dropout)
             for _ in range(n_layers)])
-        self.proj = nn.Linear(hidden_dim, output_dim)
+        self.proj = SplitKLinear(hidden_dim, output_dim)
         self.layer_norm = nn.LayerNorm(hidden_dim)
 
     def forward(self, batch_tokens: Tensor) -> Tensor:

@@ -133,7 +133,19 @@ class _SplitKLinearFn(torch.autograd.Function):
                 dw = dy2.t().matmul(x2).to(weight.dtype)
         db = None
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = dy.reshape(-1, dy.shape[-1]).sum(0)
+            dy2b = dy.reshape(-1, dy.shape[-1])
+            if dy2b.is_cuda:
+                # deterministic replay-safe column sum: ATen's outer-dim
+                # reduce corrupts bias grads under hipGraph replay at
+                # these shapes (ROCm 7 — BACKLOG hazard ledger)
+                from genrec_amd import ops
+
+                if ops.has_ext():
+                    db = ops.ext().colsum(dy2b.contiguous())
+                else:
+                    db = dy2b.sum(0)
+            else:
+                db = dy2b.sum(0)
         return dx, dw, db
 
 
@@ -141,8 +153,13 @@ class SplitKLinear(nn.Linear):
     """nn.Linear with split-K grad-weight GEMMs (see module docstring)."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        # small-K layers take the stock C++ fast path: the custom Function
-        # only pays for itself when the split-K backward will be used
+        # CUDA + bias: always the custom Function, so the bias gradient
+        # takes the replay-safe colsum kernel (ATen's fused addmm
+        # backward reduce corrupts under hipGraph replay). Otherwise the
+        # custom Function only pays for itself when the split-K backward
+        # will be used.
+        if x.is_cuda and self.bias is not None:
+            return _SplitKLinearFn.apply(x, self.weight, self.bias)
         if x.numel() // x.shape[-1] < _SPLITK_MIN_K:
             return F.linear(x, self.weight, self.bias)
         return _SplitKLinearFn.apply(x, self.weight, self.bias)
