@@ -413,3 +413,40 @@ def test_captured_dropout_replay_stays_finite():
         assert torch.isfinite(loss.detach()), f"loss NaN at replay {i}"
     assert torch.isfinite(runner.flat_grads.float()).all()
     assert torch.isfinite(runner.flat_master).all()
+
+
+def test_colsum_matches_sum_and_replays():
+    """Replay-safe bias-grad column sum: parity with ATen sum(0) and
+    bit-stable across graph replays (ATen's outer-dim reduce corrupted
+    sporadically under replay — BACKLOG hazard 5)."""
+    from genrec_amd import ops
+
+    torch.manual_seed(0)
+    for rows, cols, dt in [(2688, 1152, torch.bfloat16),
+                           (2688, 2048, torch.bfloat16),
+                           (10752, 2304, torch.bfloat16),
+                           (1000, 384, torch.float32)]:
+        x = torch.randn(rows, cols, device=DEV, dtype=dt)
+        got = ops.ext().colsum(x)
+        ref = x.float().sum(0)
+        tol = 2.0 if dt == torch.bfloat16 else 1e-3
+        assert (got.float() - ref).abs().max() < tol
+
+    x = torch.randn(2688, 1152, device=DEV, dtype=torch.bfloat16)
+    out = ops.ext().colsum(x)  # warm
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        ops.ext().colsum(x)
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        out = ops.ext().colsum(x)
+    g.replay()
+    torch.cuda.synchronize()
+    first = out.clone()
+    for _ in range(8):
+        g.replay()
+    torch.cuda.synchronize()
+    assert torch.equal(out, first)
+    assert torch.isfinite(out.float()).all()
