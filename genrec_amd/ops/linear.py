@@ -86,6 +86,7 @@ class _SplitKLinearFn(torch.autograd.Function):
                 bias: Optional[torch.Tensor]):
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
+        ctx.bias_dtype = bias.dtype if bias is not None else None
         if _use_skinny(x, weight):
             from genrec_amd import ops
 
@@ -146,6 +147,8 @@ class _SplitKLinearFn(torch.autograd.Function):
                     db = dy2b.sum(0)
             else:
                 db = dy2b.sum(0)
+            if db.dtype != ctx.bias_dtype:  # autocast: fp32 bias leaf
+                db = db.to(ctx.bias_dtype)
         return dx, dw, db
 
 
