@@ -95,10 +95,13 @@ class _SplitKLinearFn(torch.autograd.Function):
                 x2, weight.contiguous(),
                 bias if bias is not None else None)
             return out.view(*x.shape[:-1], weight.shape[0])
-        out = x.matmul(weight.t())
         if bias is not None:
-            out = out + bias
-        return out
+            # fused addmm forward (same kernel F.linear uses); only the
+            # BACKWARD bias reduce is replaced (replay-safe colsum)
+            x2 = x.reshape(-1, x.shape[-1])
+            return torch.addmm(bias, x2, weight.t()) \
+                .view(*x.shape[:-1], weight.shape[0])
+        return x.matmul(weight.t())
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
