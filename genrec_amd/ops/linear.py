@@ -21,6 +21,7 @@ transformer.py:72-124 and tiger.py:161-207 of the reference.
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -164,7 +165,8 @@ class SplitKLinear(nn.Linear):
         # backward reduce corrupts under hipGraph replay). Otherwise the
         # custom Function only pays for itself when the split-K backward
         # will be used.
-        if x.is_cuda and self.bias is not None:
+        if x.is_cuda and self.bias is not None \
+                and os.environ.get("GENREC_UNSAFE_BIAS", "0") != "1":
             return _SplitKLinearFn.apply(x, self.weight, self.bias)
         if x.numel() // x.shape[-1] < _SPLITK_MIN_K:
             return F.linear(x, self.weight, self.bias)
