@@ -113,58 +113,64 @@ std::vector<torch::Tensor> plain_dropout_fwd(
 // ROCm 7 (sporadic corrupt elements in bias grads — BACKLOG hazard
 // ledger). Two fixed-order stages keep the chip busy and the result
 // bit-stable.
-constexpr int CS_CH = 32;
-
 template <typename T>
 __global__ void colsum1_kernel(const T* __restrict__ x,
                                float* __restrict__ tmp,
-                               int64_t rows, int64_t cols) {
+                               int64_t rows, int64_t cols, int ch) {
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (tid >= (int64_t)CS_CH * cols) return;
+  if (tid >= (int64_t)ch * cols) return;
   int64_t j = tid % cols;
   int chunk = (int)(tid / cols);
   float acc = 0.f;
-  for (int64_t r = chunk; r < rows; r += CS_CH)
+  for (int64_t r = chunk; r < rows; r += ch)
     acc += to_f32(x[r * cols + j]);
   tmp[(int64_t)chunk * cols + j] = acc;
 }
 
+// one 64-lane wave per column: lanes stride the ch partials, fixed-order
+// shuffle tree finishes (deterministic)
 template <typename T>
 __global__ void colsum2_kernel(const float* __restrict__ tmp,
-                               T* __restrict__ out, int64_t cols) {
-  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+                               T* __restrict__ out, int64_t cols, int ch) {
+  int64_t j = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+      + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
   if (j >= cols) return;
   float acc = 0.f;
-#pragma unroll
-  for (int r = 0; r < CS_CH; ++r) acc += tmp[(int64_t)r * cols + j];
-  out[j] = from_f32<T>(acc);
+  for (int r = lane; r < ch; r += WAVE) acc += tmp[(int64_t)r * cols + j];
+  acc = wave_sum(acc);
+  if (lane == 0) out[j] = from_f32<T>(acc);
 }
 
 torch::Tensor colsum(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
   const int64_t rows = x.size(0), cols = x.size(1);
-  auto tmp = torch::empty({CS_CH, cols},
-                          x.options().dtype(torch::kFloat32));
+  // enough chunks to fill the chip regardless of cols (small-col bias
+  // shapes starved at a fixed 32: 46 us at [6400, 256])
+  int ch = 32;
+  while ((int64_t)ch * cols < 131072 && ch < 512 && ch * 4 < rows) ch *= 2;
+  auto tmp = torch::empty({ch, cols}, x.options().dtype(torch::kFloat32));
   auto out = torch::empty({cols}, x.options());
   dim3 block(256);
-  dim3 grid1((unsigned)((CS_CH * cols + 255) / 256));
-  dim3 grid2((unsigned)((cols + 255) / 256));
+  dim3 grid1((unsigned)(((int64_t)ch * cols + 255) / 256));
+  dim3 grid2((unsigned)((cols + 3) / 4));  // 4 waves/block -> 4 cols
   auto stream = at::cuda::getCurrentHIPStream();
   if (x.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((colsum1_kernel<float>), grid1, block, 0, stream,
                        x.data_ptr<float>(), tmp.data_ptr<float>(), rows,
-                       cols);
+                       cols, ch);
     hipLaunchKernelGGL((colsum2_kernel<float>), grid2, block, 0, stream,
-                       tmp.data_ptr<float>(), out.data_ptr<float>(), cols);
+                       tmp.data_ptr<float>(), out.data_ptr<float>(), cols,
+                       ch);
   } else if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL((colsum1_kernel<__hip_bfloat16>), grid1, block, 0,
                        stream,
                        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-                       tmp.data_ptr<float>(), rows, cols);
+                       tmp.data_ptr<float>(), rows, cols, ch);
     hipLaunchKernelGGL((colsum2_kernel<__hip_bfloat16>), grid2, block, 0,
                        stream, tmp.data_ptr<float>(),
                        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
-                       cols);
+                       cols, ch);
   } else {
     TORCH_CHECK(false, "colsum: unsupported dtype");
   }
