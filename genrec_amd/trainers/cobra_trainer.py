@@ -69,6 +69,8 @@ def train(
     decoder_num_heads: int = 6,
     decoder_dropout: float = 0.1,
     encoder_n_layers: int = 1,
+    encoder_hidden_dim: int = 768,
+    encoder_num_heads: int = 8,
     encoder_type: str = "light",
     encoder_model_name: Optional[str] = None,
     dataset=None,
@@ -108,6 +110,8 @@ def train(
                   decoder_num_heads=decoder_num_heads,
                   decoder_dropout=decoder_dropout,
                   encoder_n_layers=encoder_n_layers,
+                  encoder_hidden_dim=encoder_hidden_dim,
+                  encoder_num_heads=encoder_num_heads,
                   encoder_type=encoder_type,
                   encoder_model_name=encoder_model_name).to(device)
     broadcast_parameters(model)

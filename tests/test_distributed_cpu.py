@@ -262,3 +262,40 @@ def test_grad_reducer_accumulation_micro_steps(tmp_path):
     expect = (results["local0"] + results["local1"]) / 2
     assert torch.allclose(results["got0"], expect, atol=1e-6)
     assert torch.allclose(results["got1"], expect, atol=1e-6)
+
+
+def _run_cobra_trainer(rank, world, port, results, tmpdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from genrec_amd.data.cobra_synthetic import SyntheticCobraDataset
+    from genrec_amd.trainers import cobra_trainer
+
+    class Tiny(SyntheticCobraDataset):
+        def __init__(self, **kw):
+            kw.update(num_users=40, num_items=60, text_vocab_size=500,
+                      id_vocab_size=16)
+            super().__init__(**kw)
+
+    cobra_trainer.train(
+        dataset=Tiny, epochs=1, max_steps=2, num_workers=0, batch_size=8,
+        save_dir_root=tmpdir, amp=False, do_eval=True, eval_every_epoch=1,
+        save_every_epoch=1, n_codebooks=3, id_vocab_size=16, d_model=32,
+        decoder_n_layers=1, decoder_num_heads=4, decoder_dropout=0.0,
+        encoder_n_layers=1, eval_n_beam=4, eval_max_batches=1,
+        num_warmup_steps=1)
+    results[rank] = True
+
+
+def test_cobra_trainer_world2(tmp_path):
+    """COBRA trainer under gloo world_size=2: DDP GradReducer + the
+    all-reduced beam-fusion eval metrics (same code path over RCCL)."""
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_run_cobra_trainer, args=(2, 29931, results, str(tmp_path)),
+             nprocs=2, join=True)
+    assert results[0] and results[1]
+    import os as _os
+
+    assert _os.path.exists(_os.path.join(str(tmp_path),
+                                         "checkpoint_final.pt"))
