@@ -32,13 +32,15 @@ std::vector<torch::Tensor> attn_fwd_mfma(
     c10::optional<torch::Tensor> add_mask,
     c10::optional<torch::Tensor> query_mask,
     double scale, bool causal, int64_t act, double dropout_p, int64_t seed,
-    c10::optional<torch::Tensor> seed_dev);
+    c10::optional<torch::Tensor> seed_dev,
+    c10::optional<torch::Tensor> bias_bucket);
 std::vector<torch::Tensor> attn_bwd_mfma(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor p_saved, torch::Tensor drop_mask,
     c10::optional<torch::Tensor> query_mask,
     double scale, int64_t act, double dropout_p, int64_t seed,
-    bool bias_grad, int64_t bias_dim);
+    bool bias_grad, int64_t bias_dim,
+    c10::optional<torch::Tensor> bias_bucket, int64_t n_buckets);
 
 std::vector<torch::Tensor> hstu_attn_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -107,8 +109,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_bwd", &genrec::l2norm_bwd, "fused L2Norm backward");
   m.def("attn_fwd", &genrec::attn_fwd, "fused attention forward");
   m.def("attn_bwd", &genrec::attn_bwd, "fused attention backward");
-  m.def("attn_fwd_mfma", &genrec::attn_fwd_mfma, "MFMA attention forward");
-  m.def("attn_bwd_mfma", &genrec::attn_bwd_mfma, "MFMA attention backward");
+  m.def("attn_fwd_mfma", &genrec::attn_fwd_mfma, "MFMA attention forward",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("bias"),
+        py::arg("key_pad"), py::arg("add_mask"), py::arg("query_mask"),
+        py::arg("scale"), py::arg("causal"), py::arg("act"),
+        py::arg("dropout_p"), py::arg("seed"), py::arg("seed_dev"),
+        py::arg("bias_bucket") = py::none());
+  m.def("attn_bwd_mfma", &genrec::attn_bwd_mfma, "MFMA attention backward",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("p_saved"), py::arg("drop_mask"), py::arg("query_mask"),
+        py::arg("scale"), py::arg("act"), py::arg("dropout_p"),
+        py::arg("seed"), py::arg("bias_grad"), py::arg("bias_dim"),
+        py::arg("bias_bucket") = py::none(),
+        py::arg("n_buckets") = 0);
   m.def("hstu_attn_fwd", &genrec::hstu_attn_fwd, "HSTU fused attention fwd");
   m.def("hstu_attn_bwd", &genrec::hstu_attn_bwd, "HSTU fused attention bwd");
   m.def("softmax_ce_fwd", &genrec::softmax_ce_fwd, "fused CE forward");

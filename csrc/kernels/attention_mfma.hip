@@ -27,6 +27,8 @@
 
 namespace genrec {
 
+torch::Tensor colsum(torch::Tensor x);  // fused_elementwise.hip
+
 constexpr int TILE = 64;               // max Lq/Lk per block
 constexpr float NEG_BIG_MF = -1e9f;
 
@@ -76,6 +78,8 @@ attn_fwd_mfma_kernel(
     const __hip_bfloat16* __restrict__ k,   // [B,H,Lk,D]
     const __hip_bfloat16* __restrict__ v,   // [B,H,Lk,D]
     const void* __restrict__ bias,          // null | [H,Lq,Lk] | [B,H,Lq,Lk]
+                                            // | table [H*nb] (bucket mode)
+    const int* __restrict__ bias_bucket,    // null | [Lq,Lk] table indices
     const bool* __restrict__ key_pad,       // null | [B,Lk]
     const float* __restrict__ add_mask,     // null | [Lq,Lk]
     const float* __restrict__ query_mask,   // null | [B,Lq]
@@ -83,7 +87,7 @@ attn_fwd_mfma_kernel(
     float* __restrict__ p_saved,            // [B,H,Lq,Lk]
     unsigned char* __restrict__ drop_mask,
     const unsigned int* __restrict__ seed_dev,
-    int B, int H, int Lq, int Lk, int D,
+    int B, int H, int Lq, int Lk, int D, int n_buckets,
     int64_t q_sb, int64_t q_sh, int64_t q_sl,
     int64_t k_sb, int64_t k_sh, int64_t k_sl,
     int64_t v_sb, int64_t v_sh, int64_t v_sl,
@@ -185,6 +189,10 @@ attn_fwd_mfma_kernel(
           int64_t bi = (bias_dim == 3) ? ((int64_t)h * Lq + i) * Lk + j
                                        : IDX4M(b, h, i, j, H, Lq, Lk);
           s += bias_bf16 ? to_f32(bias_b[bi]) : bias_f[bi];
+        } else if (bias_bucket) {
+          // in-kernel rel-bias table gather (HSTU-style): no
+          // materialized [H,Lq,Lk] bias tensor traffic
+          s += bias_f[h * n_buckets + bias_bucket[(int64_t)i * Lk + j]];
         }
         if (causal && j > i) s = NEG_BIG_MF;
         if (key_pad && key_pad[(int64_t)b * Lk + j]) s = NEG_BIG_MF;
@@ -321,6 +329,9 @@ attn_bwd_ds_kernel(
     __hip_bfloat16* __restrict__ ds_saved,    // null | [B,H,Lq,Lk] bias grad
                                               // (bf16: halves the HBM
                                               // round-trip; summed fp32)
+    const int* __restrict__ bias_bucket,      // null | [Lq,Lk]
+    float* __restrict__ dtab_partial,         // null | [B*H, n_buckets]
+    int n_buckets,
     int B, int H, int Lq, int Lk, int D,
     int64_t do_sb, int64_t do_sh, int64_t do_sl,
     int64_t q_sb, int64_t q_sh, int64_t q_sl,
@@ -343,6 +354,7 @@ attn_bwd_ds_kernel(
   char* dst = dsn + TILE * 128;     // [64(j)][128B(i)] dS^T (A for dK)
   char* adt = dst + TILE * 128;     // [64(j)][128B(i)] A_d^T (A for dV)
   float* dotbuf = reinterpret_cast<float*>(adt + TILE * 128);  // [2][64]
+  float* tab_lds = dotbuf + 2 * TILE;  // [n_buckets] table-grad partials
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -350,6 +362,8 @@ attn_bwd_ds_kernel(
   const int pair = wid >> 1;         // strip owner
   const int half = wid & 1;          // fragment split
   const int fbase = half * 2;
+
+  if (dtab_partial && tid < n_buckets) tab_lds[tid] = 0.f;
 
   for (int idx = tid; idx < TILE * (TILE / 8); idx += blockDim.x) {
     int row = idx / (TILE / 8);
@@ -504,6 +518,10 @@ attn_bwd_ds_kernel(
       if (ds_saved && i < Lq && j < Lk) {
         ds_saved[IDX4M(b, h, i, j, H, Lq, Lk)] = __float2bfloat16(dval);
       }
+      if (dtab_partial && i < Lq && j < Lk) {
+        // LDS-privatized table-grad accumulation (32-entry histogram)
+        atomicAdd(&tab_lds[bias_bucket[(int64_t)i * Lk + j]], dval);
+      }
     }
     int j = (fbase + fi) * 16 + col_base;
     *reinterpret_cast<short4v*>(dst + swz(j, i0 * 2)) = dpack;
@@ -513,6 +531,11 @@ attn_bwd_ds_kernel(
   // fragments, and dK/dV read all pairs' dS^T/A_d^T columns: one
   // workgroup barrier covers every consumer below.
   __syncthreads();
+  if (dtab_partial && tid < n_buckets) {
+    // one partial row per (b,h) block; reduced deterministically on the
+    // host via colsum over [B, H*nb]
+    dtab_partial[(int64_t)bh * n_buckets + tid] = tab_lds[tid];
+  }
 
   const int nfrag_d = (D + 15) / 16;
   {  // dQ[strip rows] = (scale*dS) @ K : output d-fragments split by half
@@ -614,7 +637,8 @@ std::vector<torch::Tensor> attn_fwd_mfma(
     c10::optional<torch::Tensor> add_mask,
     c10::optional<torch::Tensor> query_mask,
     double scale, bool causal, int64_t act, double dropout_p, int64_t seed,
-    c10::optional<torch::Tensor> seed_dev) {
+    c10::optional<torch::Tensor> seed_dev,
+    c10::optional<torch::Tensor> bias_bucket) {
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
@@ -633,10 +657,19 @@ std::vector<torch::Tensor> attn_fwd_mfma(
   } else {
     dmask = torch::empty({0}, q.options().dtype(torch::kUInt8));
   }
-  torch::Tensor bias_c;
+  torch::Tensor bias_c, bucket_c;
   int bias_dim = 0;
   bool bias_bf16 = false;
-  if (bias.has_value()) {
+  int n_buckets = 0;
+  if (bias_bucket.has_value()) {
+    // bucket mode: `bias` is the fp32 table [H*nb] (flattened rel-bias
+    // embedding weight), `bias_bucket` the [Lq,Lk] int32 index map
+    TORCH_CHECK(bias.has_value() &&
+                bias->scalar_type() == torch::kFloat32);
+    bias_c = bias->contiguous();
+    bucket_c = bias_bucket->to(torch::kInt32).contiguous();
+    n_buckets = (int)(bias_c.numel() / H);
+  } else if (bias.has_value()) {
     TORCH_CHECK(bias->scalar_type() == torch::kFloat32 ||
                 bias->scalar_type() == torch::kBFloat16);
     bias_c = bias->contiguous();
@@ -659,7 +692,8 @@ std::vector<torch::Tensor> attn_fwd_mfma(
       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),                   \
       reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),                   \
       reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),                   \
-      bias_dim ? bias_c.data_ptr() : nullptr,                                  \
+      (bias_dim || n_buckets) ? bias_c.data_ptr() : nullptr,                   \
+      n_buckets ? bucket_c.data_ptr<int>() : nullptr,                          \
       key_pad.has_value() ? key_pad->data_ptr<bool>() : nullptr,               \
       add_mask.has_value() ? am_f.data_ptr<float>() : nullptr,                 \
       query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,               \
@@ -669,7 +703,7 @@ std::vector<torch::Tensor> attn_fwd_mfma(
       seed_dev.has_value()                                                     \
           ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())        \
           : nullptr,                                                           \
-      B, H, Lq, Lk, D,                                                         \
+      B, H, Lq, Lk, D, n_buckets,                                              \
       q.stride(0), q.stride(1), q.stride(2),                                   \
       k.stride(0), k.stride(1), k.stride(2),                                   \
       v.stride(0), v.stride(1), v.stride(2),                                   \
@@ -688,7 +722,8 @@ std::vector<torch::Tensor> attn_bwd_mfma(
     torch::Tensor p_saved, torch::Tensor drop_mask,
     c10::optional<torch::Tensor> query_mask,
     double scale, int64_t act, double dropout_p, int64_t seed,
-    bool bias_grad, int64_t bias_dim) {
+    bool bias_grad, int64_t bias_dim,
+    c10::optional<torch::Tensor> bias_bucket, int64_t n_buckets) {
   (void)seed;
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
@@ -699,9 +734,14 @@ std::vector<torch::Tensor> attn_bwd_mfma(
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  torch::Tensor ds_saved;
+  torch::Tensor ds_saved, bucket_c, dtab_partial;
   __hip_bfloat16* ds_ptr = nullptr;
-  if (bias_grad) {
+  const bool table_mode = bias_grad && bias_bucket.has_value();
+  if (table_mode) {
+    bucket_c = bias_bucket->to(torch::kInt32).contiguous();
+    dtab_partial = torch::empty({(int64_t)B * H, n_buckets},
+                                q.options().dtype(torch::kFloat32));
+  } else if (bias_grad) {
     ds_saved = torch::empty({B, H, Lq, Lk},
                             q.options().dtype(torch::kBFloat16));
     ds_ptr = reinterpret_cast<__hip_bfloat16*>(ds_saved.data_ptr());
@@ -711,7 +751,8 @@ std::vector<torch::Tensor> attn_bwd_mfma(
     qm_f = query_mask->to(torch::kFloat32).contiguous();
   dim3 block(512);  // 8 waves: pairs own strips, halves split fragments
   dim3 grid(B * H);
-  size_t smem = 8 * TILE * 128 + 2 * TILE * sizeof(float);
+  size_t smem = 8 * TILE * 128 + 2 * TILE * sizeof(float)
+      + (table_mode ? (size_t)n_buckets * sizeof(float) : 0);
   auto stream = at::cuda::getCurrentHIPStream();
 
 #define LAUNCH_BWD_DS(SILU)                                                    \
@@ -726,6 +767,9 @@ std::vector<torch::Tensor> attn_bwd_mfma(
       reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),                        \
       reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),                        \
       reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()), ds_ptr,                \
+      table_mode ? bucket_c.data_ptr<int>() : nullptr,                         \
+      table_mode ? dtab_partial.data_ptr<float>() : nullptr,                   \
+      (int)n_buckets,                                                          \
       B, H, Lq, Lk, D,                                                         \
       dout.stride(0), dout.stride(1), dout.stride(2),                          \
       q.stride(0), q.stride(1), q.stride(2),                                   \
@@ -741,7 +785,11 @@ std::vector<torch::Tensor> attn_bwd_mfma(
 #undef LAUNCH_BWD_DS
 
   torch::Tensor dbias;
-  if (bias_grad) {
+  if (table_mode) {
+    // deterministic reduce of the per-block table-grad partials:
+    // [B*H, nb] -> view [B, H*nb] -> replay-safe colsum -> [H*nb]
+    dbias = colsum(dtab_partial.view({B, (int64_t)H * n_buckets}));
+  } else if (bias_grad) {
     if (bias_dim == 3) {  // bf16 per-element grads, fp32 batch reduce
       const int64_t inner = (int64_t)H * Lq * Lk;
       dbias = torch::empty({H, Lq, Lk},

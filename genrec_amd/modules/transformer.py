@@ -21,6 +21,7 @@ MI355X redesign notes:
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -113,6 +114,20 @@ class T5Attention(nn.Module):
         return ops.embedding(self.rel_bias.weight, idx).view(
             self.n_heads, q_len, k_len)
 
+    def _bucket_map(self, q_len: int, k_len: int, device) -> Tensor:
+        """[q_len, k_len] int32 bucket indices WITHOUT head offsets (the
+        kernel adds h*n_buckets) — cached per shape/device."""
+        key = ("bm", q_len, k_len, str(device))
+        bm = self._bucket_cache.get(key)
+        if bm is None:
+            ctx = torch.arange(q_len, device=device)[:, None]
+            mem = torch.arange(k_len, device=device)[None, :]
+            bm = relative_position_bucket(
+                mem - ctx, self.num_relative_buckets, self.max_distance,
+                bidirectional=True).to(torch.int32).contiguous()
+            self._bucket_cache[key] = bm
+        return bm
+
     def _split(self, x: Tensor) -> Tensor:
         b, l, _ = x.shape
         return x.view(b, l, self.n_heads, self.head_dim).transpose(1, 2)
@@ -148,17 +163,30 @@ class T5Attention(nn.Module):
                 key_padding_mask: Optional[Tensor]) -> Tensor:
 
         bias = None
+        bias_table = bias_bucket = None
         if self.rel_bias is not None:
-            # cached incremental decode: queries sit at the LAST q_len
-            # positions of the k_len-long sequence, so take the bottom
-            # rows of the full bias square
-            bias = self.compute_bias(k.size(2), k.size(2), q.device)
-            if q.size(2) != k.size(2):
-                bias = bias[:, k.size(2) - q.size(2):, :]
+            if os.environ.get("GENREC_ATTN_TABLE_BIAS", "1") == "1" \
+                    and q.is_cuda:
+                # in-kernel rel-bias table gather (round 2): no
+                # materialized [H,Lq,Lk] bias, backward accumulates table
+                # grads directly (LDS histogram + deterministic colsum)
+                bias_table = self.rel_bias.weight.view(-1)
+                bm = self._bucket_map(k.size(2), k.size(2), q.device)
+                if q.size(2) != k.size(2):  # cached incremental decode
+                    bm = bm[k.size(2) - q.size(2):, :].contiguous()
+                bias_bucket = bm
+            else:
+                # cached incremental decode: queries sit at the LAST
+                # q_len positions of the k_len-long sequence, so take the
+                # bottom rows of the full bias square
+                bias = self.compute_bias(k.size(2), k.size(2), q.device)
+                if q.size(2) != k.size(2):
+                    bias = bias[:, k.size(2) - q.size(2):, :]
 
         out = ops.t5_attention(
             q, k, v, bias, key_padding_mask, attn_mask, self.scale,
             self.dropout_p, self.training,
+            bias_table=bias_table, bias_bucket=bias_bucket,
         )
         b = out.size(0)
         out = out.transpose(1, 2).reshape(b, -1, self.d_model)

@@ -131,6 +131,59 @@ class _FusedAttnFn(torch.autograd.Function):
                 None, None, None, None, None, None, None, None)
 
 
+class _FusedAttnTableBiasFn(torch.autograd.Function):
+    """MFMA attention with the rel-bias TABLE gathered in-kernel
+    (HSTU-style; round 2): no materialized [H,Lq,Lk] bias tensor in
+    forward, and backward accumulates table grads via an LDS histogram
+    + deterministic colsum instead of the [B,H,Lq,Lk] ds_saved
+    round-trip. Caller guarantees the mfma path fits."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias_table, bias_bucket, key_pad_mask,
+                additive_mask, query_mask, scale, causal, dropout_p,
+                training):
+        from genrec_amd import ops
+
+        q = q if q.stride(-1) == 1 else q.contiguous()
+        k = k if k.stride(-1) == 1 else k.contiguous()
+        v = v if v.stride(-1) == 1 else v.contiguous()
+        table_f = bias_table.float()  # kernel reads fp32
+        seed_dev = None
+        seed = 0
+        if dropout_p > 0 and training:
+            seed = _call_seed()
+            seed_dev = _seed_counter(q.device)
+        out, probs, dmask = ops.ext().attn_fwd_mfma(
+            q, k, v, table_f, key_pad_mask, additive_mask, query_mask,
+            scale, causal, 0, dropout_p if training else 0.0, seed,
+            seed_dev, bias_bucket)
+        ctx.save_for_backward(
+            q, k, v, probs, dmask,
+            query_mask if query_mask is not None else torch.empty(0),
+            bias_bucket)
+        ctx.meta = (scale, dropout_p if training else 0.0,
+                    bias_table.requires_grad,
+                    bias_table.numel() // q.size(1), bias_table.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from genrec_amd import ops
+
+        q, k, v, probs, dmask, query_mask, bucket = ctx.saved_tensors
+        scale, dropout_p, bias_grad, n_buckets, tab_dtype = ctx.meta
+        if dout.stride(-1) != 1:
+            dout = dout.contiguous()
+        dq, dk, dv, dtab = ops.ext().attn_bwd_mfma(
+            dout, q, k, v, probs, dmask,
+            query_mask if query_mask.numel() else None,
+            scale, 0, dropout_p, 0, bias_grad, 0, bucket, n_buckets)
+        if bias_grad and dtab.dtype != tab_dtype:
+            dtab = dtab.to(tab_dtype)
+        return (dq, dk, dv, dtab if bias_grad else None, None,
+                None, None, None, None, None, None, None)
+
+
 class _FlashAttnFn(torch.autograd.Function):
     """Flash-tiled attention (Lk/Lq beyond one 64-tile): running-softmax
     forward, single-pass backward via the flash identity (tile math proven
@@ -183,6 +236,8 @@ def fused_attention(
     *,
     scale: float = 1.0,
     bias: Optional[Tensor] = None,
+    bias_table: Optional[Tensor] = None,
+    bias_bucket: Optional[Tensor] = None,
     key_pad_mask: Optional[Tensor] = None,
     additive_mask: Optional[Tensor] = None,
     causal: bool = False,
@@ -199,6 +254,25 @@ def fused_attention(
             # (recompute-P) kernels to trade HBM p_saved traffic for
             # recompute — measured per-config on GPU
             and os.environ.get("GENREC_FORCE_FLASH", "0") != "1")
+    if bias_table is not None:
+        # rel-bias TABLE mode: gather in-kernel when the mfma path fits
+        # (bf16, D % 32 == 0); otherwise materialize the dense bias and
+        # fall through to the ordinary paths
+        if (fits and act == _ACT_SOFTMAX and q.dtype == torch.bfloat16
+                and q.size(3) % 32 == 0
+                and os.environ.get("GENREC_DISABLE_MFMA", "0") != "1"
+                and _kernel_available("attn_fwd_mfma", q, k, v)):
+            kp = key_pad_mask.contiguous() if key_pad_mask is not None \
+                else None
+            am = additive_mask.contiguous() if additive_mask is not None \
+                else None
+            qm = query_mask.contiguous() if query_mask is not None else None
+            return _FusedAttnTableBiasFn.apply(
+                q, k, v, bias_table, bias_bucket.contiguous(), kp, am, qm,
+                scale, causal, dropout_p, training)
+        h = q.size(1)
+        bias = bias_table.view(h, -1)[:, bias_bucket.reshape(-1)] \
+            .view(h, bias_bucket.size(0), bias_bucket.size(1))
     if fits and _kernel_available("attn_fwd", q, k, v):
         b = bias.contiguous() if bias is not None else None
         kp = key_pad_mask.contiguous() if key_pad_mask is not None else None
@@ -236,10 +310,13 @@ def sasrec_attention(q, k, v, valid_mask, scale, dropout_p, training):
 
 
 def t5_attention(q, k, v, bias, key_pad_mask, additive_mask, scale,
-                 dropout_p, training):
-    """K13: bias [H,Lq,Lk] or None; additive_mask [Lq,Lk] float or None."""
+                 dropout_p, training, bias_table=None, bias_bucket=None):
+    """K13: bias [H,Lq,Lk] or None; additive_mask [Lq,Lk] float or None.
+    bias_table [H*nb] + bias_bucket [Lq,Lk] select the in-kernel
+    rel-bias gather (no materialized bias tensor)."""
     return fused_attention(
-        q, k, v, scale=scale, bias=bias, key_pad_mask=key_pad_mask,
+        q, k, v, scale=scale, bias=bias, bias_table=bias_table,
+        bias_bucket=bias_bucket, key_pad_mask=key_pad_mask,
         additive_mask=additive_mask, score_act="softmax",
         dropout_p=dropout_p, training=training,
     )

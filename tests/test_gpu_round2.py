@@ -480,3 +480,38 @@ def test_cobra_trainer_hip_graph_mode(tmp_path):
     for v in state["runner"].values():
         if torch.is_tensor(v):
             assert torch.isfinite(v.float()).all()
+
+
+def test_t5_table_bias_matches_dense(monkeypatch):
+    """In-kernel rel-bias table gather == the materialized-bias path
+    (fwd outputs and table gradients), TIGER encoder shapes."""
+    from genrec_amd.modules.transformer import T5Attention
+
+    torch.manual_seed(0)
+    att = T5Attention(d_model=384, n_heads=6, dropout=0.0).to(DEV) \
+        .to(torch.bfloat16)
+    att.train()
+    x = torch.randn(8, 61, 384, device=DEV, dtype=torch.bfloat16)
+    kp = torch.zeros(8, 61, dtype=torch.bool, device=DEV)
+    kp[:, -9:] = True
+
+    monkeypatch.setenv("GENREC_ATTN_TABLE_BIAS", "1")
+    out_t = att(x, key_padding_mask=kp)
+    out_t.float().sum().backward()
+    g_tab = att.rel_bias.weight.grad.clone()
+    gq_tab = att.qkv.weight.grad.clone()
+    att.zero_grad(set_to_none=True)
+
+    monkeypatch.setenv("GENREC_ATTN_TABLE_BIAS", "0")
+    out_d = att(x, key_padding_mask=kp)
+    out_d.float().sum().backward()
+    g_den = att.rel_bias.weight.grad.clone()
+    gq_den = att.qkv.weight.grad.clone()
+
+    assert torch.allclose(out_t.float(), out_d.float(), atol=2e-2), \
+        (out_t.float() - out_d.float()).abs().max()
+    assert torch.allclose(g_tab.float(), g_den.float(), atol=0.5,
+                          rtol=0.05), \
+        (g_tab.float() - g_den.float()).abs().max()
+    assert torch.allclose(gq_tab.float(), gq_den.float(), atol=2e-2,
+                          rtol=0.05)
