@@ -165,7 +165,7 @@ class T5Attention(nn.Module):
         bias = None
         bias_table = bias_bucket = None
         if self.rel_bias is not None:
-            if os.environ.get("GENREC_ATTN_TABLE_BIAS", "1") == "1" \
+            if os.environ.get("GENREC_ATTN_TABLE_BIAS", "0") == "1" \
                     and q.is_cuda:
                 # in-kernel rel-bias table gather (round 2): no
                 # materialized [H,Lq,Lk] bias, backward accumulates table
