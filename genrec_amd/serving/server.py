@@ -133,7 +133,10 @@ class RecommendationService:
         state = torch.load(checkpoint_path, map_location="cpu",
                            weights_only=False)
         sd = state["model"] if "model" in state else state
-        mk = model_kwargs or {}
+        # model_config saved by the trainers wins unless overridden
+        mk = dict(state.get("model_config") or {}) \
+            if isinstance(state, dict) else {}
+        mk.update(model_kwargs or {})
         model = Tiger(**mk)
         model.load_state_dict(sd)
         sem = torch.load(sem_ids_path, map_location="cpu",
