@@ -80,13 +80,19 @@ def sinkhorn_knopp(cost: Tensor, row_marginals: Tensor, col_marginals: Tensor,
     step across the 2 x max_iter calls); the mul+reduce pair is ~10x
     faster for the same math (round-2 rocprofv3 profile).
     """
+    import os
+
     K = torch.exp(-cost / eps)
-    Kt = K  # reuse; column sums expressed on K directly
     u = torch.ones_like(row_marginals)
     v = torch.ones_like(col_marginals)
-    for _ in range(max_iter):
-        u = row_marginals / ((K * v.unsqueeze(0)).sum(1) + 1e-8)
-        v = col_marginals / ((Kt * u.unsqueeze(1)).sum(0) + 1e-8)
+    if os.environ.get("GENREC_SINKHORN_GEMV", "0") == "1":  # A/B knob
+        for _ in range(max_iter):
+            u = row_marginals / (K @ v + 1e-8)
+            v = col_marginals / (K.T @ u + 1e-8)
+    else:
+        for _ in range(max_iter):
+            u = row_marginals / ((K * v.unsqueeze(0)).sum(1) + 1e-8)
+            v = col_marginals / ((K * u.unsqueeze(1)).sum(0) + 1e-8)
     return u.unsqueeze(1) * K * v.unsqueeze(0)
 
 
