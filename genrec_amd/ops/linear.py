@@ -41,21 +41,29 @@ def _pick_chunks(k: int) -> int:
 
 
 def _use_skinny(x: torch.Tensor, weight: torch.Tensor) -> bool:
-    """Hand MFMA GEMM for the tall-skinny fwd family (BACKLOG r1 item 4):
-    bf16, K % 64 == 0, many rows. Opt-in (GENREC_SKINNY_GEMM=1) until the
-    measured win generalizes across shapes."""
+    """Hand MFMA GEMM (T14-pipelined staging) for the tall-skinny fwd
+    family where it measures FASTER than hipBLASLt: many rows, narrow
+    N <= 768 and K <= 512 (the exact-strip shapes; ~9 vs 11.7 us at
+    M=15616 N=K=384). Wider shapes re-read the A strip per 384-tile and
+    lose — they stay on hipBLASLt. GENREC_SKINNY_GEMM=0 disables,
+    =all forces every eligible shape (A/B)."""
     import os
 
-    if os.environ.get("GENREC_SKINNY_GEMM", "0") != "1":
+    mode = os.environ.get("GENREC_SKINNY_GEMM", "1")
+    if mode == "0":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16):
         return False
-    k = weight.shape[1]
+    n, k = weight.shape
     rows = x.numel() // x.shape[-1]
+    if not (k % 64 == 0 and rows >= 4096):
+        return False
+    if mode != "all" and not (n <= 768 and k <= 512):
+        return False
     from genrec_amd import ops
 
-    return k % 64 == 0 and rows >= 4096 and ops.has_ext()
+    return ops.has_ext()
 
 
 def _use_skinny_dx(dy: torch.Tensor, weight: torch.Tensor) -> bool:
