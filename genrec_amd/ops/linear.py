@@ -42,14 +42,14 @@ def _pick_chunks(k: int) -> int:
 
 def _use_skinny(x: torch.Tensor, weight: torch.Tensor) -> bool:
     """Hand MFMA GEMM (T14-pipelined staging) for the tall-skinny fwd
-    family where it measures FASTER than hipBLASLt: many rows, narrow
-    N <= 768 and K <= 512 (the exact-strip shapes; ~9 vs 11.7 us at
-    M=15616 N=K=384). Wider shapes re-read the A strip per 384-tile and
-    lose — they stay on hipBLASLt. GENREC_SKINNY_GEMM=0 disables,
-    =all forces every eligible shape (A/B)."""
+    family. Measured SLOWER than hipBLASLt in-graph even on its best
+    shapes (46.7 vs 47.4k bench with the narrow-shape-only gate; the
+    eager microbench win was launch-floor artifact), so OPT-IN:
+    GENREC_SKINNY_GEMM=1 enables narrow shapes, =all every eligible
+    shape."""
     import os
 
-    mode = os.environ.get("GENREC_SKINNY_GEMM", "1")
+    mode = os.environ.get("GENREC_SKINNY_GEMM", "0")
     if mode == "0":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16
