@@ -123,31 +123,50 @@ class GraphedTrainStep:
         self._graph = None
         self._loss = None
         if use_graph and device.type == "cuda":
+            # Warmup + capture + verification replay each run the full
+            # optimizer update on the example batch; snapshot the
+            # optimizer state first and restore it after (success OR
+            # failure), so real training starts from step 0 exactly as
+            # eager mode would.
+            snap = self._opt_snapshot()
+            graph = None
             try:
-                # Warmup + capture + verification replay each run the full
-                # optimizer update on the example batch; snapshot the
-                # optimizer state first and restore it after, so real
-                # training starts from step 0 exactly as eager mode would.
-                snap = self._opt_snapshot()
                 s = torch.cuda.Stream()
                 s.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(s):
                     for _ in range(warmup_iters):
                         self._inner()
                 torch.cuda.current_stream().wait_stream(s)
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
                     self._loss = self._inner()
-                graph.replay()
-                torch.cuda.synchronize()
-                self._opt_restore(snap)
-                self._graph = graph
-                logger.info("GraphedTrainStep: hipGraph capture active")
+                graph = g
             except Exception as e:
                 logger.warning(
                     "GraphedTrainStep: capture failed (%s); running the "
                     "same step eagerly", e)
-                self._graph = None
+            if self.world > 1:
+                # all-or-none capture across ranks: a rank whose capture
+                # failed would otherwise skip the verify replay's
+                # all_reduce (and any per-step launch asymmetries) and
+                # deadlock the job. Capture RECORDS without executing
+                # collectives, so counts are still aligned here.
+                import torch.distributed as dist
+
+                ok = torch.tensor([1 if graph is not None else 0],
+                                  device=device)
+                dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+                if int(ok.item()) == 0 and graph is not None:
+                    logger.warning(
+                        "GraphedTrainStep: capture failed on another "
+                        "rank; all ranks falling back to eager")
+                    graph = None
+            if graph is not None:
+                graph.replay()  # verify (uniform across ranks)
+                torch.cuda.synchronize()
+                self._graph = graph
+                logger.info("GraphedTrainStep: hipGraph capture active")
+            self._opt_restore(snap)
 
     def _opt_snapshot(self):
         with torch.no_grad():
