@@ -34,6 +34,88 @@ __global__ void dropout_fuse_fwd_kernel(const T* __restrict__ x,
   }
 }
 
+// bf16x4 vectorized variants (guide Guideline 13): 2-byte-per-lane loads
+// leave half the memory pipe idle, so the bf16 paths move 4 elements per
+// thread as 8-byte words (mask as a 4-byte word). The per-element RNG
+// stream (hash_rng(s, i)) and mask encoding are IDENTICAL to the scalar
+// kernels, so fwd/bwd stay paired and replays stay deterministic.
+// GENREC_SCALAR_ELEMWISE=1 forces the scalar kernels (A/B switch).
+
+union BF16x4 {
+  uint2 u;
+  __hip_bfloat16 e[4];
+};
+union U8x4 {
+  unsigned int u;
+  unsigned char e[4];
+};
+
+template <bool RELU, bool HAS_RES>
+__global__ void dropout_fuse_fwd_v4_kernel(
+    const uint2* __restrict__ x, const uint2* __restrict__ residual,
+    uint2* __restrict__ out, unsigned int* __restrict__ mask, int64_t n4,
+    float p, float inv_keep, unsigned int seed,
+    const unsigned int* __restrict__ seed_dev) {
+  unsigned int s = seed + (seed_dev ? *seed_dev : 0u);
+  const unsigned int thresh = (unsigned int)(p * 16777216.0f);
+  for (int64_t q = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; q < n4;
+       q += (int64_t)gridDim.x * blockDim.x) {
+    BF16x4 xi, ri, oo;
+    xi.u = x[q];
+    if (HAS_RES) ri.u = residual[q];
+    U8x4 mm;
+    const int64_t base = q * 4;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float v = to_f32(xi.e[k]);
+      if (RELU) v = fmaxf(v, 0.f);
+      bool keep = (hash_rng(s, base + k) & 0xFFFFFF) >= thresh;
+      mm.e[k] = keep ? (RELU ? (v > 0.f ? 3 : 1) : 1)
+                     : (RELU && v > 0.f ? 2 : 0);
+      v = keep ? v * inv_keep : 0.f;
+      if (HAS_RES) v += to_f32(ri.e[k]);
+      oo.e[k] = from_f32<__hip_bfloat16>(v);
+    }
+    out[q] = oo.u;
+    mask[q] = mm.u;
+  }
+}
+
+template <bool RELU>
+__global__ void dropout_fuse_bwd_v4_kernel(const uint2* __restrict__ dy,
+                                           const unsigned int* __restrict__ mask,
+                                           uint2* __restrict__ dx, int64_t n4,
+                                           float inv_keep) {
+  for (int64_t q = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; q < n4;
+       q += (int64_t)gridDim.x * blockDim.x) {
+    BF16x4 g, o;
+    g.u = dy[q];
+    U8x4 mm;
+    mm.u = mask[q];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      unsigned char m = mm.e[k];
+      bool keep = m & 1;
+      bool relu_pass = !RELU || (m & 2);
+      o.e[k] = from_f32<__hip_bfloat16>(
+          (keep && relu_pass) ? to_f32(g.e[k]) * inv_keep : 0.f);
+    }
+    dx[q] = o.u;
+  }
+}
+
+static inline bool elemwise_force_scalar() {
+  static const bool v = [] {
+    const char* e = getenv("GENREC_SCALAR_ELEMWISE");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
+
+static inline bool ptr_aligned8(const void* p) {
+  return (reinterpret_cast<uintptr_t>(p) & 7) == 0;
+}
+
 // mask bits: bit0 = dropout keep, bit1 = relu pass (x > 0)
 template <typename T, bool RELU>
 __global__ void dropout_fuse_bwd_kernel(const T* __restrict__ dy,
@@ -61,6 +143,36 @@ static std::vector<torch::Tensor> dropout_fuse_fwd(
   dim3 block(256);
   dim3 grid((unsigned)std::min<int64_t>((n + 255) / 256, 4096));
   auto stream = at::cuda::getCurrentHIPStream();
+
+  // bf16x4 fast path (see kernel comment)
+  if (x.scalar_type() == torch::kBFloat16 && (n & 3) == 0 && n > 0 &&
+      ptr_aligned8(x.data_ptr()) && ptr_aligned8(out.data_ptr()) &&
+      (!residual.has_value() || ptr_aligned8(residual->data_ptr())) &&
+      !elemwise_force_scalar()) {
+    const int64_t n4 = n >> 2;
+    dim3 grid4((unsigned)std::min<int64_t>((n4 + 255) / 256, 4096));
+#define LAUNCH_DF4(RELU, HAS_RES)                                              \
+  hipLaunchKernelGGL((dropout_fuse_fwd_v4_kernel<RELU, HAS_RES>), grid4,       \
+      block, 0, stream, reinterpret_cast<const uint2*>(x.data_ptr()),          \
+      residual.has_value()                                                     \
+          ? reinterpret_cast<const uint2*>(residual->data_ptr())               \
+          : nullptr,                                                           \
+      reinterpret_cast<uint2*>(out.data_ptr()),                                \
+      reinterpret_cast<unsigned int*>(mask.data_ptr<unsigned char>()), n4,     \
+      (float)p, inv_keep, (unsigned int)seed,                                  \
+      seed_dev.has_value()                                                     \
+          ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())        \
+          : nullptr)
+    if (relu) {
+      if (residual.has_value()) LAUNCH_DF4(true, true);
+      else LAUNCH_DF4(true, false);
+    } else {
+      if (residual.has_value()) LAUNCH_DF4(false, true);
+      else LAUNCH_DF4(false, false);
+    }
+#undef LAUNCH_DF4
+    return {out, mask};
+  }
 
 #define LAUNCH_DF(T, RELU)                                                     \
   hipLaunchKernelGGL((dropout_fuse_fwd_kernel<T, RELU>), grid, block, 0,       \
@@ -186,6 +298,22 @@ torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
   dim3 block(256);
   dim3 grid((unsigned)std::min<int64_t>((n + 255) / 256, 4096));
   auto stream = at::cuda::getCurrentHIPStream();
+
+  if (dy.scalar_type() == torch::kBFloat16 && (n & 3) == 0 && n > 0 &&
+      ptr_aligned8(dyc.data_ptr()) && ptr_aligned8(dx.data_ptr()) &&
+      ptr_aligned8(mask.data_ptr()) && !elemwise_force_scalar()) {
+    const int64_t n4 = n >> 2;
+    dim3 grid4((unsigned)std::min<int64_t>((n4 + 255) / 256, 4096));
+#define LAUNCH_DB4(RELU)                                                       \
+  hipLaunchKernelGGL((dropout_fuse_bwd_v4_kernel<RELU>), grid4, block, 0,      \
+      stream, reinterpret_cast<const uint2*>(dyc.data_ptr()),                  \
+      reinterpret_cast<const unsigned int*>(mask.data_ptr<unsigned char>()),   \
+      reinterpret_cast<uint2*>(dx.data_ptr()), n4, inv_keep)
+    if (relu) LAUNCH_DB4(true);
+    else LAUNCH_DB4(false);
+#undef LAUNCH_DB4
+    return dx;
+  }
 #define LAUNCH_DB(T, RELU)                                                     \
   hipLaunchKernelGGL((dropout_fuse_bwd_kernel<T, RELU>), grid, block, 0,       \
       stream, reinterpret_cast<const T*>(dyc.data_ptr()),                      \

@@ -3,13 +3,16 @@
 //
 // Design: one 64-lane wave per row (rows are small: D = 32..768 across the
 // zoo), 4 waves per 256-thread block, grid-stride over rows. bf16 rows are
-// loaded vectorized (short4 reinterpret = 8 B/lane) per Guideline 13; all
+// loaded/stored as column-pair dwords (4 B/lane) per Guideline 13; all
 // math accumulates in fp32, matching the reference's fp32-upcast semantics
 // (normalize.py:38-55, 73-95). Backward reduces dweight per-block in LDS
-// then atomically into a fp32 accumulator.
+// into one partial row per block, finished by a fixed-order two-stage
+// column reduce (deterministic, no atomics).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+
+#include <type_traits>
 
 #include "../core/common.h"
 
@@ -117,6 +120,161 @@ __global__ void rms_norm_bwd_kernel(const OutT* __restrict__ dy,
   }
 }
 
+// bf16x2-vectorized variants (guide Guideline 13): 2-byte-per-lane loads
+// leave half the memory pipe idle, so the bf16 paths load/store column
+// PAIRS (4 B/lane dwords). Math, rounding points and the dw partial
+// layout are identical to the scalar kernels; only the fp32 accumulation
+// order inside a lane changes (pairs instead of 64-strided singles).
+// GENREC_SCALAR_NORMS=1 forces the scalar kernels (A/B switch).
+
+union BF16x2 {
+  unsigned int u;
+  __hip_bfloat16 e[2];
+};
+
+template <typename WT>
+__device__ __forceinline__ void load_w2(const WT* __restrict__ w, int j2,
+                                        float* wf) {
+  if constexpr (std::is_same<WT, float>::value) {
+    float2 wv = reinterpret_cast<const float2*>(w)[j2];
+    wf[0] = wv.x;
+    wf[1] = wv.y;
+  } else {
+    BF16x2 wv;
+    wv.u = reinterpret_cast<const unsigned int*>(w)[j2];
+    wf[0] = to_f32(wv.e[0]);
+    wf[1] = to_f32(wv.e[1]);
+  }
+}
+
+template <typename WT, bool T5_STYLE>
+__global__ void rms_norm_fwd_v2_kernel(const unsigned int* __restrict__ x,
+                                       const WT* __restrict__ w,
+                                       unsigned int* __restrict__ y,
+                                       float* __restrict__ inv_rms,
+                                       int64_t n_rows, int d2, float eps,
+                                       bool w_is_half) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = gridDim.x * blockDim.x / WAVE;
+  const int d = d2 * 2;
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const unsigned int* xr = x + row * d2;
+    float ss = 0.f;
+    for (int j = lane; j < d2; j += WAVE) {
+      BF16x2 v;
+      v.u = xr[j];
+      float a = to_f32(v.e[0]), b = to_f32(v.e[1]);
+      ss += a * a + b * b;
+    }
+    ss = wave_sum(ss);
+    float r = rsqrtf(ss / d + eps);
+    if (lane == 0) inv_rms[row] = r;
+    unsigned int* yr = y + row * d2;
+    for (int j = lane; j < d2; j += WAVE) {
+      BF16x2 v, o;
+      v.u = xr[j];
+      float wf[2];
+      load_w2(w, j, wf);
+#pragma unroll
+      for (int k = 0; k < 2; ++k) {
+        float xf = to_f32(v.e[k]);
+        float out;
+        if (T5_STYLE) {
+          float t = xf * r;
+          if (w_is_half) t = to_f32(from_f32<__hip_bfloat16>(t));
+          out = wf[k] * t;
+        } else {
+          float t = xf * r;
+          t = to_f32(from_f32<__hip_bfloat16>(t));
+          out = t * wf[k];
+        }
+        o.e[k] = from_f32<__hip_bfloat16>(out);
+      }
+      yr[j] = o.u;
+    }
+  }
+}
+
+template <typename WT>
+__global__ void rms_norm_bwd_v2_kernel(const unsigned int* __restrict__ dy,
+                                       const unsigned int* __restrict__ x,
+                                       const WT* __restrict__ w,
+                                       const float* __restrict__ inv_rms,
+                                       unsigned int* __restrict__ dx,
+                                       float* __restrict__ dw,
+                                       int64_t n_rows, int d2) {
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = (int64_t)blockIdx.x * waves_per_block + wave_in_block;
+  const int64_t n_waves = (int64_t)gridDim.x * waves_per_block;
+  const int d = d2 * 2;
+
+  float dw_acc[RMS_MAX_COLS_PER_LANE];
+#pragma unroll
+  for (int c = 0; c < RMS_MAX_COLS_PER_LANE; ++c) dw_acc[c] = 0.f;
+
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const unsigned int* xr = x + row * d2;
+    const unsigned int* dyr = dy + row * d2;
+    float r = inv_rms[row];
+    float dot = 0.f;
+    for (int j = lane; j < d2; j += WAVE) {
+      BF16x2 xv, gv;
+      xv.u = xr[j];
+      gv.u = dyr[j];
+      float wf[2];
+      load_w2(w, j, wf);
+      dot += wf[0] * to_f32(gv.e[0]) * to_f32(xv.e[0]) +
+             wf[1] * to_f32(gv.e[1]) * to_f32(xv.e[1]);
+    }
+    dot = wave_sum(dot);
+    float c = r * r * r / d * dot;
+    unsigned int* dxr = dx + row * d2;
+    int ci = 0;
+    for (int j = lane; j < d2; j += WAVE, ci += 2) {
+      BF16x2 xv, gv, o;
+      xv.u = xr[j];
+      gv.u = dyr[j];
+      float wf[2];
+      load_w2(w, j, wf);
+#pragma unroll
+      for (int k = 0; k < 2; ++k) {
+        float xf = to_f32(xv.e[k]);
+        float dyf = to_f32(gv.e[k]);
+        o.e[k] = from_f32<__hip_bfloat16>(wf[k] * dyf * r - c * xf);
+        dw_acc[ci + k] += dyf * xf * r;
+      }
+      dxr[j] = o.u;
+    }
+  }
+  extern __shared__ __attribute__((aligned(16))) float dw_lds[];  // [4][d]
+  int ci = 0;
+  for (int j = lane; j < d2; j += WAVE, ci += 2) {
+    dw_lds[wave_in_block * d + 2 * j] = dw_acc[ci];
+    dw_lds[wave_in_block * d + 2 * j + 1] = dw_acc[ci + 1];
+  }
+  __syncthreads();
+  for (int j = (int)threadIdx.x; j < d; j += (int)blockDim.x) {
+    float s = dw_lds[j] + dw_lds[d + j] + dw_lds[2 * d + j] +
+              dw_lds[3 * d + j];
+    dw[(int64_t)blockIdx.x * d + j] = s;
+  }
+}
+
+static inline bool norms_force_scalar() {
+  static const bool v = [] {
+    const char* e = getenv("GENREC_SCALAR_NORMS");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
+
+static inline bool ptr_aligned4(const void* p) {
+  return (reinterpret_cast<uintptr_t>(p) & 3) == 0;
+}
+
 // ---------------------------------------------------------------- L2Norm
 
 template <typename T>
@@ -204,6 +362,31 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
   dim3 grid(grid_for_rows(n_rows, 4));
   auto stream = at::cuda::getCurrentHIPStream();
 
+  // bf16x2 fast path: dword loads/stores (see kernel comment)
+  if (x.scalar_type() == torch::kBFloat16 && out_dtype == torch::kBFloat16 &&
+      (d & 1) == 0 && ptr_aligned4(x.data_ptr()) &&
+      ptr_aligned4(y.data_ptr()) &&
+      (reinterpret_cast<uintptr_t>(wc.data_ptr()) & (w_bf16 ? 3 : 7)) == 0 &&
+      !norms_force_scalar()) {
+#define LAUNCH_RMS_V2(WT, T5)                                                  \
+  hipLaunchKernelGGL((rms_norm_fwd_v2_kernel<WT, T5>), grid, block, 0,         \
+                     stream,                                                   \
+                     reinterpret_cast<const unsigned int*>(x.data_ptr()),      \
+                     reinterpret_cast<const WT*>(wc.data_ptr()),               \
+                     reinterpret_cast<unsigned int*>(y.data_ptr()),            \
+                     inv_rms.data_ptr<float>(), n_rows, d / 2, (float)eps,     \
+                     w_half)
+    if (w_bf16) {
+      if (t5_style) LAUNCH_RMS_V2(__hip_bfloat16, true);
+      else LAUNCH_RMS_V2(__hip_bfloat16, false);
+    } else {
+      if (t5_style) LAUNCH_RMS_V2(float, true);
+      else LAUNCH_RMS_V2(float, false);
+    }
+#undef LAUNCH_RMS_V2
+    return {y, inv_rms};
+  }
+
 #define LAUNCH_RMS(T, WT, OutT, T5)                                            \
   hipLaunchKernelGGL((rms_norm_fwd_kernel<T, WT, OutT, T5>), grid, block, 0,   \
                      stream, reinterpret_cast<const T*>(x.data_ptr()),         \
@@ -240,31 +423,33 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
 // sum(0) + .to() pair (~12+5 us per call, ~22 calls/step on TIGER).
 // Two stages so the reduce has enough resident waves (a single-stage
 // d-thread loop left only 6 waves on the whole chip and was LOSING to
-// ATen): stage 1 folds n_part rows 32-fold with CH*d threads, stage 2
-// finishes 32 rows with d threads. Fixed-order loops: deterministic.
-constexpr int RED_CH = 32;
+// ATen). Chunk count is adaptive like colsum's (a fixed 32 chunks left
+// only 12k threads at d=384 — 7.3 us for a 768 KB read): stage 1 folds
+// n_part rows ch-fold with ch*d threads, stage 2 finishes ch rows with
+// one 64-lane wave per column. Fixed-order loops: deterministic.
 
 __global__ void rms_dw_reduce1_kernel(const float* __restrict__ part,
                                       float* __restrict__ tmp,
-                                      int n_part, int d) {
+                                      int n_part, int d, int ch) {
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (tid >= (int64_t)RED_CH * d) return;
+  if (tid >= (int64_t)ch * d) return;
   int j = (int)(tid % d);
   int chunk = (int)(tid / d);
   float acc = 0.f;
-  for (int r = chunk; r < n_part; r += RED_CH) acc += part[(int64_t)r * d + j];
+  for (int r = chunk; r < n_part; r += ch) acc += part[(int64_t)r * d + j];
   tmp[(int64_t)chunk * d + j] = acc;
 }
 
 template <typename WT>
 __global__ void rms_dw_reduce2_kernel(const float* __restrict__ tmp,
-                                      WT* __restrict__ out, int d) {
-  int j = blockIdx.x * blockDim.x + threadIdx.x;
+                                      WT* __restrict__ out, int d, int ch) {
+  int j = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
   if (j >= d) return;
   float acc = 0.f;
-#pragma unroll
-  for (int r = 0; r < RED_CH; ++r) acc += tmp[(int64_t)r * d + j];
-  out[j] = from_f32<WT>(acc);
+  for (int r = lane; r < ch; r += WAVE) acc += tmp[(int64_t)r * d + j];
+  acc = wave_sum(acc);
+  if (lane == 0) out[j] = from_f32<WT>(acc);
 }
 
 std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
@@ -288,6 +473,27 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   size_t smem = 4 * (size_t)d * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
 
+  bool vec2_ok = x.scalar_type() == torch::kBFloat16 &&
+                 dy.scalar_type() == torch::kBFloat16 && (d & 1) == 0 &&
+                 ptr_aligned4(x.data_ptr()) && ptr_aligned4(dy.data_ptr()) &&
+                 ptr_aligned4(dx.data_ptr()) &&
+                 (reinterpret_cast<uintptr_t>(wc.data_ptr()) &
+                  (w_bf16 ? 3 : 7)) == 0 &&
+                 !norms_force_scalar();
+  if (vec2_ok) {
+#define LAUNCH_RMSB_V2(WT)                                                     \
+  hipLaunchKernelGGL((rms_norm_bwd_v2_kernel<WT>), grid, block, smem, stream,  \
+                     reinterpret_cast<const unsigned int*>(dy.data_ptr()),     \
+                     reinterpret_cast<const unsigned int*>(x.data_ptr()),      \
+                     reinterpret_cast<const WT*>(wc.data_ptr()),               \
+                     inv_rms.data_ptr<float>(),                                \
+                     reinterpret_cast<unsigned int*>(dx.data_ptr()),           \
+                     dw.data_ptr<float>(), n_rows, d / 2)
+    if (w_bf16) LAUNCH_RMSB_V2(__hip_bfloat16);
+    else LAUNCH_RMSB_V2(float);
+#undef LAUNCH_RMSB_V2
+  }
+
 #define LAUNCH_RMSB(T, WT, OutT)                                               \
   hipLaunchKernelGGL((rms_norm_bwd_kernel<T, WT, OutT>), grid, block, smem,    \
                      stream, reinterpret_cast<const OutT*>(dy.data_ptr()),     \
@@ -303,7 +509,9 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
     else LAUNCH_RMSB(T, float, OutT);                                          \
   } while (0)
 
-  if (x.scalar_type() == torch::kFloat32) {
+  if (vec2_ok) {
+    // already launched above
+  } else if (x.scalar_type() == torch::kFloat32) {
     LAUNCH_RMSB_W(float, float);
   } else if (x.scalar_type() == torch::kBFloat16) {
     if (dy.scalar_type() == torch::kFloat32)
@@ -315,23 +523,25 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
 #undef LAUNCH_RMSB_W
 #undef LAUNCH_RMSB
   auto dw_out = torch::empty({(int64_t)d}, w.options());
-  auto dw_tmp = torch::empty({RED_CH, (int64_t)d},
+  int ch = 32;
+  while ((int64_t)ch * d < 131072 && ch < 512 && ch * 4 < n_blocks) ch *= 2;
+  auto dw_tmp = torch::empty({ch, (int64_t)d},
                              x.options().dtype(torch::kFloat32));
   dim3 rblock(256);
-  dim3 rgrid1((unsigned)(((int64_t)RED_CH * d + 255) / 256));
+  dim3 rgrid1((unsigned)(((int64_t)ch * d + 255) / 256));
   hipLaunchKernelGGL(rms_dw_reduce1_kernel, rgrid1, rblock, 0, stream,
                      dw.data_ptr<float>(), dw_tmp.data_ptr<float>(),
-                     n_blocks, d);
-  dim3 rgrid2((d + 255) / 256);
+                     n_blocks, d, ch);
+  dim3 rgrid2((unsigned)((d + 3) / 4));  // 4 waves/block, one wave per col
   if (w_bf16) {
     hipLaunchKernelGGL((rms_dw_reduce2_kernel<__hip_bfloat16>), rgrid2,
                        rblock, 0, stream, dw_tmp.data_ptr<float>(),
                        reinterpret_cast<__hip_bfloat16*>(dw_out.data_ptr()),
-                       d);
+                       d, ch);
   } else {
     hipLaunchKernelGGL((rms_dw_reduce2_kernel<float>), rgrid2, rblock, 0,
                        stream, dw_tmp.data_ptr<float>(),
-                       dw_out.data_ptr<float>(), d);
+                       dw_out.data_ptr<float>(), d, ch);
   }
   return {dx, dw_out};
 }
