@@ -76,6 +76,7 @@ std::vector<torch::Tensor> plain_dropout_fwd(
     torch::Tensor x, double p, int64_t seed,
     c10::optional<torch::Tensor> seed_dev);
 torch::Tensor colsum(torch::Tensor x);
+torch::Tensor chunk_sum(torch::Tensor x);
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
@@ -135,6 +136,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "graph-replay-safe plain dropout fwd");
   m.def("colsum", &genrec::colsum,
         "deterministic replay-safe column sum (bias grads)");
+  m.def("chunk_sum", &genrec::chunk_sum,
+        "small-row-count column sum (split-K dW partials)");
   m.def("dropout_fuse_bwd", &genrec::dropout_fuse_bwd, "fused dropout bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
   m.def("fused_adamw", &genrec::fused_adamw,

@@ -138,10 +138,24 @@ class _SplitKLinearFn(torch.autograd.Function):
                 part = torch.bmm(
                     dy2.view(nc, k // nc, -1).transpose(1, 2),
                     x2.view(nc, k // nc, -1))
-                # ATen reduces bf16 sums in fp32 (acc_type), so one
-                # bf16-out sum keeps single-rounding numerics without a
-                # separate cast kernel
-                dw = part.sum(0).to(weight.dtype)
+                # fp32-accumulated chunk reduce. GENREC_CHUNK_SUM=1 swaps
+                # ATen's sum(0) for the genrec chunk_sum kernel — BITWISE
+                # identical (both accumulate bf16 in fp32, same row
+                # order) but measured perf-NEUTRAL on the TIGER bench
+                # (49.0k vs 48.9k same-box): the ~6 us per-dispatch
+                # average rocprof shows for ATen's reduce is launch-ramp
+                # inflation, not stream time. Opt-in, same policy as the
+                # skinny GEMM.
+                from genrec_amd import ops
+
+                n_el = part.shape[1] * part.shape[2]
+                if part.is_cuda and ops.has_ext() and n_el % 4 == 0 \
+                        and os.environ.get("GENREC_CHUNK_SUM") == "1":
+                    dw = ops.ext().chunk_sum(part.view(nc, n_el)) \
+                        .view(part.shape[1], part.shape[2]) \
+                        .to(weight.dtype)
+                else:
+                    dw = part.sum(0).to(weight.dtype)
             else:
                 dw = dy2.t().matmul(x2).to(weight.dtype)
         db = None
