@@ -471,25 +471,25 @@ def test_chunk_sum_matches_sum():
             assert (got - ref).abs().max() < 1e-3
 
 
-def test_splitk_dw_uses_chunk_sum(monkeypatch):
-    """SplitKLinear's chunked dW path with the opt-in chunk_sum reduce
-    matches the plain-GEMM gradient (relative tolerance: per-chunk bf16
-    partials round before the final sum)."""
+def test_splitk_dw_chunk_sum_matches_aten_reduce(monkeypatch):
+    """The opt-in chunk_sum reduce gives BITWISE the same dW as the
+    default ATen sum(0) over the same chunked bf16 partials (both
+    accumulate fp32 in the same row order). Split-K-vs-plain-GEMM
+    numerics are covered by the graph-vs-eager step tests; this pins
+    only what GENREC_CHUNK_SUM changes."""
     from genrec_amd.ops.linear import SplitKLinear
 
-    monkeypatch.setenv("GENREC_CHUNK_SUM", "1")
     torch.manual_seed(4)
     lin = SplitKLinear(384, 384, bias=False).to(DEV, torch.bfloat16)
-    x = torch.randn(15616, 384, device=DEV, dtype=torch.bfloat16,
-                    requires_grad=True)
-    lin(x).float().sum().backward()
-    dw_fast = lin.weight.grad.clone().float()
-    w = lin.weight.detach().clone().requires_grad_(True)
-    y = x.detach().requires_grad_(True).matmul(w.t())
-    y.float().sum().backward()
-    ref = w.grad.float()
-    rel = (dw_fast - ref).abs() / (ref.abs() + 1.0)
-    assert rel.max() < 2e-2
+    x = torch.randn(15616, 384, device=DEV, dtype=torch.bfloat16)
+    grads = {}
+    for mode in ("0", "1"):
+        monkeypatch.setenv("GENREC_CHUNK_SUM", mode)
+        lin.weight.grad = None
+        xi = x.clone().requires_grad_(True)
+        lin(xi).float().sum().backward()
+        grads[mode] = lin.weight.grad.clone()
+    assert torch.equal(grads["0"], grads["1"])
 
 
 def test_cobra_trainer_hip_graph_mode(tmp_path):
