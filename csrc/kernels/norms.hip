@@ -463,10 +463,18 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
   bool w_bf16 = w.scalar_type() == torch::kBFloat16;
   auto dx = torch::empty_like(x);
   dim3 block(256);
-  // 512 blocks keeps 2048 waves for the main pass (still ~8x over-
-  // subscribed at TIGER shapes) while halving the dw partial matrix the
-  // two-stage reduce has to chew through
-  int n_blocks = std::min(grid_for_rows(n_rows, 4), 512);
+  // Block cap: 1024 blocks = 4096 waves (4 blocks/CU) hide the per-row
+  // wave_sum serialization at the big [15616, 384] shapes — same-box A/B
+  // vs the old 512 cap: 48.3k -> 49.5k samples/s on the TIGER bench
+  // (2048 flat vs 1024). The dw partial matrix grows with the cap but
+  // the adaptive two-stage reduce absorbs it. GENREC_RMS_BWD_CAP
+  // overrides for A/B.
+  static const int cap = [] {
+    const char* e = getenv("GENREC_RMS_BWD_CAP");
+    int v = e ? atoi(e) : 1024;
+    return (v >= 64 && v <= 4096) ? v : 1024;
+  }();
+  int n_blocks = std::min(grid_for_rows(n_rows, 4), cap);
   dim3 grid(n_blocks);
   auto dw = torch::empty({n_blocks, (int64_t)d},
                          x.options().dtype(torch::kFloat32));
