@@ -22,6 +22,15 @@
 
 namespace genrec {
 
+// Element-quad layout (float4 / 8-byte-bf16x4 loads): the scalar version
+// moved 28 B/element through seven dword-or-narrower streams and measured
+// 70 us/step on TIGER (~3.5x the ~20 us HBM roofline). The last partial
+// quad falls back to per-element accesses in the same kernel.
+union ADW_BF4 {
+  uint2 u;
+  __hip_bfloat16 e[4];
+};
+
 __global__ void fused_adamw_kernel(
     float* __restrict__ master, const __hip_bfloat16* __restrict__ grad,
     float* __restrict__ m, float* __restrict__ v,
@@ -35,17 +44,49 @@ __global__ void fused_adamw_kernel(
   const float bc1 = 1.0f - powf(beta1, st);
   const float bc2 = 1.0f - powf(beta2, st);
   const float decay = 1.0f - lr * wd;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    float g = gscale * to_f32(grad[i]);
-    float p = master[i] * decay;
-    float mi = beta1 * m[i] + (1.0f - beta1) * g;
-    float vi = beta2 * v[i] + (1.0f - beta2) * g * g;
-    m[i] = mi;
-    v[i] = vi;
-    p -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
-    master[i] = p;
-    out_p[i] = from_f32<__hip_bfloat16>(p);
+  const int64_t nq = (n + 3) >> 2;
+  for (int64_t q = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; q < nq;
+       q += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t i0 = q * 4;
+    if (i0 + 3 < n) {
+      ADW_BF4 g4;
+      g4.u = reinterpret_cast<const uint2*>(grad)[q];
+      float4 p4 = reinterpret_cast<const float4*>(master)[q];
+      float4 m4 = reinterpret_cast<const float4*>(m)[q];
+      float4 v4 = reinterpret_cast<const float4*>(v)[q];
+      float* pp = &p4.x;
+      float* mp = &m4.x;
+      float* vp = &v4.x;
+      ADW_BF4 o4;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float g = gscale * to_f32(g4.e[k]);
+        float p = pp[k] * decay;
+        float mi = beta1 * mp[k] + (1.0f - beta1) * g;
+        float vi = beta2 * vp[k] + (1.0f - beta2) * g * g;
+        mp[k] = mi;
+        vp[k] = vi;
+        p -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        pp[k] = p;
+        o4.e[k] = from_f32<__hip_bfloat16>(p);
+      }
+      reinterpret_cast<float4*>(m)[q] = m4;
+      reinterpret_cast<float4*>(v)[q] = v4;
+      reinterpret_cast<float4*>(master)[q] = p4;
+      reinterpret_cast<uint2*>(out_p)[q] = o4.u;
+    } else {
+      for (int64_t i = i0; i < n; ++i) {
+        float g = gscale * to_f32(grad[i]);
+        float p = master[i] * decay;
+        float mi = beta1 * m[i] + (1.0f - beta1) * g;
+        float vi = beta2 * v[i] + (1.0f - beta2) * g * g;
+        m[i] = mi;
+        v[i] = vi;
+        p -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        master[i] = p;
+        out_p[i] = from_f32<__hip_bfloat16>(p);
+      }
+    }
   }
 }
 
@@ -72,7 +113,9 @@ void fused_adamw(torch::Tensor master, torch::Tensor grad, torch::Tensor m,
   hipLaunchKernelGGL(step_inc_kernel, dim3(1), dim3(64), 0, stream,
                      step.data_ptr<int>());
   const int threads = 256;
-  const int blocks = (int)std::min<int64_t>(8192, (n + threads - 1) / threads);
+  const int64_t nq = (n + 3) / 4;  // one thread per element quad
+  const int blocks =
+      (int)std::min<int64_t>(8192, (nq + threads - 1) / threads);
   const float* scale_ptr =
       scale.defined() && scale.numel() ? scale.data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(
