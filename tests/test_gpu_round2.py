@@ -587,3 +587,18 @@ def test_recommendation_service_gpu(tmp_path):
                 assert len(r["sem_ids"]) == 3
     # repeated batch shapes replay captured graphs (at most 3 graphs)
     assert len(svc._graphed._graphs) <= 3
+
+
+def test_vectorized_kernels_match_scalar():
+    """Vectorized (bf16x2/x4) norm+dropout kernels vs the scalar paths:
+    dropout bitwise-equal, rms within one bf16 ulp. Runs the A/B tool
+    (subprocess per mode — the extension caches the env gate once)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools", "ab_vec_kernels.py")],
+        capture_output=True, text=True, timeout=300, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "AB_VEC PASS" in out.stdout, out.stdout[-2000:]
