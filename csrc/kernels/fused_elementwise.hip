@@ -308,6 +308,29 @@ __global__ void colsum1_kernel(const T* __restrict__ x,
   tmp[(int64_t)chunk * cols + j] = acc;
 }
 
+// bf16x4 stage 1 (column quads, 8-byte loads + float4 partial stores).
+// Still deterministic (fixed-order loops); the adaptive chunk count
+// differs from the scalar path's, so outputs agree within one bf16 ulp
+// rather than bitwise. 2-byte-per-lane loads made the scalar version
+// COBRA's single hottest kernel (25 us x 14 bias grads/step).
+__global__ void colsum1_v4_kernel(const uint2* __restrict__ x,
+                                  float4* __restrict__ tmp,
+                                  int64_t rows, int64_t cols4, int ch) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= (int64_t)ch * cols4) return;
+  int64_t j4 = tid % cols4;
+  int chunk = (int)(tid / cols4);
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int64_t r = chunk; r < rows; r += ch) {
+    BF16x4 v;
+    v.u = x[r * cols4 + j4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) acc[k] += to_f32(v.e[k]);
+  }
+  tmp[(int64_t)chunk * cols4 + j4] =
+      make_float4(acc[0], acc[1], acc[2], acc[3]);
+}
+
 // one 64-lane wave per column: lanes stride the ch partials, fixed-order
 // shuffle tree finishes (deterministic)
 template <typename T>
@@ -326,10 +349,16 @@ __global__ void colsum2_kernel(const float* __restrict__ tmp,
 torch::Tensor colsum(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
   const int64_t rows = x.size(0), cols = x.size(1);
+  const bool vec4 = x.scalar_type() == torch::kBFloat16 && (cols & 3) == 0 &&
+                    ptr_aligned8(x.data_ptr()) && !elemwise_force_scalar();
   // enough chunks to fill the chip regardless of cols (small-col bias
-  // shapes starved at a fixed 32: 46 us at [6400, 256])
+  // shapes starved at a fixed 32: 46 us at [6400, 256]). The vec4 stage-1
+  // kernel has 4x fewer threads per chunk, so it targets the same wave
+  // count via a 4x-smaller per-chunk column count.
+  const int64_t stage1_cols = vec4 ? cols / 4 : cols;
   int ch = 32;
-  while ((int64_t)ch * cols < 131072 && ch < 512 && ch * 4 < rows) ch *= 2;
+  while ((int64_t)ch * stage1_cols < 131072 && ch < 512 && ch * 4 < rows)
+    ch *= 2;
   auto tmp = torch::empty({ch, cols}, x.options().dtype(torch::kFloat32));
   auto out = torch::empty({cols}, x.options());
   dim3 block(256);
@@ -344,10 +373,20 @@ torch::Tensor colsum(torch::Tensor x) {
                        tmp.data_ptr<float>(), out.data_ptr<float>(), cols,
                        ch);
   } else if (x.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL((colsum1_kernel<__hip_bfloat16>), grid1, block, 0,
-                       stream,
-                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-                       tmp.data_ptr<float>(), rows, cols, ch);
+    if (vec4) {
+      const int64_t cols4 = cols >> 2;
+      dim3 g1v((unsigned)(((int64_t)ch * cols4 + 255) / 256));
+      hipLaunchKernelGGL(colsum1_v4_kernel, g1v, block, 0, stream,
+                         reinterpret_cast<const uint2*>(x.data_ptr()),
+                         reinterpret_cast<float4*>(tmp.data_ptr<float>()),
+                         rows, cols4, ch);
+    } else {
+      hipLaunchKernelGGL((colsum1_kernel<__hip_bfloat16>), grid1, block, 0,
+                         stream,
+                         reinterpret_cast<const __hip_bfloat16*>(
+                             x.data_ptr()),
+                         tmp.data_ptr<float>(), rows, cols, ch);
+    }
     hipLaunchKernelGGL((colsum2_kernel<__hip_bfloat16>), grid2, block, 0,
                        stream, tmp.data_ptr<float>(),
                        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),

@@ -39,11 +39,14 @@ wf = torch.randn(768, device=dev, dtype=torch.float32)
 y2, inv2 = ext.rms_norm_fwd(x, wf, 1e-6, False)
 # fp32 weight + bf16 dy: exercises the vec2<float> weight loads
 dx3, dw3 = ext.rms_norm_bwd(dy.to(torch.bfloat16), x, wf, inv2, False)
+# colsum: vec4 stage 1 uses a different adaptive chunk count, so partials
+# regroup -> compare within bf16 tolerance, not bitwise
+cs = ext.colsum(torch.randn(2560, 1536, device=dev, dtype=torch.bfloat16))
 torch.save({"o1": o1.cpu(), "m1": m1.cpu(), "dx1": dx1.cpu(),
             "o2": o2.cpu(), "m2": m2.cpu(), "dx2": dx2.cpu(),
             "y": y.cpu(), "inv": inv.cpu(), "dxx": dxx.cpu(),
             "dw": dw.cpu(), "y2": y2.cpu(), "dx3": dx3.cpu(),
-            "dw3": dw3.cpu()}, sys.argv[1])
+            "dw3": dw3.cpu(), "cs": cs.cpu()}, sys.argv[1])
 """
 
 
@@ -68,7 +71,7 @@ def main():
         if not torch.equal(a[k], b[k]):
             print(f"FAIL {k}: dropout scalar vs vec NOT bitwise equal")
             ok = False
-    for k in ("y", "inv", "dxx", "dw", "y2", "dx3", "dw3"):
+    for k in ("y", "inv", "dxx", "dw", "y2", "dx3", "dw3", "cs"):
         d = (a[k].float() - b[k].float()).abs().max().item()
         ref = a[k].float().abs().max().item() + 1e-6
         if d > 2e-2 * ref:
