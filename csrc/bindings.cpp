@@ -77,6 +77,11 @@ std::vector<torch::Tensor> plain_dropout_fwd(
     c10::optional<torch::Tensor> seed_dev);
 torch::Tensor colsum(torch::Tensor x);
 torch::Tensor chunk_sum(torch::Tensor x);
+std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor w,
+                                          torch::Tensor b, double eps);
+std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor w, torch::Tensor mean,
+                                          torch::Tensor rstd);
 torch::Tensor dropout_fuse_bwd(torch::Tensor dy, torch::Tensor mask,
                                double p, bool relu);
 torch::Tensor topk_hit_ranks(torch::Tensor actual, torch::Tensor topk);
@@ -138,6 +143,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "deterministic replay-safe column sum (bias grads)");
   m.def("chunk_sum", &genrec::chunk_sum,
         "small-row-count column sum (split-K dW partials)");
+  m.def("layer_norm_fwd", &genrec::layer_norm_fwd, "fused LayerNorm fwd");
+  m.def("layer_norm_bwd", &genrec::layer_norm_bwd, "fused LayerNorm bwd");
   m.def("dropout_fuse_bwd", &genrec::dropout_fuse_bwd, "fused dropout bwd");
   m.def("topk_hit_ranks", &genrec::topk_hit_ranks, "first-match ranks");
   m.def("fused_adamw", &genrec::fused_adamw,

@@ -275,6 +275,224 @@ static inline bool ptr_aligned4(const void* p) {
   return (reinterpret_cast<uintptr_t>(p) & 3) == 0;
 }
 
+// -------------------------------------------------------------- LayerNorm
+// nn.LayerNorm semantics (elementwise affine, fp32 stats like ATen's
+// vectorized_layer_norm_kernel<BFloat16, float>): ATen's backward pair
+// (cuComputeGradInput 68 us + cuComputePartGradGammaBeta) was ~7% of the
+// COBRA step. Same wave-per-row structure as the RMSNorm kernels; dw/db
+// partials go through the adaptive two-stage reduce ([n_blocks, 2d]).
+
+static int grid_for_rows(int64_t n_rows, int waves_per_block);
+__global__ void rms_dw_reduce1_kernel(const float* __restrict__ part,
+                                      float* __restrict__ tmp, int n_part,
+                                      int d, int ch);
+template <typename WT>
+__global__ void rms_dw_reduce2_kernel(const float* __restrict__ tmp,
+                                      WT* __restrict__ out, int d, int ch);
+
+template <typename T, typename WT>
+__global__ void layer_norm_fwd_kernel(const T* __restrict__ x,
+                                      const WT* __restrict__ w,
+                                      const WT* __restrict__ b,
+                                      T* __restrict__ y,
+                                      float* __restrict__ mean_arr,
+                                      float* __restrict__ rstd_arr,
+                                      int64_t n_rows, int d, float eps) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = gridDim.x * blockDim.x / WAVE;
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const T* xr = x + row * d;
+    float s = 0.f, ss = 0.f;
+    for (int j = lane; j < d; j += WAVE) {
+      float v = to_f32(xr[j]);
+      s += v;
+      ss += v * v;
+    }
+    s = wave_sum(s);
+    ss = wave_sum(ss);
+    float mean = s / d;
+    float var = ss / d - mean * mean;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (lane == 0) {
+      mean_arr[row] = mean;
+      rstd_arr[row] = rstd;
+    }
+    T* yr = y + row * d;
+    for (int j = lane; j < d; j += WAVE) {
+      float xhat = (to_f32(xr[j]) - mean) * rstd;
+      yr[j] = from_f32<T>(xhat * to_f32(w[j]) + to_f32(b[j]));
+    }
+  }
+}
+
+template <typename T, typename WT>
+__global__ void layer_norm_bwd_kernel(const T* __restrict__ dy,
+                                      const T* __restrict__ x,
+                                      const WT* __restrict__ w,
+                                      const float* __restrict__ mean_arr,
+                                      const float* __restrict__ rstd_arr,
+                                      T* __restrict__ dx,
+                                      float* __restrict__ dwdb,
+                                      int64_t n_rows, int d) {
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = (int64_t)blockIdx.x * waves_per_block + wave_in_block;
+  const int64_t n_waves = (int64_t)gridDim.x * waves_per_block;
+
+  float dw_acc[RMS_MAX_COLS_PER_LANE];
+  float db_acc[RMS_MAX_COLS_PER_LANE];
+#pragma unroll
+  for (int c = 0; c < RMS_MAX_COLS_PER_LANE; ++c) dw_acc[c] = db_acc[c] = 0.f;
+
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const T* xr = x + row * d;
+    const T* dyr = dy + row * d;
+    float mean = mean_arr[row];
+    float rstd = rstd_arr[row];
+    float c1 = 0.f, c2 = 0.f;
+    for (int j = lane; j < d; j += WAVE) {
+      float g = to_f32(dyr[j]) * to_f32(w[j]);
+      float xhat = (to_f32(xr[j]) - mean) * rstd;
+      c1 += g;
+      c2 += g * xhat;
+    }
+    c1 = wave_sum(c1) / d;
+    c2 = wave_sum(c2) / d;
+    T* dxr = dx + row * d;
+    int ci = 0;
+    for (int j = lane; j < d; j += WAVE, ++ci) {
+      float dyf = to_f32(dyr[j]);
+      float g = dyf * to_f32(w[j]);
+      float xhat = (to_f32(xr[j]) - mean) * rstd;
+      dxr[j] = from_f32<T>(rstd * (g - c1 - xhat * c2));
+      dw_acc[ci] += dyf * xhat;
+      db_acc[ci] += dyf;
+    }
+  }
+  // one [2, d] partial row per block (dw then db), LDS block-reduced
+  extern __shared__ __attribute__((aligned(16))) float ln_lds[];  // [4][2d]
+  int ci = 0;
+  for (int j = lane; j < d; j += WAVE, ++ci) {
+    ln_lds[wave_in_block * 2 * d + j] = dw_acc[ci];
+    ln_lds[wave_in_block * 2 * d + d + j] = db_acc[ci];
+  }
+  __syncthreads();
+  for (int j = (int)threadIdx.x; j < 2 * d; j += (int)blockDim.x) {
+    float sum = ln_lds[j] + ln_lds[2 * d + j] + ln_lds[4 * d + j] +
+                ln_lds[6 * d + j];
+    dwdb[(int64_t)blockIdx.x * 2 * d + j] = sum;
+  }
+}
+
+std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor w,
+                                          torch::Tensor b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int d = x.size(-1);
+  const int64_t n_rows = x.numel() / d;
+  TORCH_CHECK(d <= WAVE * 16, "layer_norm: d too large");
+  auto wc = w.contiguous();
+  auto bc = b.contiguous();
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              b.scalar_type() == x.scalar_type(),
+              "layer_norm: dtype mismatch");
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({n_rows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({n_rows}, x.options().dtype(torch::kFloat32));
+  dim3 block(256);
+  dim3 grid(grid_for_rows(n_rows, 4));
+  auto stream = at::cuda::getCurrentHIPStream();
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((layer_norm_fwd_kernel<__hip_bfloat16, __hip_bfloat16>),
+                       grid, block, 0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(wc.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(bc.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       n_rows, d, (float)eps);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((layer_norm_fwd_kernel<float, float>), grid, block, 0,
+                       stream, x.data_ptr<float>(), wc.data_ptr<float>(),
+                       bc.data_ptr<float>(), y.data_ptr<float>(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       n_rows, d, (float)eps);
+  } else {
+    TORCH_CHECK(false, "layer_norm: unsupported dtype");
+  }
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor w, torch::Tensor mean,
+                                          torch::Tensor rstd) {
+  const int d = x.size(-1);
+  const int64_t n_rows = x.numel() / d;
+  TORCH_CHECK(d <= WAVE * 16, "layer_norm_bwd: d too large");
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(dy.scalar_type() == x.scalar_type() &&
+              w.scalar_type() == x.scalar_type());
+  auto wc = w.contiguous();
+  auto dx = torch::empty_like(x);
+  dim3 block(256);
+  static const int cap = [] {
+    const char* e = getenv("GENREC_RMS_BWD_CAP");
+    int v = e ? atoi(e) : 1024;
+    return (v >= 64 && v <= 4096) ? v : 1024;
+  }();
+  int n_blocks = std::min(grid_for_rows(n_rows, 4), cap);
+  dim3 grid(n_blocks);
+  auto dwdb = torch::empty({n_blocks, 2 * (int64_t)d},
+                           x.options().dtype(torch::kFloat32));
+  size_t smem = 8 * (size_t)d * sizeof(float);
+  auto stream = at::cuda::getCurrentHIPStream();
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((layer_norm_bwd_kernel<__hip_bfloat16, __hip_bfloat16>),
+                       grid, block, smem, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(wc.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()),
+                       dwdb.data_ptr<float>(), n_rows, d);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((layer_norm_bwd_kernel<float, float>), grid, block,
+                       smem, stream, dy.data_ptr<float>(),
+                       x.data_ptr<float>(), wc.data_ptr<float>(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       dx.data_ptr<float>(), dwdb.data_ptr<float>(), n_rows,
+                       d);
+  } else {
+    TORCH_CHECK(false, "layer_norm_bwd: unsupported dtype");
+  }
+  // adaptive two-stage column reduce over the [n_blocks, 2d] partials
+  auto dwdb_out = torch::empty({2 * (int64_t)d}, w.options());
+  const int d2x = 2 * d;
+  int ch = 32;
+  while ((int64_t)ch * d2x < 131072 && ch < 512 && ch * 4 < n_blocks) ch *= 2;
+  auto tmp = torch::empty({ch, (int64_t)d2x},
+                          x.options().dtype(torch::kFloat32));
+  dim3 rblock(256);
+  dim3 rgrid1((unsigned)(((int64_t)ch * d2x + 255) / 256));
+  hipLaunchKernelGGL(rms_dw_reduce1_kernel, rgrid1, rblock, 0, stream,
+                     dwdb.data_ptr<float>(), tmp.data_ptr<float>(), n_blocks,
+                     d2x, ch);
+  dim3 rgrid2((unsigned)((d2x + 3) / 4));
+  if (w.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((rms_dw_reduce2_kernel<__hip_bfloat16>), rgrid2,
+                       rblock, 0, stream, tmp.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(dwdb_out.data_ptr()),
+                       d2x, ch);
+  } else {
+    hipLaunchKernelGGL((rms_dw_reduce2_kernel<float>), rgrid2, rblock, 0,
+                       stream, tmp.data_ptr<float>(),
+                       dwdb_out.data_ptr<float>(), d2x, ch);
+  }
+  auto parts = dwdb_out.split(d, 0);
+  return {dx, parts[0], parts[1]};
+}
+
 // ---------------------------------------------------------------- L2Norm
 
 template <typename T>

@@ -39,6 +39,7 @@ from genrec_amd.ops.linear import SplitKLinear
 
 from genrec_amd.config import ginlite
 from genrec_amd.modules.encoders import LightT5Encoder
+from genrec_amd.modules.norms import FusedLayerNorm
 
 
 class CobraOutput(NamedTuple):
@@ -157,9 +158,9 @@ class CobraDecoderLayer(nn.Module):
         self.out = SplitKLinear(d_model, d_model, bias=True)
         self.linear1 = SplitKLinear(d_model, ff_dim, bias=True)
         self.linear2 = SplitKLinear(ff_dim, d_model, bias=True)
-        self.norm1 = nn.LayerNorm(d_model)
-        self.norm2 = nn.LayerNorm(d_model)
-        self.norm3 = nn.LayerNorm(d_model)
+        self.norm1 = FusedLayerNorm(d_model)
+        self.norm2 = FusedLayerNorm(d_model)
+        self.norm3 = FusedLayerNorm(d_model)
         self.dropout_p = dropout
 
     def forward(self, x: Tensor, key_pad_mask: Optional[Tensor],

@@ -20,6 +20,7 @@ import torch.nn.functional as F
 from torch import Tensor, nn
 
 from genrec_amd.ops.linear import SplitKLinear
+from genrec_amd.modules.norms import FusedLayerNorm
 
 
 class CaptureSafeEncoderLayer(nn.Module):
@@ -43,8 +44,8 @@ class CaptureSafeEncoderLayer(nn.Module):
         self.out = SplitKLinear(d_model, d_model)
         self.linear1 = SplitKLinear(d_model, ff_dim)
         self.linear2 = SplitKLinear(ff_dim, d_model)
-        self.norm1 = nn.LayerNorm(d_model)
-        self.norm2 = nn.LayerNorm(d_model)
+        self.norm1 = FusedLayerNorm(d_model)
+        self.norm2 = FusedLayerNorm(d_model)
         self.dropout_p = dropout
 
     def forward(self, x: Tensor,
@@ -102,7 +103,7 @@ class LightT5Encoder(nn.Module):
             CaptureSafeEncoderLayer(hidden_dim, num_heads, ff_dim, dropout)
             for _ in range(n_layers)])
         self.proj = SplitKLinear(hidden_dim, output_dim)
-        self.layer_norm = nn.LayerNorm(hidden_dim)
+        self.layer_norm = FusedLayerNorm(hidden_dim)
 
     def forward(self, batch_tokens: Tensor) -> Tensor:
         """batch_tokens: (B, T, L) or (B, L); 0 = pad. Returns L2-normalized

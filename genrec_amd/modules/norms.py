@@ -79,3 +79,12 @@ class SwishLayerNorm(nn.Module):
 
     def forward(self, x: Tensor) -> Tensor:
         return ops.swish_layer_norm(x, self.ln.weight, self.ln.bias, self.ln.eps)
+
+
+class FusedLayerNorm(nn.LayerNorm):
+    """nn.LayerNorm whose forward dispatches to the genrec LayerNorm
+    kernels on GPU (eager F.layer_norm elsewhere). State-dict compatible
+    with nn.LayerNorm (same parameter names), normalizes the last dim."""
+
+    def forward(self, x: Tensor) -> Tensor:
+        return ops.layer_norm(x, self.weight, self.bias, self.eps)

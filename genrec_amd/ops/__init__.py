@@ -75,7 +75,9 @@ def use_hip(*tensors: torch.Tensor) -> bool:
 
 
 from genrec_amd.ops import eager  # noqa: E402
-from genrec_amd.ops.norms import rms_norm, l2norm_op, swish_layer_norm  # noqa: E402
+from genrec_amd.ops.norms import (  # noqa: E402
+    layer_norm, rms_norm, l2norm_op, swish_layer_norm,
+)
 from genrec_amd.ops.attention import (  # noqa: E402
     fused_attention,
     sasrec_attention,
@@ -97,6 +99,7 @@ __all__ = [
     "rms_norm",
     "l2norm_op",
     "swish_layer_norm",
+    "layer_norm",
     "fused_attention",
     "sasrec_attention",
     "t5_attention",

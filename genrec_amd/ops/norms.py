@@ -80,3 +80,39 @@ def swish_layer_norm(x: Tensor, weight: Tensor, bias: Tensor, eps: float) -> Ten
     # Fused silu(LayerNorm(x)). HIP kernel pending; composed form is used on
     # GPU meanwhile (LayerNorm + silu are both ATen/MIOpen kernels).
     return eager.swish_layer_norm(x, weight, bias, eps)
+
+
+class _LayerNormFn(torch.autograd.Function):
+    """nn.LayerNorm (elementwise affine) on the genrec kernel: ATen's
+    backward pair (cuComputeGradInput + cuComputePartGradGammaBeta) was
+    ~7% of the COBRA step at [2560, 384]-class shapes."""
+
+    @staticmethod
+    def forward(ctx, x: Tensor, weight: Tensor, bias: Tensor, eps: float):
+        from genrec_amd import ops
+
+        x2 = x.contiguous()
+        y, mean, rstd = ops.ext().layer_norm_fwd(x2, weight, bias, eps)
+        ctx.save_for_backward(x2, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        from genrec_amd import ops
+
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ops.ext().layer_norm_bwd(
+            dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, db, None
+
+
+def layer_norm(x: Tensor, weight: Tensor, bias: Tensor,
+               eps: float = 1e-5) -> Tensor:
+    """Drop-in F.layer_norm over the last dim (affine required)."""
+    if (_hip_ok(x) and weight is not None and bias is not None
+            and x.dtype == weight.dtype and x.dtype == bias.dtype
+            and x.dtype in (torch.bfloat16, torch.float32)
+            and x.shape[-1] <= 1024):
+        return _LayerNormFn.apply(x, weight, bias, eps)
+    return torch.nn.functional.layer_norm(
+        x, (x.shape[-1],), weight, bias, eps)

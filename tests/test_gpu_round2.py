@@ -602,3 +602,35 @@ def test_vectorized_kernels_match_scalar():
         capture_output=True, text=True, timeout=300, cwd=repo)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "AB_VEC PASS" in out.stdout, out.stdout[-2000:]
+
+
+def test_layer_norm_kernel_parity():
+    """genrec LayerNorm fwd/bwd vs F.layer_norm in fp32 (dx, dw, db)."""
+    from genrec_amd import ops
+
+    torch.manual_seed(5)
+    for rows, d, dt in [(2560, 384, torch.bfloat16),
+                        (1000, 768, torch.bfloat16),
+                        (511, 640, torch.float32),
+                        (3, 64, torch.bfloat16)]:
+        x = torch.randn(rows, d, device=DEV, dtype=dt, requires_grad=True)
+        w = torch.randn(d, device=DEV, dtype=dt, requires_grad=True)
+        b = torch.randn(d, device=DEV, dtype=dt, requires_grad=True)
+        y = ops.layer_norm(x, w, b, 1e-5)
+        assert "LayerNormFn" in type(y.grad_fn).__name__
+        dy = torch.randn_like(y)
+        y.backward(dy)
+
+        xf = x.detach().float().requires_grad_(True)
+        wf = w.detach().float().requires_grad_(True)
+        bf = b.detach().float().requires_grad_(True)
+        yf = torch.nn.functional.layer_norm(xf, (d,), wf, bf, 1e-5)
+        yf.backward(dy.float())
+
+        tol = 3e-2 if dt == torch.bfloat16 else 1e-4
+        for got, ref, name in [(y, yf, "y"), (x.grad, xf.grad, "dx"),
+                               (w.grad, wf.grad, "dw"),
+                               (b.grad, bf.grad, "db")]:
+            scale = ref.abs().max().item() + 1e-3
+            err = (got.float() - ref).abs().max().item() / scale
+            assert err < tol, (name, rows, d, dt, err)
