@@ -23,6 +23,7 @@ import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
 
+from genrec_amd.modules.norms import FusedLayerNorm
 from genrec_amd import ops
 from genrec_amd.ops.linear import SplitKLinear
 from genrec_amd.config import ginlite
@@ -101,13 +102,13 @@ class HSTULayer(nn.Module):
         if use_temporal_bias:
             self.temporal_bias = TemporalBias(num_buckets=num_time_buckets,
                                               num_heads=num_heads)
-        self.attn_norm = nn.LayerNorm(embed_dim)
+        self.attn_norm = FusedLayerNorm(embed_dim)
         self.ffn = nn.Sequential(
             SplitKLinear(embed_dim, 4 * embed_dim), nn.SiLU(),
             nn.Dropout(dropout), SplitKLinear(4 * embed_dim, embed_dim),
             nn.Dropout(dropout),
         )
-        self.ffn_norm = nn.LayerNorm(embed_dim)
+        self.ffn_norm = FusedLayerNorm(embed_dim)
         self.dropout = nn.Dropout(dropout)
 
     def forward(self, x: Tensor, padding_mask: Tensor,
@@ -168,7 +169,7 @@ class HSTU(nn.Module):
                       use_temporal_bias=use_temporal_bias)
             for _ in range(num_blocks)
         ])
-        self.final_norm = nn.LayerNorm(embed_dim)
+        self.final_norm = FusedLayerNorm(embed_dim)
         self._init_weights()
 
     def _init_weights(self) -> None:

@@ -25,6 +25,7 @@ import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
 
+from genrec_amd.modules.norms import FusedLayerNorm
 from genrec_amd import ops
 from genrec_amd.ops.linear import SplitKLinear
 from genrec_amd.config import ginlite
@@ -86,8 +87,8 @@ class SASRecBlock(nn.Module):
         super().__init__()
         self.attention = MultiHeadAttention(embed_dim, num_heads, dropout)
         self.ffn = PointWiseFeedForward(embed_dim, ffn_dim, dropout)
-        self.norm1 = nn.LayerNorm(embed_dim, eps=1e-8)
-        self.norm2 = nn.LayerNorm(embed_dim, eps=1e-8)
+        self.norm1 = FusedLayerNorm(embed_dim, eps=1e-8)
+        self.norm2 = FusedLayerNorm(embed_dim, eps=1e-8)
 
     def forward(self, x: Tensor, mask: Tensor) -> Tensor:
         x = self.attention(self.norm1(x), x, mask)
@@ -116,7 +117,7 @@ class SASRec(nn.Module):
             SASRecBlock(embed_dim, num_heads, ffn_dim, dropout)
             for _ in range(num_blocks)
         ])
-        self.final_norm = nn.LayerNorm(embed_dim, eps=1e-8)
+        self.final_norm = FusedLayerNorm(embed_dim, eps=1e-8)
         self._init_weights()
 
     def _init_weights(self) -> None:
