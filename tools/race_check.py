@@ -118,6 +118,20 @@ def build_cases():
     def ce():
         return ops.ext().softmax_ce_fwd(logits, targets, -100)[:2]
 
+    b = torch.randn(384, device=dev, dtype=torch.bfloat16)
+
+    def ln_fwd():
+        return ops.ext().layer_norm_fwd(x, w, b, 1e-5)
+
+    lf = ops.ext().layer_norm_fwd(x, w, b, 1e-5)
+    dy_ln = torch.randn_like(x)
+
+    def ln_bwd():
+        return ops.ext().layer_norm_bwd(dy_ln, x, w, lf[1], lf[2])
+
+    def csum():
+        return ops.ext().colsum(x)
+
     return [
         ("attn_fwd_mfma", attn_fwd, 0.0),
         ("attn_bwd_ds(8-wave)", attn_bwd, 0.0),
@@ -126,6 +140,9 @@ def build_cases():
         # dependent rounding; tolerance instead of bitwise
         ("attn_bwd_flash", flash_bwd, 2e-3),
         ("rms_norm", rms, 0.0),
+        ("layer_norm_fwd", ln_fwd, 0.0),
+        ("layer_norm_bwd", ln_bwd, 0.0),
+        ("colsum(vec4)", csum, 0.0),
         ("softmax_ce", ce, 0.0),
     ]
 
