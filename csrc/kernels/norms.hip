@@ -386,6 +386,131 @@ __global__ void layer_norm_bwd_kernel(const T* __restrict__ dy,
   }
 }
 
+// bf16x2 variants (column-pair dwords, same math/rounding as scalar —
+// only the per-lane fp32 accumulation order changes, as for RMSNorm)
+__global__ void layer_norm_fwd_v2_kernel(const unsigned int* __restrict__ x,
+                                         const unsigned int* __restrict__ w,
+                                         const unsigned int* __restrict__ b,
+                                         unsigned int* __restrict__ y,
+                                         float* __restrict__ mean_arr,
+                                         float* __restrict__ rstd_arr,
+                                         int64_t n_rows, int d2, float eps) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = gridDim.x * blockDim.x / WAVE;
+  const int d = d2 * 2;
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const unsigned int* xr = x + row * d2;
+    float s = 0.f, ss = 0.f;
+    for (int j = lane; j < d2; j += WAVE) {
+      BF16x2 v;
+      v.u = xr[j];
+      float a = to_f32(v.e[0]), c = to_f32(v.e[1]);
+      s += a + c;
+      ss += a * a + c * c;
+    }
+    s = wave_sum(s);
+    ss = wave_sum(ss);
+    float mean = s / d;
+    float var = ss / d - mean * mean;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (lane == 0) {
+      mean_arr[row] = mean;
+      rstd_arr[row] = rstd;
+    }
+    unsigned int* yr = y + row * d2;
+    for (int j = lane; j < d2; j += WAVE) {
+      BF16x2 v, wv, bv, o;
+      v.u = xr[j];
+      wv.u = w[j];
+      bv.u = b[j];
+#pragma unroll
+      for (int k = 0; k < 2; ++k) {
+        float xhat = (to_f32(v.e[k]) - mean) * rstd;
+        o.e[k] = from_f32<__hip_bfloat16>(
+            xhat * to_f32(wv.e[k]) + to_f32(bv.e[k]));
+      }
+      yr[j] = o.u;
+    }
+  }
+}
+
+__global__ void layer_norm_bwd_v2_kernel(const unsigned int* __restrict__ dy,
+                                         const unsigned int* __restrict__ x,
+                                         const unsigned int* __restrict__ w,
+                                         const float* __restrict__ mean_arr,
+                                         const float* __restrict__ rstd_arr,
+                                         unsigned int* __restrict__ dx,
+                                         float* __restrict__ dwdb,
+                                         int64_t n_rows, int d2) {
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = (int64_t)blockIdx.x * waves_per_block + wave_in_block;
+  const int64_t n_waves = (int64_t)gridDim.x * waves_per_block;
+  const int d = d2 * 2;
+
+  float dw_acc[RMS_MAX_COLS_PER_LANE];
+  float db_acc[RMS_MAX_COLS_PER_LANE];
+#pragma unroll
+  for (int c = 0; c < RMS_MAX_COLS_PER_LANE; ++c) dw_acc[c] = db_acc[c] = 0.f;
+
+  for (int64_t row = wave_id; row < n_rows; row += n_waves) {
+    const unsigned int* xr = x + row * d2;
+    const unsigned int* dyr = dy + row * d2;
+    float mean = mean_arr[row];
+    float rstd = rstd_arr[row];
+    float c1 = 0.f, c2 = 0.f;
+    for (int j = lane; j < d2; j += WAVE) {
+      BF16x2 xv, gv, wv;
+      xv.u = xr[j];
+      gv.u = dyr[j];
+      wv.u = w[j];
+#pragma unroll
+      for (int k = 0; k < 2; ++k) {
+        float g = to_f32(gv.e[k]) * to_f32(wv.e[k]);
+        float xhat = (to_f32(xv.e[k]) - mean) * rstd;
+        c1 += g;
+        c2 += g * xhat;
+      }
+    }
+    c1 = wave_sum(c1) / d;
+    c2 = wave_sum(c2) / d;
+    unsigned int* dxr = dx + row * d2;
+    int ci = 0;
+    for (int j = lane; j < d2; j += WAVE, ci += 2) {
+      BF16x2 xv, gv, wv, o;
+      xv.u = xr[j];
+      gv.u = dyr[j];
+      wv.u = w[j];
+#pragma unroll
+      for (int k = 0; k < 2; ++k) {
+        float dyf = to_f32(gv.e[k]);
+        float g = dyf * to_f32(wv.e[k]);
+        float xhat = (to_f32(xv.e[k]) - mean) * rstd;
+        o.e[k] = from_f32<__hip_bfloat16>(rstd * (g - c1 - xhat * c2));
+        dw_acc[ci + k] += dyf * xhat;
+        db_acc[ci + k] += dyf;
+      }
+      dxr[j] = o.u;
+    }
+  }
+  extern __shared__ __attribute__((aligned(16))) float ln_lds[];  // [4][2d]
+  int ci = 0;
+  for (int j = lane; j < d2; j += WAVE, ci += 2) {
+    ln_lds[wave_in_block * 2 * d + 2 * j] = dw_acc[ci];
+    ln_lds[wave_in_block * 2 * d + 2 * j + 1] = dw_acc[ci + 1];
+    ln_lds[wave_in_block * 2 * d + d + 2 * j] = db_acc[ci];
+    ln_lds[wave_in_block * 2 * d + d + 2 * j + 1] = db_acc[ci + 1];
+  }
+  __syncthreads();
+  for (int j = (int)threadIdx.x; j < 2 * d; j += (int)blockDim.x) {
+    float sum = ln_lds[j] + ln_lds[2 * d + j] + ln_lds[4 * d + j] +
+                ln_lds[6 * d + j];
+    dwdb[(int64_t)blockIdx.x * 2 * d + j] = sum;
+  }
+}
+
 std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor w,
                                           torch::Tensor b, double eps) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
@@ -404,6 +529,18 @@ std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor w,
   dim3 grid(grid_for_rows(n_rows, 4));
   auto stream = at::cuda::getCurrentHIPStream();
   if (x.scalar_type() == torch::kBFloat16) {
+    if ((d & 1) == 0 && ptr_aligned4(x.data_ptr()) &&
+        ptr_aligned4(wc.data_ptr()) && ptr_aligned4(bc.data_ptr()) &&
+        !norms_force_scalar()) {
+      hipLaunchKernelGGL(layer_norm_fwd_v2_kernel, grid, block, 0, stream,
+                         reinterpret_cast<const unsigned int*>(x.data_ptr()),
+                         reinterpret_cast<const unsigned int*>(wc.data_ptr()),
+                         reinterpret_cast<const unsigned int*>(bc.data_ptr()),
+                         reinterpret_cast<unsigned int*>(y.data_ptr()),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         n_rows, d / 2, (float)eps);
+      return {y, mean, rstd};
+    }
     hipLaunchKernelGGL((layer_norm_fwd_kernel<__hip_bfloat16, __hip_bfloat16>),
                        grid, block, 0, stream,
                        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
@@ -447,7 +584,17 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
                            x.options().dtype(torch::kFloat32));
   size_t smem = 8 * (size_t)d * sizeof(float);
   auto stream = at::cuda::getCurrentHIPStream();
-  if (x.scalar_type() == torch::kBFloat16) {
+  if (x.scalar_type() == torch::kBFloat16 && (d & 1) == 0 &&
+      ptr_aligned4(dy.data_ptr()) && ptr_aligned4(x.data_ptr()) &&
+      ptr_aligned4(wc.data_ptr()) && !norms_force_scalar()) {
+    hipLaunchKernelGGL(layer_norm_bwd_v2_kernel, grid, block, smem, stream,
+                       reinterpret_cast<const unsigned int*>(dy.data_ptr()),
+                       reinterpret_cast<const unsigned int*>(x.data_ptr()),
+                       reinterpret_cast<const unsigned int*>(wc.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       reinterpret_cast<unsigned int*>(dx.data_ptr()),
+                       dwdb.data_ptr<float>(), n_rows, d / 2);
+  } else if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL((layer_norm_bwd_kernel<__hip_bfloat16, __hip_bfloat16>),
                        grid, block, smem, stream,
                        reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
