@@ -361,3 +361,29 @@ def test_cosine_warmup_schedule_matches_closed_form():
         assert abs(lr - expect) < 1e-9, (step, lr, expect)
         opt.step()
         sched.step()
+
+
+def test_fused_layer_norm_cpu_fallback_and_state_dict():
+    """FusedLayerNorm == nn.LayerNorm on CPU (exact F.layer_norm
+    fallback) and state-dict compatible in both directions."""
+    import torch
+    from torch import nn
+
+    from genrec_amd.modules.norms import FusedLayerNorm
+
+    torch.manual_seed(0)
+    ref = nn.LayerNorm(48, eps=1e-8)
+    fused = FusedLayerNorm(48, eps=1e-8)
+    fused.load_state_dict(ref.state_dict())
+    x = torch.randn(7, 48)
+    assert torch.equal(fused(x), ref(x))
+    # grads flow identically
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    fused(x1).sum().backward()
+    ref(x2).sum().backward()
+    assert torch.equal(x1.grad, x2.grad)
+    # round-trip the other way
+    ref2 = nn.LayerNorm(48, eps=1e-8)
+    ref2.load_state_dict(fused.state_dict())
+    assert torch.equal(ref2.weight, fused.weight)
