@@ -7,9 +7,9 @@
 // the identity dot_i = rowsum(dO * O)) is validated at fragment
 // granularity against autograd in tools/sim_flash_tiles.py (~1e-15).
 //
-// STAGED FOR ROUND 2: compiled and bound, but dispatched only with
-// GENREC_ATTN_FLASH=1 (ops/attention.py) until GPU-validated — the
-// <=64-tile kernels in attention_mfma.hip remain the production path.
+// DEFAULT for Lk>64 since round 2 (GPU-validated; routes the COBRA
+// decoder). Stride-aware on the input side: q/k/v (+dout in backward)
+// may be [B,L,H,D]-style transpose views with d innermost.
 //
 // Backward saves scores S (post-mask) + per-row (m, l) instead of
 // normalized P; P is reconstructed as exp(S - m)/l in the epilogue.
@@ -71,7 +71,10 @@ attn_fwd_flash_kernel(
     const unsigned int* __restrict__ seed_dev,
     int B, int H, int Lq, int Lk, int D,
     float scale, int bias_dim, bool bias_bf16, bool causal,
-    float dropout_p, unsigned int seed) {
+    float dropout_p, unsigned int seed,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl,
+    int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl) {
   if (seed_dev) seed += *seed_dev;
   const float* bias_f = reinterpret_cast<const float*>(bias);
   const __hip_bfloat16* bias_b = reinterpret_cast<const __hip_bfloat16*>(bias);
@@ -101,7 +104,7 @@ attn_fwd_flash_kernel(
     int qi = q0 + row;
     if (qi < Lq && d0 < D) {
       val = *reinterpret_cast<const short8f*>(
-          &q[FIDX(b, h, qi, d0, H, Lq, D)]);
+          &q[(int64_t)b * q_sb + h * q_sh + qi * q_sl + d0]);
     }
     *reinterpret_cast<short8f*>(qs + fswz(row, d0 * 2)) = val;
   }
@@ -121,13 +124,13 @@ attn_fwd_flash_kernel(
       short8f val = {};
       if (kj < Lk && d0 < D) {
         val = *reinterpret_cast<const short8f*>(
-            &k[FIDX(b, h, kj, d0, H, Lk, D)]);
+            &k[(int64_t)b * k_sb + h * k_sh + kj * k_sl + d0]);
       }
       *reinterpret_cast<short8f*>(ks + fswz(row, d0 * 2)) = val;
       short8f vv = {};
       if (kj < Lk && d0 < D) {
         vv = *reinterpret_cast<const short8f*>(
-            &v[FIDX(b, h, kj, d0, H, Lk, D)]);
+            &v[(int64_t)b * v_sb + h * v_sh + kj * v_sl + d0]);
       }
       const int g = (lane >> 3) & 7;
       short tv[8];
@@ -287,7 +290,11 @@ attn_bwd_flash_kernel(
     __hip_bfloat16* __restrict__ dv_out,
     float* __restrict__ ds_saved,             // null | [B,H,Lq,Lk]
     int B, int H, int Lq, int Lk, int D,
-    float scale, float dropout_p) {
+    float scale, float dropout_p,
+    int64_t do_sb, int64_t do_sh, int64_t do_sl,
+    int64_t q_sb, int64_t q_sh, int64_t q_sl,
+    int64_t k_sb, int64_t k_sh, int64_t k_sl,
+    int64_t v_sb, int64_t v_sh, int64_t v_sl) {
   const int bh = blockIdx.x;
   const int b = bh / H, h = bh % H;
   const int kt0 = blockIdx.y * FTILE;
@@ -321,9 +328,9 @@ attn_bwd_flash_kernel(
     short8f kk8 = {};
     if (kj < Lk && d0 < D) {
       val = *reinterpret_cast<const short8f*>(
-          &v[FIDX(b, h, kj, d0, H, Lk, D)]);
+          &v[(int64_t)b * v_sb + h * v_sh + kj * v_sl + d0]);
       kk8 = *reinterpret_cast<const short8f*>(
-          &k[FIDX(b, h, kj, d0, H, Lk, D)]);
+          &k[(int64_t)b * k_sb + h * k_sh + kj * k_sl + d0]);
     }
     *reinterpret_cast<short8f*>(vs + fswz(row, d0 * 2)) = val;
     const int g = (lane >> 3) & 7;
@@ -351,9 +358,9 @@ attn_bwd_flash_kernel(
       short8f dd8 = {}, qq8 = {};
       if (qi < Lq && d0 < D) {
         dd8 = *reinterpret_cast<const short8f*>(
-            &dout[FIDX(b, h, qi, d0, H, Lq, D)]);
+            &dout[(int64_t)b * do_sb + h * do_sh + qi * do_sl + d0]);
         qq8 = *reinterpret_cast<const short8f*>(
-            &q[FIDX(b, h, qi, d0, H, Lq, D)]);
+            &q[(int64_t)b * q_sb + h * q_sh + qi * q_sl + d0]);
       }
       *reinterpret_cast<short8f*>(dos + fswz(row, d0 * 2)) = dd8;
       const int g = (lane >> 3) & 7;
@@ -488,12 +495,22 @@ std::vector<torch::Tensor> attn_fwd_flash(
     c10::optional<torch::Tensor> query_mask,
     double scale, bool causal, double dropout_p, int64_t seed,
     c10::optional<torch::Tensor> seed_dev) {
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 && q.is_contiguous() &&
-              k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
   TORCH_CHECK(D <= FTILE && D % 32 == 0);
-  auto out = torch::empty_like(q);
+  // stride-aware over [B,H,L,D] views (d must be innermost; 16-byte
+  // short8 loads need 8-element-aligned batch/head/row strides — true
+  // for every [B,L,H,D] transpose view at D % 32 == 0). Anything else
+  // is copied here, not in the op layer.
+  auto ok_strides = [](const torch::Tensor& t) {
+    return t.stride(3) == 1 && t.stride(0) % 8 == 0 &&
+           t.stride(1) % 8 == 0 && t.stride(2) % 8 == 0;
+  };
+  torch::Tensor qn = ok_strides(q) ? q : q.contiguous();
+  torch::Tensor kn = ok_strides(k) ? k : k.contiguous();
+  torch::Tensor vn = ok_strides(v) ? v : v.contiguous();
+  auto out = torch::empty({B, H, Lq, D}, q.options());
   auto opts_f = q.options().dtype(torch::kFloat32);
   auto s_saved = torch::empty({B, H, Lq, Lk}, opts_f);
   auto ml = torch::empty({B, H, Lq, 2}, opts_f);
@@ -520,9 +537,9 @@ std::vector<torch::Tensor> attn_fwd_flash(
   size_t smem = 4 * FTILE * 128;
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(attn_fwd_flash_kernel, grid, block, smem, stream,
-      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(qn.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(kn.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(vn.data_ptr()),
       bias_dim ? bias_c.data_ptr() : nullptr,
       key_pad.has_value() ? key_pad->data_ptr<bool>() : nullptr,
       add_mask.has_value() ? am_f.data_ptr<float>() : nullptr,
@@ -534,7 +551,10 @@ std::vector<torch::Tensor> attn_fwd_flash(
           ? reinterpret_cast<const unsigned int*>(seed_dev->data_ptr())
           : nullptr,
       B, H, Lq, Lk, D, (float)scale, bias_dim, bias_bf16, causal,
-      (float)dropout_p, (unsigned int)seed);
+      (float)dropout_p, (unsigned int)seed,
+      qn.stride(0), qn.stride(1), qn.stride(2),
+      kn.stride(0), kn.stride(1), kn.stride(2),
+      vn.stride(0), vn.stride(1), vn.stride(2));
   return {out, s_saved, ml, dmask};
 }
 
@@ -545,14 +565,21 @@ std::vector<torch::Tensor> attn_bwd_flash(
     double scale, double dropout_p, bool bias_grad, int64_t bias_dim) {
   const int B = q.size(0), H = q.size(1), Lq = q.size(2), D = q.size(3);
   const int Lk = k.size(2);
-  dout = dout.contiguous();
+  auto ok_strides = [](const torch::Tensor& t) {
+    return t.stride(3) == 1 && t.stride(0) % 8 == 0 &&
+           t.stride(1) % 8 == 0 && t.stride(2) % 8 == 0;
+  };
+  torch::Tensor don = ok_strides(dout) ? dout : dout.contiguous();
+  torch::Tensor qn = ok_strides(q) ? q : q.contiguous();
+  torch::Tensor kn = ok_strides(k) ? k : k.contiguous();
+  torch::Tensor vn = ok_strides(v) ? v : v.contiguous();
   // flash identity: dot_i = rowsum(dO * O) (tools/sim_flash_tiles.py)
-  auto dot_row = (dout.to(torch::kFloat32) * out.to(torch::kFloat32))
+  auto dot_row = (don.to(torch::kFloat32) * out.to(torch::kFloat32))
                      .sum(-1).contiguous();
   auto dq_f32 = torch::zeros({B, H, Lq, D},
                              q.options().dtype(torch::kFloat32));
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  auto dk = torch::empty({B, H, Lk, D}, k.options());
+  auto dv = torch::empty({B, H, Lk, D}, v.options());
   torch::Tensor ds_saved;
   float* ds_ptr = nullptr;
   if (bias_grad) {
@@ -568,10 +595,10 @@ std::vector<torch::Tensor> attn_bwd_flash(
   size_t smem = 8 * FTILE * 128;
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(attn_bwd_flash_kernel, grid, block, smem, stream,
-      reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(don.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(qn.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(kn.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(vn.data_ptr()),
       s_saved.data_ptr<float>(), ml.data_ptr<float>(),
       dot_row.data_ptr<float>(),
       query_mask.has_value() ? qm_f.data_ptr<float>() : nullptr,
@@ -579,7 +606,11 @@ std::vector<torch::Tensor> attn_bwd_flash(
       dq_f32.data_ptr<float>(),
       reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),
       reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()), ds_ptr,
-      B, H, Lq, Lk, D, (float)scale, (float)dropout_p);
+      B, H, Lq, Lk, D, (float)scale, (float)dropout_p,
+      don.stride(0), don.stride(1), don.stride(2),
+      qn.stride(0), qn.stride(1), qn.stride(2),
+      kn.stride(0), kn.stride(1), kn.stride(2),
+      vn.stride(0), vn.stride(1), vn.stride(2));
   auto dq = dq_f32.to(torch::kBFloat16);
   torch::Tensor dbias;
   if (bias_grad) {

@@ -170,8 +170,10 @@ class CobraDecoderLayer(nn.Module):
         b, l, d = x.shape
         qkv = self.qkv(x).view(b, l, 3, self.h, self.hd) \
             .permute(2, 0, 3, 1, 4)
+        # zero-copy: the flash/MFMA kernels are stride-aware, so the
+        # permuted qkv views pass straight through (round-2 late sweep)
         att = ops.fused_attention(
-            qkv[0].contiguous(), qkv[1].contiguous(), qkv[2].contiguous(),
+            qkv[0], qkv[1], qkv[2],
             scale=self.scale, causal=True, key_pad_mask=key_pad_mask,
             query_mask=query_mask, dropout_p=self.dropout_p,
             training=self.training)

@@ -55,8 +55,9 @@ class CaptureSafeEncoderLayer(nn.Module):
         b, l, d = x.shape
         qkv = self.qkv(x).view(b, l, 3, self.h, self.hd) \
             .permute(2, 0, 3, 1, 4)
+        # zero-copy: the attention kernels are stride-aware
         att = ops.fused_attention(
-            qkv[0].contiguous(), qkv[1].contiguous(), qkv[2].contiguous(),
+            qkv[0], qkv[1], qkv[2],
             scale=self.scale, key_pad_mask=src_key_padding_mask,
             dropout_p=self.dropout_p, training=self.training)
         att = self.out(att.transpose(1, 2).reshape(b, l, d))

@@ -195,7 +195,9 @@ class _FlashAttnFn(torch.autograd.Function):
                 scale, causal, dropout_p, training):
         from genrec_amd import ops
 
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        # stride-aware: [B,L,H,D] transpose views pass straight through
+        # (the host copies only when d isn't innermost / strides are
+        # unaligned), killing the q/k/v copy kernels per decoder layer
         seed_dev = None
         seed = 0
         if dropout_p > 0 and training:
@@ -220,7 +222,7 @@ class _FlashAttnFn(torch.autograd.Function):
         q, k, v, out, s_saved, ml, dmask, query_mask = ctx.saved_tensors
         scale, dropout_p, bias_grad, bias_dim, bias_dtype = ctx.meta
         dq, dk, dv, dbias = ops.ext().attn_bwd_flash(
-            dout.contiguous(), q, k, v, out, s_saved, ml, dmask,
+            dout, q, k, v, out, s_saved, ml, dmask,
             query_mask if query_mask.numel() else None,
             scale, dropout_p, bias_grad, bias_dim)
         if bias_grad and dbias.dtype != bias_dtype:
