@@ -153,7 +153,9 @@ def _parse_value(text: str) -> Any:
                 items.append("".join(cur))
             vals = [_parse_value(it) for it in items]
             return tuple(vals) if text.startswith("(") else vals
-        raise GinError(f"cannot parse value: {text!r}")
+        raise GinError(
+            f"cannot parse value: {text!r} (string values need quotes, "
+            f'e.g. --gin \'key="{text}"\' — same as gin-config)')
 
 
 def bind(key: str, value_text_or_value: Any, *, raw: bool = False) -> None:

@@ -288,3 +288,12 @@ def test_mkdocs_nav_files_exist():
     for entry in cfg["nav"]:
         for _, path in entry.items():
             assert os.path.exists(os.path.join(root, path)), path
+
+
+def test_unquoted_string_value_error_hints_quoting():
+    import pytest
+
+    from genrec_amd.config import ginlite
+
+    with pytest.raises(ginlite.GinError, match="need quotes"):
+        ginlite.bind("train.save_dir_root", "/tmp/nope")
